@@ -1,0 +1,181 @@
+/* Batched multi-set POA driver.
+ *
+ * The reference processes one read set at a time (abpoa_poa,
+ * abpoa_align.c:313-353); a single alignment cannot fill a 256-CU GPU, so the
+ * MI355X-native driver advances MANY independent read sets in lockstep
+ * "rounds": in round r every unfinished set aligns its r-th read against its
+ * own graph in ONE batched kernel launch (thousands of wavefronts), then the
+ * CIGAR folds + topological re-sorts run on host threads. Per-set results are
+ * identical to the sequential path: sets are fully independent
+ * (SURVEY.md §5/§8e).
+ */
+#include <pthread.h>
+#include "abpoa_amd.h"
+#include "abamd_util.h"
+
+/* from gpu_align.cpp */
+typedef struct {
+    abpoa_t *ab;
+    abpoa_para_t *abpt;
+    int beg_node_id, end_node_id;
+    uint8_t *query;
+    int qlen;
+    abpoa_res_t *res;
+} abamd_batch_job_t;
+int abamd_gpu_align_batch(abamd_batch_job_t *batch, int n_jobs);
+
+typedef struct {
+    abpoa_t *ab;
+    const int *seq_lens;
+    const uint8_t *const *seqs;
+    int n_seqs;
+    abpoa_res_t res;
+    int *weight_buf; int weight_cap;
+    int active; /* has a job this round */
+} set_state_t;
+
+typedef struct {
+    set_state_t *sets;
+    int n_sets;
+    abpoa_para_t *abpt;
+    int round;
+    int next;             /* work-stealing cursor */
+    pthread_mutex_t mu;
+} fold_work_t;
+
+static int *ones_weight(set_state_t *st, int len) {
+    if (len > st->weight_cap) {
+        st->weight_buf = (int*)abamd_realloc(st->weight_buf, (size_t)len * sizeof(int));
+        for (int i = st->weight_cap; i < len; ++i) st->weight_buf[i] = 1;
+        for (int i = 0; i < len; ++i) st->weight_buf[i] = 1;
+        st->weight_cap = len;
+    }
+    return st->weight_buf;
+}
+
+static void fold_one(fold_work_t *w, int si) {
+    set_state_t *st = &w->sets[si];
+    int r = w->round;
+    if (r >= st->n_seqs) return;
+    int qlen = st->seq_lens[r];
+    uint8_t *q = (uint8_t*)st->seqs[r];
+    abpoa_add_graph_alignment(st->ab, w->abpt, q, ones_weight(st, qlen), qlen, NULL,
+                              st->res, r, st->n_seqs, 1);
+    if (st->res.n_cigar) { free(st->res.graph_cigar); st->res.graph_cigar = NULL; st->res.n_cigar = 0; }
+}
+
+static void *fold_worker(void *arg) {
+    fold_work_t *w = (fold_work_t*)arg;
+    for (;;) {
+        pthread_mutex_lock(&w->mu);
+        int si = w->next++;
+        pthread_mutex_unlock(&w->mu);
+        if (si >= w->n_sets) break;
+        if (w->sets[si].active) fold_one(w, si);
+    }
+    return NULL;
+}
+
+typedef struct {
+    set_state_t *sets;
+    int n_sets, next;
+    abpoa_para_t *abpt;
+    abpoa_amd_cons_cb cb; void *user;
+    pthread_mutex_t mu;
+} cons_work_t;
+
+static void *cons_worker(void *arg) {
+    cons_work_t *w = (cons_work_t*)arg;
+    for (;;) {
+        pthread_mutex_lock(&w->mu);
+        int si = w->next++;
+        pthread_mutex_unlock(&w->mu);
+        if (si >= w->n_sets) break;
+        w->sets[si].ab->abs->n_seq = w->sets[si].n_seqs; /* consensus reads n_seq */
+        abpoa_generate_consensus(w->sets[si].ab, w->abpt);
+    }
+    return NULL;
+}
+
+int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
+                        const int *const *seq_lens, const uint8_t *const *const *seqs,
+                        abpoa_amd_cons_cb cb, void *user, int n_host_threads) {
+    if (n_sets <= 0) return 0;
+    if (n_host_threads < 1) n_host_threads = 1;
+    int i, r, max_reads = 0;
+    set_state_t *sets = (set_state_t*)abamd_calloc(n_sets, sizeof(set_state_t));
+    for (i = 0; i < n_sets; ++i) {
+        sets[i].ab = abpoa_init();
+        abpoa_reset(sets[i].ab, abpt, 1024);
+        sets[i].seq_lens = seq_lens[i];
+        sets[i].seqs = seqs[i];
+        sets[i].n_seqs = n_seqs[i];
+        if (n_seqs[i] > max_reads) max_reads = n_seqs[i];
+    }
+    abamd_batch_job_t *jobs = (abamd_batch_job_t*)abamd_malloc((size_t)n_sets * sizeof(abamd_batch_job_t));
+    int *job_set = (int*)abamd_malloc((size_t)n_sets * sizeof(int));
+    pthread_t *tids = (pthread_t*)abamd_malloc((size_t)n_host_threads * sizeof(pthread_t));
+
+    /* GPU memory budget per launch (arena is the dominant term) */
+    double mem_gb = 48.0;
+    {
+        const char *s = getenv("ABPOA_AMD_MEM_GB");
+        if (s && *s) mem_gb = atof(s);
+    }
+    const double budget_bytes = mem_gb * 1e9;
+
+    for (r = 0; r < max_reads; ++r) {
+        /* collect this round's alignment jobs */
+        int n_jobs = 0;
+        for (i = 0; i < n_sets; ++i) {
+            sets[i].active = 0;
+            if (r >= sets[i].n_seqs) continue;
+            sets[i].active = 1;
+            memset(&sets[i].res, 0, sizeof(abpoa_res_t));
+            if (sets[i].ab->abg->node_n <= 2) continue; /* first read folds directly */
+            if (sets[i].ab->abg->is_topological_sorted == 0)
+                abpoa_topological_sort(sets[i].ab->abg, abpt);
+            abamd_batch_job_t *J = &jobs[n_jobs];
+            J->ab = sets[i].ab; J->abpt = abpt;
+            J->beg_node_id = ABPOA_SRC_NODE_ID; J->end_node_id = ABPOA_SINK_NODE_ID;
+            J->query = (uint8_t*)sets[i].seqs[r];
+            J->qlen = sets[i].seq_lens[r];
+            J->res = &sets[i].res;
+            job_set[n_jobs] = i;
+            ++n_jobs;
+        }
+        /* launch in memory-bounded chunks */
+        int done = 0;
+        while (done < n_jobs) {
+            double acc = 0.0;
+            int take = 0;
+            while (done + take < n_jobs) {
+                abamd_batch_job_t *J = &jobs[done + take];
+                int w = abpt->wb < 0 ? J->qlen : abpt->wb + (int)(abpt->wf * J->qlen);
+                double est = (double)(J->ab->abg->node_n) * (2.0 * w + 160.0) * 5 * 2; /* int16 bytes */
+                if (take > 0 && acc + est > budget_bytes) break;
+                acc += est; ++take;
+            }
+            abamd_gpu_align_batch(&jobs[done], take);
+            done += take;
+        }
+        /* fold on host threads */
+        fold_work_t fw = { sets, n_sets, abpt, r, 0, PTHREAD_MUTEX_INITIALIZER };
+        int nt = n_host_threads;
+        for (i = 0; i < nt; ++i) pthread_create(&tids[i], NULL, fold_worker, &fw);
+        for (i = 0; i < nt; ++i) pthread_join(tids[i], NULL);
+    }
+
+    /* consensus on host threads, then emit callbacks in order */
+    cons_work_t cw = { sets, n_sets, 0, abpt, cb, user, PTHREAD_MUTEX_INITIALIZER };
+    for (i = 0; i < n_host_threads; ++i) pthread_create(&tids[i], NULL, cons_worker, &cw);
+    for (i = 0; i < n_host_threads; ++i) pthread_join(tids[i], NULL);
+    if (cb) for (i = 0; i < n_sets; ++i) cb(i, sets[i].ab->abc, user);
+
+    for (i = 0; i < n_sets; ++i) {
+        free(sets[i].weight_buf);
+        abpoa_free(sets[i].ab);
+    }
+    free(sets); free(jobs); free(job_set); free(tids);
+    return 0;
+}

@@ -1,0 +1,458 @@
+/* Host side of the CDNA4 aligner core.
+ *
+ * Packs one alignment job (compact reachable rows of the topo-sorted graph +
+ * query + parameters) into a device slab, launches the DP+backtrack kernel
+ * (gpu_kernels.hip), and unpacks abpoa_res_t. Fails loudly when no AMD GPU is
+ * usable — there is no CPU fallback on this path.
+ *
+ * Width selection and inf_min mirror the reference dispatch
+ * (abpoa_align_simd.c:1284-1302): int16 unless the score bound overflows,
+ * then int32. Arena overflow (adaptive band wider than the reserved slab)
+ * retries with a doubled arena.
+ *
+ * Batching: abamd_gpu_align_batch() packs many independent jobs into one
+ * launch so the 256-CU chip sees thousands of wavefronts; the single-job seam
+ * entry is a batch of one.
+ */
+#include <hip/hip_runtime.h>
+#include <atomic>
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <vector>
+#include <cmath>
+#include "abpoa_amd.h"
+#include "gpu_core.h"
+
+#define HIP_CHECK(x) do { hipError_t _e = (x); if (_e != hipSuccess) { \
+    fprintf(stderr, "[abpoa_amd] HIP error %s at %s:%d: %s\n", hipGetErrorName(_e), __FILE__, __LINE__, hipGetErrorString(_e)); \
+    exit(EXIT_FAILURE); } } while (0)
+
+static std::atomic<uint64_t> g_dp_cells{0}, g_kernel_ns{0}, g_launches{0};
+
+extern "C" void abpoa_amd_get_stats(uint64_t *dp_cells, uint64_t *kernel_ns, uint64_t *n_launches) {
+    if (dp_cells) *dp_cells = g_dp_cells.load();
+    if (kernel_ns) *kernel_ns = g_kernel_ns.load();
+    if (n_launches) *n_launches = g_launches.load();
+}
+extern "C" void abpoa_amd_reset_stats(void) { g_dp_cells = 0; g_kernel_ns = 0; g_launches = 0; }
+
+/* ---------------- device context (per thread) ---------------- */
+
+namespace {
+
+struct DevBuf {
+    void *p = nullptr;
+    size_t cap = 0;
+    void ensure(size_t n) {
+        if (n <= cap) return;
+        size_t want = cap ? cap : 4096;
+        while (want < n) want <<= 1;
+        if (p) HIP_CHECK(hipFree(p));
+        HIP_CHECK(hipMalloc(&p, want));
+        cap = want;
+    }
+};
+
+struct HostBuf {
+    std::vector<uint8_t> v;
+    size_t used = 0;
+    void reset() { used = 0; }
+    /* reserve n bytes aligned to 16 and return the offset */
+    size_t alloc(size_t n) {
+        size_t off = (used + 15) & ~(size_t)15;
+        used = off + n;
+        if (used > v.size()) v.resize(used * 2 + 4096);
+        return off;
+    }
+    uint8_t *at(size_t off) { return v.data() + off; }
+};
+
+struct GpuCtx {
+    bool init = false;
+    hipStream_t stream;
+    hipEvent_t ev0, ev1;
+    DevBuf slab;        /* all per-job inputs + metadata */
+    DevBuf arena;       /* DP plane arena */
+    DevBuf jobs, results, cigars;
+    HostBuf hb;
+    std::vector<abamd_gpu_job_t> hjobs;
+    std::vector<abamd_gpu_res_t> hres;
+    std::vector<uint64_t> hcig;
+    void ensure_init() {
+        if (init) return;
+        int n = 0;
+        hipError_t e = hipGetDeviceCount(&n);
+        if (e != hipSuccess || n == 0) {
+            fprintf(stderr, "[abpoa_amd] FATAL: no usable AMD GPU (hipGetDeviceCount: %s). "
+                            "The abpoa_amd aligner is GPU-only; there is no CPU fallback.\n",
+                    hipGetErrorString(e));
+            exit(EXIT_FAILURE);
+        }
+        HIP_CHECK(hipStreamCreate(&stream));
+        HIP_CHECK(hipEventCreate(&ev0));
+        HIP_CHECK(hipEventCreate(&ev1));
+        init = true;
+    }
+};
+
+thread_local GpuCtx g_ctx;
+
+/* host-side packed job (offsets into the slab) */
+struct JobPack {
+    abamd_gpu_job_t jb;     /* device pointers filled after upload */
+    size_t o_query, o_base, o_nodeid, o_preoff, o_preidx, o_preps, o_outoff, o_outidx, o_remain;
+    size_t o_ml, o_mr, o_beg, o_end, o_roff;
+    int64_t arena_off;      /* cells */
+    int cigar_off;          /* entries */
+    int n_rows, qlen;
+    int n_pre, n_out;
+};
+
+/* Build the compact-row job into ctx.hb; returns per-job arena demand (cells). */
+static int64_t pack_job(GpuCtx &C, JobPack &P, abpoa_t *ab, abpoa_para_t *abpt,
+                        int beg_node_id, int end_node_id, uint8_t *query, int qlen) {
+    abpoa_graph_t *g = ab->abg;
+    int beg_index = g->node_id_to_index[beg_node_id];
+    int end_index = g->node_id_to_index[end_node_id];
+    int span = end_index - beg_index + 1;
+
+    /* reachability map (abpoa_align_simd.c:1259-1269) */
+    std::vector<uint8_t> imap(g->node_n, 0);
+    imap[beg_index] = imap[end_index] = 1;
+    for (int i = beg_index; i < end_index - 1; ++i) {
+        if (!imap[i]) continue;
+        int nid = g->index_to_node_id[i];
+        for (int j = 0; j < g->node[nid].out_edge_n; ++j)
+            imap[g->node_id_to_index[g->node[nid].out_id[j]]] = 1;
+    }
+    /* compact row numbering */
+    std::vector<int> idx2row(span, -1);
+    int n_rows = 0;
+    for (int i = 0; i < span; ++i)
+        if (imap[beg_index + i]) idx2row[i] = n_rows++;
+
+    P.n_rows = n_rows; P.qlen = qlen;
+
+    size_t o_query = C.hb.alloc(qlen);
+    size_t o_base = C.hb.alloc(n_rows);
+    size_t o_nodeid = C.hb.alloc((size_t)n_rows * 4);
+    size_t o_preoff = C.hb.alloc((size_t)(n_rows + 1) * 4);
+    size_t o_outoff = C.hb.alloc((size_t)(n_rows + 1) * 4);
+    size_t o_remain = C.hb.alloc((size_t)n_rows * 4);
+    memcpy(C.hb.at(o_query), query, qlen);
+    uint8_t *base = C.hb.at(o_base);
+    int *nodeid = (int*)C.hb.at(o_nodeid);
+    int *preoff = (int*)C.hb.at(o_preoff);
+    int *outoff = (int*)C.hb.at(o_outoff);
+    int *remain = (int*)C.hb.at(o_remain);
+
+    int n_pre = 0, n_out = 0;
+    for (int i = 0, r = 0; i < span; ++i) {
+        if (idx2row[i] < 0) continue;
+        int nid = g->index_to_node_id[beg_index + i];
+        base[r] = g->node[nid].base;
+        nodeid[r] = nid;
+        remain[r] = (abpt->wb >= 0 || abpt->zdrop > 0) ? g->node_id_to_max_remain[nid] : 0;
+        preoff[r] = n_pre; outoff[r] = n_out;
+        if (r > 0) {
+            for (int j = 0; j < g->node[nid].in_edge_n; ++j) {
+                int pidx = g->node_id_to_index[g->node[nid].in_id[j]];
+                if (pidx >= beg_index && pidx <= end_index && imap[pidx] && idx2row[pidx - beg_index] >= 0)
+                    ++n_pre;
+            }
+        }
+        for (int j = 0; j < g->node[nid].out_edge_n; ++j) {
+            int oidx = g->node_id_to_index[g->node[nid].out_id[j]];
+            if (oidx >= beg_index && oidx <= end_index && imap[oidx] && idx2row[oidx - beg_index] >= 0)
+                ++n_out;
+        }
+        ++r;
+    }
+    preoff[n_rows] = n_pre; outoff[n_rows] = n_out;
+
+    size_t o_preidx = C.hb.alloc((size_t)(n_pre > 0 ? n_pre : 1) * 4);
+    size_t o_preps = C.hb.alloc((size_t)(n_pre > 0 ? n_pre : 1) * 4);
+    size_t o_outidx = C.hb.alloc((size_t)(n_out > 0 ? n_out : 1) * 4);
+    int *preidx = (int*)C.hb.at(o_preidx);
+    int *preps = (int*)C.hb.at(o_preps);
+    int *outidx = (int*)C.hb.at(o_outidx);
+    /* re-read offsets (hb.alloc may have resized the vector) */
+    base = C.hb.at(o_base); nodeid = (int*)C.hb.at(o_nodeid);
+    preoff = (int*)C.hb.at(o_preoff); outoff = (int*)C.hb.at(o_outoff);
+    remain = (int*)C.hb.at(o_remain);
+
+    n_pre = 0; n_out = 0;
+    for (int i = 0, r = 0; i < span; ++i) {
+        if (idx2row[i] < 0) continue;
+        int nid = g->index_to_node_id[beg_index + i];
+        if (r > 0) {
+            for (int j = 0; j < g->node[nid].in_edge_n; ++j) {
+                int pidx = g->node_id_to_index[g->node[nid].in_id[j]];
+                if (pidx >= beg_index && pidx <= end_index && imap[pidx] && idx2row[pidx - beg_index] >= 0) {
+                    preidx[n_pre] = idx2row[pidx - beg_index];
+                    preps[n_pre] = 0; /* inc_path_score handled below */
+                    if (abpt->inc_path_score) {
+                        /* abpoa_get_incre_path_score (abpoa_graph.c:429-437) */
+                        int pre_id = g->node[nid].in_id[j], node_w = 0;
+                        for (int t = 0; t < g->node[pre_id].out_edge_n; ++t) node_w += g->node[pre_id].out_edge_weight[t];
+                        int edge_w = g->node[nid].in_edge_weight[j];
+                        int sc = 0;
+                        if (node_w != 0 && edge_w != 0) {
+                            double r2 = (double)edge_w / (double)node_w;
+                            sc = (int)lround(log(r2));
+                            if (sc < -20) sc = -20;
+                        }
+                        preps[n_pre] = sc;
+                    }
+                    ++n_pre;
+                }
+            }
+        }
+        for (int j = 0; j < g->node[nid].out_edge_n; ++j) {
+            int oidx = g->node_id_to_index[g->node[nid].out_id[j]];
+            if (oidx >= beg_index && oidx <= end_index && imap[oidx] && idx2row[oidx - beg_index] >= 0)
+                outidx[n_out++] = idx2row[oidx - beg_index];
+        }
+        ++r;
+    }
+
+    /* working arrays (device-initialized) */
+    size_t o_ml = C.hb.alloc((size_t)n_rows * 4);
+    size_t o_mr = C.hb.alloc((size_t)n_rows * 4);
+    size_t o_beg = C.hb.alloc((size_t)n_rows * 4);
+    size_t o_end = C.hb.alloc((size_t)n_rows * 4);
+    size_t o_roff = C.hb.alloc((size_t)n_rows * 8);
+
+    P.o_query = o_query; P.o_base = o_base; P.o_nodeid = o_nodeid;
+    P.o_preoff = o_preoff; P.o_preidx = o_preidx; P.o_preps = o_preps;
+    P.o_outoff = o_outoff; P.o_outidx = o_outidx; P.o_remain = o_remain;
+    P.o_ml = o_ml; P.o_mr = o_mr; P.o_beg = o_beg; P.o_end = o_end; P.o_roff = o_roff;
+    P.n_pre = n_pre; P.n_out = n_out;
+
+    abamd_gpu_job_t &jb = P.jb;
+    memset(&jb, 0, sizeof(jb));
+    jb.n_rows = n_rows; jb.qlen = qlen; jb.m = abpt->m;
+    jb.w = abpt->wb < 0 ? qlen : abpt->wb + (int)(abpt->wf * qlen);
+    jb.banded = abpt->wb >= 0;
+    jb.o1 = abpt->gap_open1; jb.e1 = abpt->gap_ext1;
+    jb.o2 = abpt->gap_open2; jb.e2 = abpt->gap_ext2;
+    jb.oe1 = abpt->gap_open1 + abpt->gap_ext1;
+    jb.oe2 = abpt->gap_open2 + abpt->gap_ext2;
+    jb.align_mode = abpt->align_mode;
+    jb.put_gap_on_right = abpt->put_gap_on_right;
+    jb.put_gap_at_end = abpt->put_gap_at_end;
+    jb.zdrop = abpt->zdrop;
+    jb.inc_path_score = abpt->inc_path_score;
+    jb.node_n_init = g->node_n;
+    jb.ret_cigar = abpt->ret_cigar;
+
+    /* arena demand estimate: adaptive bands are ~2w + drift; reserve slack */
+    int64_t est = (int64_t)n_rows * (2 * (int64_t)jb.w + 160) + qlen + 64;
+    return est;
+}
+
+struct BatchJob {
+    abpoa_t *ab;
+    abpoa_para_t *abpt;
+    int beg_node_id, end_node_id;
+    uint8_t *query;
+    int qlen;
+    abpoa_res_t *res;
+};
+
+/* score-width pick (abpoa_align_simd.c:1284-1302); assumes uniform paras */
+static void pick_width(abpoa_para_t *abpt, int qlen, int gn, int *bits, int *inf_min) {
+    int32_t gap_oe1 = abpt->gap_open1 + abpt->gap_ext1;
+    int32_t gap_oe2 = abpt->gap_open2 + abpt->gap_ext2;
+    int len = qlen > gn ? qlen : gn;
+    int32_t max_score = (int32_t)qlen * abpt->max_mat;
+    int32_t alt = (int32_t)len * abpt->gap_ext1 + abpt->gap_open1;
+    if (alt > max_score) max_score = alt;
+    int32_t ext_max = abpt->gap_ext1 > abpt->gap_ext2 ? abpt->gap_ext1 : abpt->gap_ext2;
+    if (max_score <= INT16_MAX - abpt->min_mis - gap_oe1 - gap_oe2) {
+        int32_t im = INT16_MIN + abpt->min_mis;
+        if (INT16_MIN + gap_oe1 > im) im = INT16_MIN + gap_oe1;
+        if (INT16_MIN + gap_oe2 > im) im = INT16_MIN + gap_oe2;
+        *inf_min = im + 512 * ext_max; *bits = 16;
+    } else {
+        int32_t im = INT32_MIN + abpt->min_mis;
+        if (INT32_MIN + gap_oe1 > im) im = INT32_MIN + gap_oe1;
+        if (INT32_MIN + gap_oe2 > im) im = INT32_MIN + gap_oe2;
+        *inf_min = im + 512 * ext_max; *bits = 32;
+    }
+}
+
+} // namespace
+
+/* Align a batch of independent jobs in one kernel launch. Returns 0.
+ * All jobs must share gap_mode/align_mode (they do: one abpoa_para_t per run). */
+extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs);
+
+extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
+    if (n_jobs <= 0) return 0;
+    GpuCtx &C = g_ctx;
+    C.ensure_init();
+    abpoa_para_t *abpt = batch[0].abpt;
+    if (abpt->gap_mode != ABPOA_CONVEX_GAP || abpt->align_mode != ABPOA_GLOBAL_MODE) {
+        fprintf(stderr, "[abpoa_amd] GPU core: gap_mode %d / align_mode %d not implemented yet "
+                        "(convex-gap global alignment only in this build)\n",
+                abpt->gap_mode, abpt->align_mode);
+        exit(EXIT_FAILURE);
+    }
+
+    C.hb.reset();
+    C.hjobs.resize(n_jobs);
+    C.hres.resize(n_jobs);
+    std::vector<JobPack> packs(n_jobs);
+    std::vector<int64_t> arena_est(n_jobs);
+
+    /* pack matrix once (shared paras) */
+    size_t o_mat = C.hb.alloc((size_t)abpt->m * abpt->m * 4);
+    memcpy(C.hb.at(o_mat), abpt->mat, (size_t)abpt->m * abpt->m * 4);
+
+    int bits_max = 16;
+    for (int i = 0; i < n_jobs; ++i) {
+        BatchJob &B = batch[i];
+        arena_est[i] = pack_job(C, packs[i], B.ab, B.abpt, B.beg_node_id, B.end_node_id, B.query, B.qlen);
+        int bits, inf_min;
+        int span = B.ab->abg->node_id_to_index[B.end_node_id] - B.ab->abg->node_id_to_index[B.beg_node_id] + 1;
+        pick_width(B.abpt, B.qlen, span, &bits, &inf_min);
+        packs[i].jb.inf_min = inf_min;
+        if (bits > bits_max) bits_max = bits;
+        packs[i].cigar_off = 0; /* filled below */
+    }
+    /* a mixed batch runs at the widest type; widths are identical across jobs
+     * of one workload in practice (same scoring paras, similar qlen) */
+    int bits = bits_max;
+    if (bits == 32) {
+        /* recompute inf_min at 32-bit for every job */
+        for (int i = 0; i < n_jobs; ++i) {
+            int b2, im;
+            BatchJob &B = batch[i];
+            int span = B.ab->abg->node_id_to_index[B.end_node_id] - B.ab->abg->node_id_to_index[B.beg_node_id] + 1;
+            (void)span; (void)b2;
+            int32_t gap_oe1 = abpt->gap_open1 + abpt->gap_ext1, gap_oe2 = abpt->gap_open2 + abpt->gap_ext2;
+            int32_t ext_max = abpt->gap_ext1 > abpt->gap_ext2 ? abpt->gap_ext1 : abpt->gap_ext2;
+            im = INT32_MIN + abpt->min_mis;
+            if (INT32_MIN + gap_oe1 > im) im = INT32_MIN + gap_oe1;
+            if (INT32_MIN + gap_oe2 > im) im = INT32_MIN + gap_oe2;
+            packs[i].jb.inf_min = im + 512 * ext_max;
+        }
+    }
+    const size_t ssz = bits == 16 ? 2 : 4;
+
+    int attempt = 0;
+    for (;;) {
+        /* arena layout */
+        int64_t arena_cells = 0;
+        for (int i = 0; i < n_jobs; ++i) {
+            packs[i].arena_off = arena_cells;
+            arena_cells += arena_est[i];
+        }
+        C.arena.ensure((size_t)arena_cells * 5 * ssz);
+
+        /* cigar buffers */
+        int cig_total = 0;
+        for (int i = 0; i < n_jobs; ++i) {
+            packs[i].cigar_off = cig_total;
+            cig_total += 2 * packs[i].qlen + 1024;
+        }
+        C.cigars.ensure((size_t)cig_total * 8);
+
+        /* upload slab + jobs */
+        C.slab.ensure(C.hb.used);
+        HIP_CHECK(hipMemcpyAsync(C.slab.p, C.hb.v.data(), C.hb.used, hipMemcpyHostToDevice, C.stream));
+        uint8_t *S = (uint8_t*)C.slab.p;
+        for (int i = 0; i < n_jobs; ++i) {
+            JobPack &P = packs[i];
+            abamd_gpu_job_t &jb = P.jb;
+            jb.query = S + P.o_query;
+            jb.row_base = S + P.o_base;
+            jb.row_node_id = (int*)(S + P.o_nodeid);
+            jb.pre_off = (int*)(S + P.o_preoff);
+            jb.pre_idx = (int*)(S + P.o_preidx);
+            jb.pre_ps = (int*)(S + P.o_preps);
+            jb.out_off = (int*)(S + P.o_outoff);
+            jb.out_idx = (int*)(S + P.o_outidx);
+            jb.max_remain = (int*)(S + P.o_remain);
+            jb.max_left = (int*)(S + P.o_ml);
+            jb.max_right = (int*)(S + P.o_mr);
+            jb.dp_beg = (int*)(S + P.o_beg);
+            jb.dp_end = (int*)(S + P.o_end);
+            jb.row_off = (int64_t*)(S + P.o_roff);
+            jb.mat = (int*)(S + o_mat);
+            jb.arena = (uint8_t*)C.arena.p + (size_t)P.arena_off * 5 * ssz;
+            jb.arena_cap = arena_est[i];
+            jb.cigar = (uint64_t*)C.cigars.p + P.cigar_off;
+            jb.cigar_cap = 2 * P.qlen + 1024;
+            C.hjobs[i] = jb;
+        }
+        C.jobs.ensure((size_t)n_jobs * sizeof(abamd_gpu_job_t));
+        C.results.ensure((size_t)n_jobs * sizeof(abamd_gpu_res_t));
+        HIP_CHECK(hipMemcpyAsync(C.jobs.p, C.hjobs.data(), (size_t)n_jobs * sizeof(abamd_gpu_job_t), hipMemcpyHostToDevice, C.stream));
+
+        HIP_CHECK(hipEventRecord(C.ev0, C.stream));
+        if (bits == 16)
+            abamd_launch_cg_i16((abamd_gpu_job_t*)C.jobs.p, (abamd_gpu_res_t*)C.results.p, n_jobs, C.stream);
+        else
+            abamd_launch_cg_i32((abamd_gpu_job_t*)C.jobs.p, (abamd_gpu_res_t*)C.results.p, n_jobs, C.stream);
+        HIP_CHECK(hipGetLastError());
+        HIP_CHECK(hipEventRecord(C.ev1, C.stream));
+        HIP_CHECK(hipMemcpyAsync(C.hres.data(), C.results.p, (size_t)n_jobs * sizeof(abamd_gpu_res_t), hipMemcpyDeviceToHost, C.stream));
+        HIP_CHECK(hipStreamSynchronize(C.stream));
+
+        float ms = 0.f;
+        HIP_CHECK(hipEventElapsedTime(&ms, C.ev0, C.ev1));
+        g_kernel_ns += (uint64_t)(ms * 1e6);
+        g_launches += 1;
+
+        /* retry on arena overflow with doubled reservations */
+        bool overflow = false;
+        for (int i = 0; i < n_jobs; ++i)
+            if (C.hres[i].status == ABAMD_JOB_ARENA_OVERFLOW) { arena_est[i] *= 2; overflow = true; }
+        if (!overflow) break;
+        if (++attempt > 8) {
+            fprintf(stderr, "[abpoa_amd] arena overflow persists after %d retries\n", attempt);
+            exit(EXIT_FAILURE);
+        }
+    }
+
+    /* unpack results */
+    C.hcig.clear();
+    for (int i = 0; i < n_jobs; ++i) {
+        abamd_gpu_res_t &R = C.hres[i];
+        if (R.status != ABAMD_JOB_OK) {
+            fprintf(stderr, "[abpoa_amd] GPU job %d failed with status %d\n", i, R.status);
+            exit(EXIT_FAILURE);
+        }
+        g_dp_cells += (uint64_t)R.cells;
+        abpoa_res_t *res = batch[i].res;
+        res->best_score = R.best_score;
+        if (batch[i].abpt->ret_cigar && R.n_cigar > 0) {
+            uint64_t *tmp = (uint64_t*)malloc((size_t)R.n_cigar * 8);
+            HIP_CHECK(hipMemcpy(tmp, (uint64_t*)C.cigars.p + packs[i].cigar_off, (size_t)R.n_cigar * 8, hipMemcpyDeviceToHost));
+            if (!batch[i].abpt->rev_cigar) { /* reverse to front-to-back order */
+                for (int a = 0, b = R.n_cigar - 1; a < b; ++a, --b) {
+                    uint64_t t = tmp[a]; tmp[a] = tmp[b]; tmp[b] = t;
+                }
+            }
+            res->graph_cigar = tmp;
+            res->n_cigar = R.n_cigar; res->m_cigar = R.n_cigar;
+        } else { res->n_cigar = 0; res->graph_cigar = nullptr; }
+        res->n_aln_bases += R.n_aln_bases;
+        res->n_matched_bases += R.n_matched_bases;
+        res->node_s = R.node_s; res->node_e = R.node_e;
+        res->query_s = R.query_s; res->query_e = R.query_e;
+    }
+    return 0;
+}
+
+extern "C" int abamd_gpu_align_sequence_to_subgraph(abpoa_t *ab, abpoa_para_t *abpt,
+        int beg_node_id, int end_node_id, uint8_t *query, int qlen, abpoa_res_t *res) {
+    BatchJob b;
+    b.ab = ab; b.abpt = abpt;
+    b.beg_node_id = beg_node_id; b.end_node_id = end_node_id;
+    b.query = query; b.qlen = qlen; b.res = res;
+    return abamd_gpu_align_batch(&b, 1);
+}

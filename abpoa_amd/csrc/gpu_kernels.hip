@@ -1,0 +1,423 @@
+/* CDNA4 (gfx950) aligner core: adaptive-banded sequence-to-graph DP.
+ *
+ * One wavefront (64 lanes) per alignment job. The wave sweeps the graph's
+ * topologically-sorted rows; within a row the band is processed in 64-cell
+ * chunks held in registers:
+ *   - M/E gather from predecessor rows (coalesced 2- or 4-byte loads from the
+ *     banded HBM arena),
+ *   - the F (insertion) recurrence is a wavefront max-plus log-scan
+ *     (__shfl_up, 6 steps) with a sequential carry between chunks,
+ *   - the five planes (H,E1,E2,F1,F2) stream back to the arena with coalesced
+ *     stores (they are re-read once by the in-kernel backtrack),
+ *   - the row argmax (adaptive band steering) is a wavefront reduction.
+ * Integer max-plus throughout: MFMA does not apply; the kernel is designed to
+ * be HBM-bound on the 10 B/cell plane traffic.
+ *
+ * Numerics: results are bit-identical to oracle/ref_core.c (and therefore to
+ * the reference x86 build): int16 arithmetic wraps (non-saturating), achieved
+ * by truncating to the score type after every add/sub. Recurrence and
+ * backtrack follow abpoa_align_simd.c:935-1074 / :309-458; band formulas
+ * abpoa_align.h:34-35.
+ *
+ * This file covers global alignment with convex gaps (the north-star path),
+ * int16 with int32 overflow rescore. Local/extension and affine/linear gap
+ * variants are host-dispatched and added next.
+ */
+#include <hip/hip_runtime.h>
+#include "gpu_core.h"
+
+#define WAVE 64
+
+template <typename S> struct ScoreLim;
+template <> struct ScoreLim<int16_t> { static constexpr int maxv = 32767; };
+template <> struct ScoreLim<int32_t> { static constexpr int maxv = 2147483647; };
+
+template <typename S>
+__device__ __forceinline__ S smax(S a, S b) { return a > b ? a : b; }
+
+/* device push_cigar, matching abpoa_align.h:54-73 (run-length merge for I/S/H) */
+__device__ static int dev_push_cigar(uint64_t *cig, int *n_c, int cap, int op, int len,
+                                     int node_id, int query_id, int *status) {
+    uint64_t l = (uint64_t)len;
+    if (*n_c == 0 || (op != 1 /*I*/ && op != 4 && op != 5) || op != (int)(cig[*n_c - 1] & 0xf)) {
+        if (*n_c >= cap) { *status = ABAMD_JOB_CIGAR_OVERFLOW; return -1; }
+        uint64_t n_id = (uint64_t)(uint32_t)node_id, q_id = (uint64_t)(uint32_t)query_id;
+        if (op == 0 || op == 3) cig[(*n_c)++] = n_id << 34 | q_id << 4 | (uint64_t)op;
+        else if (op == 1 || op == 4 || op == 5) cig[(*n_c)++] = q_id << 34 | l << 4 | (uint64_t)op;
+        else cig[(*n_c)++] = n_id << 34 | l << 4 | (uint64_t)op; /* D */
+    } else cig[*n_c - 1] += l << 4;
+    return 0;
+}
+
+/* ------------------------------------------------------------------ */
+/* The DP + backtrack kernel (global alignment, convex gaps).          */
+/* ------------------------------------------------------------------ */
+template <typename S>
+__global__ __launch_bounds__(WAVE)
+void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
+                      abamd_gpu_res_t *__restrict__ results, int n_jobs) {
+    const int jid = blockIdx.x;
+    if (jid >= n_jobs) return;
+    const abamd_gpu_job_t jb = jobs[jid];
+    abamd_gpu_res_t *res = &results[jid];
+    const int lane = threadIdx.x;
+
+    __shared__ int mat_lds[27 * 27];
+    const int m = jb.m;
+    for (int i = lane; i < m * m; i += WAVE) mat_lds[i] = jb.mat[i];
+    /* single wave: no barrier needed; LDS writes are visible after lgkmcnt */
+    __builtin_amdgcn_s_waitcnt(0); /* drain lds+vm counters conservatively */
+
+    const int qlen = jb.qlen, n_rows = jb.n_rows, w = jb.w;
+    const S inf_min = (S)jb.inf_min;
+    const S o1 = (S)jb.o1, o2 = (S)jb.o2, e1 = (S)jb.e1, e2 = (S)jb.e2;
+    const S oe1 = (S)jb.oe1, oe2 = (S)jb.oe2;
+    const int end_remain = jb.max_remain[n_rows - 1];
+    S *arena = (S*)jb.arena;
+    const uint8_t *__restrict__ query = jb.query;
+
+    if (lane == 0) {
+        res->status = ABAMD_JOB_OK;
+        res->n_cigar = 0;
+    }
+
+    /* adaptive band state init (abpoa_topological_sort:347-353 + first_dp) */
+    for (int i = lane; i < n_rows; i += WAVE) {
+        jb.max_left[i] = jb.node_n_init;
+        jb.max_right[i] = 0;
+    }
+    if (lane == 0) { jb.max_left[0] = 0; jb.max_right[0] = 0; }
+    for (int k = jb.out_off[0] + lane; k < jb.out_off[1]; k += WAVE) {
+        int o = jb.out_idx[k];
+        jb.max_left[o] = 1; jb.max_right[o] = 1;
+    }
+
+    /* ---- first row (simd_abpoa_cg_first_dp) ---- */
+    int64_t used;
+    {
+        int mr = jb.max_remain[0] - end_remain - 1;
+        int end0;
+        if (jb.banded) {
+            int t = jb.max_right[0] > qlen - mr ? jb.max_right[0] : qlen - mr;
+            end0 = (qlen < t + w) ? qlen : t + w;
+        } else end0 = qlen;
+        if (lane == 0) { jb.dp_beg[0] = 0; jb.dp_end[0] = end0; jb.row_off[0] = 0; }
+        int64_t bw = end0 + 1;
+        used = bw;
+        S *H = arena, *E1 = arena + bw, *E2 = arena + 2 * bw, *F1 = arena + 3 * bw, *F2 = arena + 4 * bw;
+        for (int j = lane; j <= end0; j += WAVE) {
+            if (j == 0) {
+                H[0] = 0; E1[0] = (S)(0 - oe1); E2[0] = (S)(0 - oe2);
+                F1[0] = inf_min; F2[0] = inf_min;
+            } else {
+                S f1 = (S)(-(jb.o1 + jb.e1 * j));
+                S f2 = (S)(-(jb.o2 + jb.e2 * j));
+                F1[j] = f1; F2[j] = f2;
+                H[j] = smax(f1, f2);
+                E1[j] = inf_min; E2[j] = inf_min;
+            }
+        }
+    }
+
+    /* ---- main row loop ---- */
+    for (int r = 1; r < n_rows - 1; ++r) {
+        const int pk0 = jb.pre_off[r], pk1 = jb.pre_off[r + 1];
+        int beg, end;
+        {
+            int mr = jb.max_remain[r] - end_remain - 1;
+            if (jb.banded) {
+                int ml = jb.max_left[r], mrr = jb.max_right[r];
+                int lo = ml < qlen - mr ? ml : qlen - mr;
+                beg = lo - w; if (beg < 0) beg = 0;
+                int hi = mrr > qlen - mr ? mrr : qlen - mr;
+                end = hi + w; if (end > qlen) end = qlen;
+                int min_pre_beg = 0x7fffffff;
+                for (int k = pk0; k < pk1; ++k) {
+                    int pb = jb.dp_beg[jb.pre_idx[k]];
+                    if (pb < min_pre_beg) min_pre_beg = pb;
+                }
+                if (beg < min_pre_beg) beg = min_pre_beg;
+            } else { beg = 0; end = qlen; }
+        }
+        const int64_t bw = end - beg + 1;
+        if (used + bw > jb.arena_cap) { if (lane == 0) res->status = ABAMD_JOB_ARENA_OVERFLOW; return; }
+        const int64_t off = used;
+        if (lane == 0) { jb.dp_beg[r] = beg; jb.dp_end[r] = end; jb.row_off[r] = off; }
+        used += bw;
+        S *H = arena + off * 5, *E1r = H + bw, *E2r = E1r + bw, *F1r = E2r + bw, *F2r = F1r + bw;
+        const uint8_t base = jb.row_base[r];
+        const int *mrow = &mat_lds[base * m];
+
+        S carry_h = inf_min, f1c = inf_min, f2c = inf_min;
+        S lmax = inf_min; int lleft = -1, lright = -1;
+
+        for (int cs = beg; cs <= end; cs += WAVE) {
+            const int j = cs + lane;
+            const bool act = j <= end;
+            S h = inf_min, e1v = inf_min, e2v = inf_min;
+            for (int k = pk0; k < pk1; ++k) {
+                const int p = jb.pre_idx[k];
+                const S ps = (S)jb.pre_ps[k];
+                const int pbeg = jb.dp_beg[p], pend = jb.dp_end[p];
+                const int64_t pbw = pend - pbeg + 1;
+                const S *__restrict__ pH = arena + jb.row_off[p] * 5;
+                const S *__restrict__ pE1 = pH + pbw;
+                const S *__restrict__ pE2 = pE1 + pbw;
+                if (act) {
+                    if (j - 1 >= pbeg && j - 1 <= pend) {
+                        S v = (S)(pH[j - 1 - pbeg] + ps);
+                        if (v > h) h = v;
+                    }
+                    if (j >= pbeg && j <= pend) {
+                        S v1 = (S)(pE1[j - pbeg] + ps);
+                        S v2 = (S)(pE2[j - pbeg] + ps);
+                        if (v1 > e1v) e1v = v1;
+                        if (v2 > e2v) e2v = v2;
+                    }
+                }
+            }
+            const S q = (S)((j == 0 || !act) ? 0 : mrow[query[j - 1]]);
+            S hpre = (S)(h + q);
+            hpre = smax(hpre, smax(e1v, e2v));
+            if (!act) hpre = inf_min;
+
+            /* F scan: candidates c[j] = max(Hpre[j-1]-oe, chain) */
+            S hshift = (S)__shfl_up((int)hpre, 1);
+            S c1, c2;
+            if (lane == 0) {
+                if (cs == beg) { c1 = (S)(inf_min - oe1); c2 = (S)(inf_min - oe2); }
+                else {
+                    c1 = smax((S)(carry_h - oe1), (S)(f1c - e1));
+                    c2 = smax((S)(carry_h - oe2), (S)(f2c - e2));
+                }
+            } else {
+                c1 = (S)(hshift - oe1);
+                c2 = (S)(hshift - oe2);
+            }
+            S f1 = c1, f2 = c2;
+            #pragma unroll
+            for (int sft = 1; sft < WAVE; sft <<= 1) {
+                S t1 = (S)__shfl_up((int)f1, sft);
+                S t2 = (S)__shfl_up((int)f2, sft);
+                if (lane >= sft) {
+                    f1 = smax(f1, (S)(t1 - (S)(sft * jb.e1)));
+                    f2 = smax(f2, (S)(t2 - (S)(sft * jb.e2)));
+                }
+            }
+            carry_h = (S)__shfl((int)hpre, WAVE - 1);
+            f1c = (S)__shfl((int)f1, WAVE - 1);
+            f2c = (S)__shfl((int)f2, WAVE - 1);
+
+            S hf = smax(hpre, smax(f1, f2));
+            S e1n = smax((S)(e1v - e1), (S)(hf - oe1));
+            S e2n = smax((S)(e2v - e2), (S)(hf - oe2));
+            if (act) {
+                H[j - beg] = hf; E1r[j - beg] = e1n; E2r[j - beg] = e2n;
+                F1r[j - beg] = f1; F2r[j - beg] = f2;
+                if (hf > lmax) { lmax = hf; lleft = j; lright = j; }
+                else if (hf == lmax) { lright = j; }
+            }
+        }
+
+        /* row argmax reduce + adaptive band push (simd_abpoa_max_in_row / ada_max_i) */
+        if (jb.banded) {
+            int mv = (int)lmax;
+            #pragma unroll
+            for (int sft = 32; sft >= 1; sft >>= 1) {
+                int o = __shfl_xor(mv, sft);
+                if (o > mv) mv = o;
+            }
+            int ll = ((int)lmax == mv && lleft >= 0) ? lleft : 0x7fffffff;
+            int rr = ((int)lmax == mv && lright >= 0) ? lright : -1;
+            #pragma unroll
+            for (int sft = 32; sft >= 1; sft >>= 1) {
+                int lo = __shfl_xor(ll, sft); if (lo < ll) ll = lo;
+                int ro = __shfl_xor(rr, sft); if (ro > rr) rr = ro;
+            }
+            for (int k = jb.out_off[r] + lane; k < jb.out_off[r + 1]; k += WAVE) {
+                int o = jb.out_idx[k];
+                if (rr + 1 > jb.max_right[o]) jb.max_right[o] = rr + 1;
+                if (ll + 1 < jb.max_left[o]) jb.max_left[o] = ll + 1;
+            }
+        }
+    }
+
+    if (lane == 0) res->cells = used;
+
+    /* ---- final best over the end row's predecessors + backtrack (lane 0) ---- */
+    if (lane != 0) return;
+
+    int32_t best_score = jb.inf_min;
+    int best_i = 0, best_j = 0;
+    for (int k = jb.pre_off[n_rows - 1]; k < jb.pre_off[n_rows]; ++k) {
+        const int p = jb.pre_idx[k];
+        int e = jb.dp_end[p] < qlen ? jb.dp_end[p] : qlen;
+        const int pbeg = jb.dp_beg[p];
+        const int64_t pbw = jb.dp_end[p] - pbeg + 1;
+        const S *pH = arena + jb.row_off[p] * 5;
+        int32_t sc = (e >= pbeg) ? (int32_t)pH[e - pbeg] : jb.inf_min;
+        (void)pbw;
+        if (sc > best_score) { best_score = sc; best_i = p; best_j = e; }
+    }
+    res->best_score = best_score;
+    res->best_i = best_i; res->best_j = best_j;
+    if (!jb.ret_cigar) return;
+
+    /* backtrack: transcription of simd_abpoa_cg_backtrack (:309-458) */
+    {
+        int bi = best_i, bj = best_j, start_i = best_i, start_j = best_j;
+        int cur_op = 0x1f, n_c = 0, status = ABAMD_JOB_OK;
+        int look_end = jb.put_gap_at_end, put_right = jb.put_gap_on_right;
+        int n_aln = 0, n_matched = 0;
+        uint64_t *cig = jb.cigar;
+        int id = jb.row_node_id[bi];
+        if (best_j < qlen) dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, qlen - best_j, -1, qlen - 1, &status);
+        while (bi > 0 && bj > 0 && status == ABAMD_JOB_OK) {
+            const int rb = jb.dp_beg[bi], re = jb.dp_end[bi];
+            const int64_t bw = re - rb + 1;
+            const S *H = arena + jb.row_off[bi] * 5;
+            const S *E1r = H + bw, *E2r = E1r + bw, *F1r = E2r + bw, *F2r = F1r + bw;
+            const S Hj = (bj >= rb && bj <= re) ? H[bj - rb] : inf_min;
+            const S Hjm1 = (bj - 1 >= rb && bj - 1 <= re) ? H[bj - 1 - rb] : inf_min;
+            const S E1j = (bj >= rb && bj <= re) ? E1r[bj - rb] : inf_min;
+            const S E2j = (bj >= rb && bj <= re) ? E2r[bj - rb] : inf_min;
+            const S F1j = (bj >= rb && bj <= re) ? F1r[bj - rb] : inf_min;
+            const S F2j = (bj >= rb && bj <= re) ? F2r[bj - rb] : inf_min;
+            const S F1jm1 = (bj - 1 >= rb && bj - 1 <= re) ? F1r[bj - 1 - rb] : inf_min;
+            const S F2jm1 = (bj - 1 >= rb && bj - 1 <= re) ? F2r[bj - 1 - rb] : inf_min;
+            start_i = bi; start_j = bj;
+            const int pq0 = jb.pre_off[bi], pq1 = jb.pre_off[bi + 1];
+            const S s = (S)mat_lds[m * jb.row_base[bi] + query[bj - 1]];
+            const int is_match = jb.row_base[bi] == query[bj - 1];
+            int hit = 0;
+            if (put_right == 0 && look_end == 0 && (cur_op & 0x1)) {
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    if (bj - 1 < jb.dp_beg[p] || bj - 1 > jb.dp_end[p]) continue;
+                    const S *pH = arena + jb.row_off[p] * 5;
+                    if ((S)(pH[bj - 1 - jb.dp_beg[p]] + s + ps) == Hj) {
+                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
+                        bi = p; --bj; id = jb.row_node_id[bi]; hit = 1;
+                        cur_op = 0x1f; ++n_aln; n_matched += is_match;
+                        break;
+                    }
+                }
+            }
+            if (!hit && (cur_op & 0x6)) { /* deletion */
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    if (bj < jb.dp_beg[p] || bj > jb.dp_end[p]) continue;
+                    const int poffc = bj - jb.dp_beg[p];
+                    const int64_t pbw = jb.dp_end[p] - jb.dp_beg[p] + 1;
+                    const S *pH = arena + jb.row_off[p] * 5;
+                    const S *pE1 = pH + pbw, *pE2 = pE1 + pbw;
+                    if (cur_op & 0x2) {
+                        if (cur_op & 0x1) {
+                            if (Hj == (S)(pE1[poffc] + ps)) {
+                                cur_op = ((S)(pH[poffc] - oe1) == pE1[poffc]) ? (0x1 | 0x18) : 0x2;
+                                hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                                bi = p; id = jb.row_node_id[bi];
+                                if (look_end) look_end = 0;
+                                break;
+                            }
+                        } else {
+                            if (E1j == (S)(pE1[poffc] - e1 + ps)) {
+                                cur_op = ((S)(pH[poffc] - oe1) == pE1[poffc]) ? (0x1 | 0x18) : 0x2;
+                                hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                                bi = p; id = jb.row_node_id[bi];
+                                if (look_end) look_end = 0;
+                                break;
+                            }
+                        }
+                    }
+                    if (cur_op & 0x4) {
+                        if (cur_op & 0x1) {
+                            if (Hj == (S)(pE2[poffc] + ps)) {
+                                cur_op = ((S)(pH[poffc] - oe2) == pE2[poffc]) ? (0x1 | 0x18) : 0x4;
+                                hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                                bi = p; id = jb.row_node_id[bi];
+                                if (look_end) look_end = 0;
+                                break;
+                            }
+                        } else {
+                            if (E2j == (S)(pE2[poffc] - e2 + ps)) {
+                                cur_op = ((S)(pH[poffc] - oe2) == pE2[poffc]) ? (0x1 | 0x18) : 0x4;
+                                hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                                bi = p; id = jb.row_node_id[bi];
+                                if (look_end) look_end = 0;
+                                break;
+                            }
+                        }
+                    }
+                }
+            }
+            if (!hit && (cur_op & 0x18)) { /* insertion */
+                if (cur_op & 0x8) {
+                    if (cur_op & 0x1) {
+                        if (Hj == F1j) {
+                            if ((S)(Hjm1 - oe1) == F1j) { cur_op = 0x1 | 0x6; hit = 1; }
+                            else if ((S)(F1jm1 - e1) == F1j) { cur_op = 0x8; hit = 1; }
+                        }
+                    } else {
+                        if ((S)(Hjm1 - oe1) == F1j) { cur_op = 0x1 | 0x6; hit = 1; }
+                        else if ((S)(F1jm1 - e1) == F1j) { cur_op = 0x8; hit = 1; }
+                    }
+                }
+                if (!hit && (cur_op & 0x10)) {
+                    if (cur_op & 0x1) {
+                        if (Hj == F2j) {
+                            if ((S)(Hjm1 - oe2) == F2j) { cur_op = 0x1 | 0x6; hit = 1; }
+                            else if ((S)(F2jm1 - e2) == F2j) { cur_op = 0x10; hit = 1; }
+                        }
+                    } else {
+                        if ((S)(Hjm1 - oe2) == F2j) { cur_op = 0x1 | 0x6; hit = 1; }
+                        else if ((S)(F2jm1 - e2) == F2j) { cur_op = 0x10; hit = 1; }
+                    }
+                }
+                if (hit) {
+                    dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, 1, id, bj - 1, &status);
+                    --bj;
+                    if (look_end) look_end = 0;
+                    ++n_aln;
+                }
+            }
+            if (!hit && (cur_op & 0x1)) {
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    if (bj - 1 < jb.dp_beg[p] || bj - 1 > jb.dp_end[p]) continue;
+                    const S *pH = arena + jb.row_off[p] * 5;
+                    if ((S)(pH[bj - 1 - jb.dp_beg[p]] + s + ps) == Hj) {
+                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
+                        bi = p; --bj; id = jb.row_node_id[bi]; hit = 1;
+                        cur_op = 0x1f; ++n_aln; n_matched += is_match;
+                        look_end = 0;
+                        break;
+                    }
+                }
+            }
+            if (!hit) { status = ABAMD_JOB_BT_DEAD_END; break; }
+        }
+        if (status == ABAMD_JOB_OK && bj > 0)
+            dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, bj, -1, bj - 1, &status);
+        res->status = status;
+        res->n_cigar = n_c;
+        res->n_aln_bases = n_aln;
+        res->n_matched_bases = n_matched;
+        res->node_e = jb.row_node_id[best_i]; res->query_e = best_j - 1;
+        res->node_s = jb.row_node_id[start_i]; res->query_s = start_j - 1;
+    }
+}
+
+extern "C" void abamd_launch_cg_i16(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
+                                    int n_jobs, void *stream) {
+    hipLaunchKernelGGL((cg_global_kernel<int16_t>), dim3(n_jobs), dim3(WAVE), 0,
+                       (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
+}
+extern "C" void abamd_launch_cg_i32(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
+                                    int n_jobs, void *stream) {
+    hipLaunchKernelGGL((cg_global_kernel<int32_t>), dim3(n_jobs), dim3(WAVE), 0,
+                       (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
+}

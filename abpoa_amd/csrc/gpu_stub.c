@@ -18,3 +18,19 @@ void abpoa_amd_get_stats(uint64_t *dp_cells, uint64_t *kernel_ns, uint64_t *n_la
     if (n_launches) *n_launches = 0;
 }
 void abpoa_amd_reset_stats(void) {}
+
+typedef struct {
+    abpoa_t *ab; abpoa_para_t *abpt;
+    int beg_node_id, end_node_id;
+    uint8_t *query; int qlen; abpoa_res_t *res;
+} abamd_batch_job_t;
+
+int abamd_gpu_align_batch(abamd_batch_job_t *batch, int n_jobs) {
+    /* CPU test build: route each job through the dispatching seam so an
+     * injected test oracle serves batched runs too. */
+    int i;
+    for (i = 0; i < n_jobs; ++i)
+        simd_abpoa_align_sequence_to_subgraph(batch[i].ab, batch[i].abpt,
+            batch[i].beg_node_id, batch[i].end_node_id, batch[i].query, batch[i].qlen, batch[i].res);
+    return 0;
+}

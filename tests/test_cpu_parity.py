@@ -1,0 +1,56 @@
+"""CPU-side parity: host pipeline + scalar oracle DP vs the reference's
+committed golden outputs, and (when the reference tree is present) vs the
+live reference binary on fresh synthetic inputs.
+
+These tests pin the oracle per SURVEY.md §8(c): every golden fixture the
+reference's own regression suite checks, plus synthetic read sets at several
+shapes. The oracle is injected explicitly (ABPOA_AMD_TEST_ALIGNER_SO); the
+product GPU path is covered by tests/test_gpu_parity.py.
+"""
+import os
+import subprocess
+import pytest
+
+from conftest import GOLDEN, ROOT, run_stdout
+
+CASES = [
+    ("seq.fa", [], "expected_seq_cons.txt"),
+    ("seq.fa", ["-a1"], "expected_seq_msa.txt"),
+    ("seq.fa", ["-r1"], "expected_seq_r1.txt"),
+    ("seq.fa", ["-r2"], "expected_seq_r2.txt"),
+    ("seq.fa", ["-m1"], "expected_seq_local.txt"),
+    ("seq.fa", ["-m2"], "expected_seq_extend.txt"),
+    ("seq.fa", ["-r3"], "expected_seq_gfa.txt"),
+    ("seq.fa", ["-r5"], "expected_seq_fq.txt"),
+    ("test.fa", [], "expected_test_cons.txt"),
+    ("heter.fa", [], "expected_heter_cons.txt"),
+]
+
+
+@pytest.mark.parametrize("fa,opts,expected", CASES,
+                         ids=["%s%s" % (c[0], "".join(c[1])) for c in CASES])
+def test_golden(cputest_bin, oracle_env, fa, opts, expected):
+    out = run_stdout([cputest_bin, os.path.join(GOLDEN, fa)] + opts, env=oracle_env)
+    want = open(os.path.join(GOLDEN, expected), "rb").read()
+    assert out == want
+
+
+@pytest.mark.parametrize("seed,length,depth", [(1, 500, 20), (2, 1000, 30), (3, 2000, 10)])
+def test_synthetic_vs_reference(cputest_bin, oracle_env, ref_bin, tmp_path, seed, length, depth):
+    fa = tmp_path / "s.fa"
+    subprocess.run(["python3", os.path.join(ROOT, "tests", "make_synth.py"), str(fa),
+                    "--seed", str(seed), "--len", str(length), "--depth", str(depth)],
+                   check=True, stderr=subprocess.DEVNULL)
+    for opts in ([], ["-r1"]):
+        ref = run_stdout([ref_bin, str(fa)] + opts)
+        got = run_stdout([cputest_bin, str(fa)] + opts, env=oracle_env)
+        assert got == ref, "divergence at seed=%d len=%d opts=%r" % (seed, length, opts)
+
+
+def test_product_fails_loudly_without_aligner(cputest_bin):
+    """The CPU test binary must refuse to align without an injected oracle
+    (mirrors the product's no-silent-fallback guarantee)."""
+    p = subprocess.run([cputest_bin, os.path.join(GOLDEN, "seq.fa")],
+                       stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+    assert p.returncode != 0
+    assert b"no GPU aligner" in p.stderr or b"CPU-only TEST build" in p.stderr

@@ -1,0 +1,69 @@
+"""GPU parity: the HIP/CDNA4 hot path must reproduce the reference outputs
+byte-exactly. Runs only on a machine with an AMD GPU (-m gpu).
+
+The comparison anchors are committed golden fixtures (generated from the
+unmodified reference binary in the dev container) and the CPU oracle run live
+next to the GPU on identical synthetic inputs — /root/reference itself is not
+needed at run time."""
+import os
+import subprocess
+import pytest
+
+from conftest import GOLDEN, ROOT, GPU_BIN, run_stdout
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def gpu_bin():
+    assert os.path.exists(GPU_BIN), "abpoa_amd not built (run __graft_entry__.build())"
+    return GPU_BIN
+
+
+def test_gpu_consensus_golden(gpu_bin):
+    out = run_stdout([gpu_bin, os.path.join(GOLDEN, "seq.fa")])
+    want = open(os.path.join(GOLDEN, "expected_seq_cons.txt"), "rb").read()
+    assert out == want
+
+
+def test_gpu_msa_golden(gpu_bin):
+    out = run_stdout([gpu_bin, os.path.join(GOLDEN, "seq.fa"), "-r1"])
+    want = open(os.path.join(GOLDEN, "expected_seq_r1.txt"), "rb").read()
+    assert out == want
+
+
+def test_gpu_mf_consensus_golden(gpu_bin):
+    out = run_stdout([gpu_bin, os.path.join(GOLDEN, "seq.fa"), "-a1"])
+    want = open(os.path.join(GOLDEN, "expected_seq_msa.txt"), "rb").read()
+    assert out == want
+
+
+def test_gpu_test_fa_golden(gpu_bin):
+    out = run_stdout([gpu_bin, os.path.join(GOLDEN, "test.fa")])
+    want = open(os.path.join(GOLDEN, "expected_test_cons.txt"), "rb").read()
+    assert out == want
+
+
+@pytest.mark.parametrize("seed,length,depth", [(11, 500, 20), (12, 1000, 50), (13, 3000, 30)])
+def test_gpu_vs_oracle_synthetic(gpu_bin, cputest_bin, oracle_env, tmp_path, seed, length, depth):
+    """GPU hot path vs CPU oracle on identical fresh inputs; both consensus
+    and RC-MSA bytes (the RC-MSA exercises every read's CIGAR)."""
+    fa = tmp_path / "s.fa"
+    subprocess.run(["python3", os.path.join(ROOT, "tests", "make_synth.py"), str(fa),
+                    "--seed", str(seed), "--len", str(length), "--depth", str(depth)],
+                   check=True, stderr=subprocess.DEVNULL)
+    for opts in ([], ["-r1"]):
+        gpu = run_stdout([gpu_bin, str(fa)] + opts)
+        cpu = run_stdout([cputest_bin, str(fa)] + opts, env=oracle_env)
+        assert gpu == cpu, "GPU/oracle divergence seed=%d len=%d opts=%r" % (seed, length, opts)
+
+
+def test_gpu_10kbp_set(gpu_bin, cputest_bin, oracle_env, tmp_path):
+    """Full north-star shape: one 50x10kbp set, GPU vs oracle consensus."""
+    fa = tmp_path / "s10k.fa"
+    subprocess.run(["python3", os.path.join(ROOT, "tests", "make_synth.py"), str(fa),
+                    "--seed", "42", "--len", "10000", "--depth", "50"],
+                   check=True, stderr=subprocess.DEVNULL)
+    gpu = run_stdout([gpu_bin, str(fa)])
+    cpu = run_stdout([cputest_bin, str(fa)], env=oracle_env)
+    assert gpu == cpu
