@@ -38,3 +38,69 @@ def get_stats():
 
 def reset_stats():
     lib().abpoa_amd_reset_stats()
+
+
+class ConsT(ctypes.Structure):
+    """Mirror of abpoa_cons_t (include/abpoa_amd.h)."""
+    _fields_ = [
+        ("n_cons", ctypes.c_int), ("n_seq", ctypes.c_int), ("msa_len", ctypes.c_int),
+        ("clu_n_seq", ctypes.POINTER(ctypes.c_int)),
+        ("clu_read_ids", ctypes.POINTER(ctypes.POINTER(ctypes.c_int))),
+        ("cons_len", ctypes.POINTER(ctypes.c_int)),
+        ("cons_node_ids", ctypes.POINTER(ctypes.POINTER(ctypes.c_int))),
+        ("cons_base", ctypes.POINTER(ctypes.POINTER(ctypes.c_uint8))),
+        ("msa_base", ctypes.POINTER(ctypes.POINTER(ctypes.c_uint8))),
+        ("cons_cov", ctypes.POINTER(ctypes.POINTER(ctypes.c_int))),
+        ("cons_phred_score", ctypes.POINTER(ctypes.POINTER(ctypes.c_int))),
+    ]
+
+
+CONS_CB = ctypes.CFUNCTYPE(None, ctypes.c_int, ctypes.POINTER(ConsT), ctypes.c_void_p)
+
+
+def msa_batch_consensus(sets, n_threads=4):
+    """Run the batched GPU driver over `sets` (list of list of bytes, codes
+    0..3) and return each set's consensus as an ACGT string."""
+    L = lib()
+    L.abpoa_init_para.restype = ctypes.c_void_p
+    L.abpoa_post_set_para.argtypes = [ctypes.c_void_p]
+    L.abpoa_free_para.argtypes = [ctypes.c_void_p]
+    L.abpoa_amd_msa_batch.argtypes = [
+        ctypes.c_void_p, ctypes.c_int,
+        ctypes.POINTER(ctypes.c_int),
+        ctypes.POINTER(ctypes.POINTER(ctypes.c_int)),
+        ctypes.POINTER(ctypes.POINTER(ctypes.POINTER(ctypes.c_uint8))),
+        CONS_CB, ctypes.c_void_p, ctypes.c_int]
+    para = L.abpoa_init_para()
+    L.abpoa_post_set_para(para)
+
+    n_sets = len(sets)
+    NSeqs = (ctypes.c_int * n_sets)(*[len(s) for s in sets])
+    lens_keep, ptrs_keep, bufs_keep = [], [], []
+    for s in sets:
+        lens = (ctypes.c_int * len(s))(*[len(r) for r in s])
+        lens_keep.append(lens)
+        bufs = [ctypes.create_string_buffer(r, len(r)) for r in s]
+        bufs_keep.append(bufs)
+        ptrs = (ctypes.POINTER(ctypes.c_uint8) * len(s))(
+            *[ctypes.cast(b, ctypes.POINTER(ctypes.c_uint8)) for b in bufs])
+        ptrs_keep.append(ptrs)
+    LensTop = (ctypes.POINTER(ctypes.c_int) * n_sets)(*lens_keep)
+    SeqsTop = (ctypes.POINTER(ctypes.POINTER(ctypes.c_uint8)) * n_sets)(*ptrs_keep)
+
+    out = [None] * n_sets
+    ACGT = "ACGTN-"
+
+    @CONS_CB
+    def cb(idx, cons_p, _user):
+        c = cons_p.contents
+        seqs = []
+        for ci in range(c.n_cons):
+            ln = c.cons_len[ci]
+            seqs.append("".join(ACGT[c.cons_base[ci][j]] for j in range(ln)))
+        out[idx] = seqs[0] if len(seqs) == 1 else seqs
+
+    rc = L.abpoa_amd_msa_batch(para, n_sets, NSeqs, LensTop, SeqsTop, cb, None, n_threads)
+    L.abpoa_free_para(para)
+    assert rc == 0
+    return out

@@ -1,0 +1,242 @@
+#!/usr/bin/env python3
+"""Benchmark: read-sets/sec (and Gcells/sec) for 50x10 kbp global convex-gap
+POA on MI355X — the reference's headline workload (BASELINE.json).
+
+One "step" = one batched pass of the hot path over --sets-per-step synthetic
+read sets (each 50 reads x 10 kbp at 10% ONT-style error, generated on the
+fly, resident in HBM when the timed kernels run). Whole-job value is the
+aggregate over all ranks (one process per GPU; sets shard across ranks with
+no data-path collective: weak scaling).
+
+Emits ONE JSON line from rank 0, including:
+  roofline:     dominant kernel (cg_global_kernel) algorithmic HBM bytes/s
+                measured with HIP events on the library stream (10 B per DP
+                cell: 5 int16 planes written once, re-read once by backtrack)
+  cpu_baseline: the unmodified reference binary (oracle/_ref/abpoa) timed on
+                this box's host cores on a bounded sample.
+"""
+import argparse
+import ctypes
+import json
+import os
+import subprocess
+import sys
+import time
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, ROOT)
+
+WORKLOAD = "1000 synthetic sets x 50 reads x 10 kbp, 10% ONT-style error, global convex gap, 1 MI355X (BASELINE.json configs[1])"
+HBM_PEAK_GBPS = 8000.0  # gfx950 spec peak (MI355X_MICROARCH.md)
+ALG_BYTES_PER_CELL = 10.0  # 5 planes x int16, written once (backtrack re-read counted separately)
+
+
+def gen_sets(rng, n_sets, depth, qlen, p_sub=0.045, p_del=0.03, p_ins=0.025):
+    """Generate encoded (0..3) read sets with numpy; returns list of list of bytes.
+    Error model matches tests/make_synth.py: per base, substitute with p_sub,
+    delete with p_del, keep-and-insert-one with p_ins, else keep."""
+    import numpy as np
+    sets = []
+    for _ in range(n_sets):
+        ref = rng.integers(0, 4, size=qlen, dtype=np.uint8)
+        reads = []
+        for _ in range(depth):
+            r = rng.random(qlen)
+            sub = r < p_sub
+            mask_del = (r >= p_sub) & (r < p_sub + p_del)
+            ins = (r >= p_sub + p_del) & (r < p_sub + p_del + p_ins)
+            out = ref.copy()
+            nsub = int(sub.sum())
+            if nsub:
+                out[sub] = (ref[sub] + rng.integers(1, 4, size=nsub, dtype=np.uint8)) % 4
+            kept = out[~mask_del]
+            ins_pos = np.nonzero(ins[~mask_del])[0]
+            if len(ins_pos):
+                ins_bases = rng.integers(0, 4, size=len(ins_pos), dtype=np.uint8)
+                kept = np.insert(kept, ins_pos + 1, ins_bases)
+            reads.append(np.ascontiguousarray(kept).tobytes())
+        sets.append(reads)
+    return sets
+
+
+def build_batch_args(lib, sets):
+    n_sets = len(sets)
+    NSeqs = (ctypes.c_int * n_sets)(*[len(s) for s in sets])
+    seq_len_arrays = []
+    seq_ptr_arrays = []
+    keepalive = []
+    for s in sets:
+        lens = (ctypes.c_int * len(s))(*[len(r) for r in s])
+        seq_len_arrays.append(lens)
+        bufs = [ctypes.create_string_buffer(r, len(r)) for r in s]
+        keepalive.append(bufs)
+        ptrs = (ctypes.POINTER(ctypes.c_uint8) * len(s))(
+            *[ctypes.cast(b, ctypes.POINTER(ctypes.c_uint8)) for b in bufs])
+        seq_ptr_arrays.append(ptrs)
+    LensTop = (ctypes.POINTER(ctypes.c_int) * n_sets)(*seq_len_arrays)
+    SeqsTop = (ctypes.POINTER(ctypes.POINTER(ctypes.c_uint8)) * n_sets)(*seq_ptr_arrays)
+    return NSeqs, LensTop, SeqsTop, (seq_len_arrays, seq_ptr_arrays, keepalive)
+
+
+class Para(ctypes.Structure):
+    pass  # opaque; created/freed by the library
+
+
+def native_setup():
+    import abpoa_amd
+    lib = abpoa_amd.lib()
+    lib.abpoa_init_para.restype = ctypes.c_void_p
+    lib.abpoa_post_set_para.argtypes = [ctypes.c_void_p]
+    lib.abpoa_free_para.argtypes = [ctypes.c_void_p]
+    lib.abpoa_amd_msa_batch.argtypes = [
+        ctypes.c_void_p, ctypes.c_int,
+        ctypes.POINTER(ctypes.c_int),
+        ctypes.POINTER(ctypes.POINTER(ctypes.c_int)),
+        ctypes.POINTER(ctypes.POINTER(ctypes.POINTER(ctypes.c_uint8))),
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int]
+    lib.abpoa_amd_msa_batch.restype = ctypes.c_int
+    return lib
+
+
+def run_step(lib, para, sets, n_threads):
+    NSeqs, LensTop, SeqsTop, keep = build_batch_args(lib, sets)
+    rc = lib.abpoa_amd_msa_batch(para, len(sets), NSeqs, LensTop, SeqsTop,
+                                 None, None, n_threads)
+    assert rc == 0
+    del keep
+
+
+def cpu_baseline_leg(depth, qlen):
+    """Time the unmodified reference binary on one set (bounded sample)."""
+    ref_bin = os.path.join(ROOT, "oracle", "_ref", "abpoa")
+    if not os.path.exists(ref_bin) or os.environ.get("ABPOA_BENCH_SKIP_CPU"):
+        return None
+    import tempfile
+    with tempfile.TemporaryDirectory() as td:
+        fa = os.path.join(td, "cpu.fa")
+        subprocess.run([sys.executable, os.path.join(ROOT, "tests", "make_synth.py"), fa,
+                        "--seed", "99", "--len", str(qlen), "--depth", str(depth)],
+                       check=True, stderr=subprocess.DEVNULL)
+        t0 = time.monotonic()
+        subprocess.run([ref_bin, fa], check=True, stdout=subprocess.DEVNULL,
+                       stderr=subprocess.DEVNULL)
+        dt = time.monotonic() - t0
+    return {"value": 1.0 / dt, "unit": "sets/s", "cores": 1, "kind": "reference",
+            "sample": "1 set (%d reads x %d bp), reference binary, cold arena" % (depth, qlen)}
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=2)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--sets-per-step", type=int, default=48)
+    ap.add_argument("--depth", type=int, default=50)
+    ap.add_argument("--qlen", type=int, default=10000)
+    ap.add_argument("--threads", type=int, default=max(2, (os.cpu_count() or 8) - 2))
+    ap.add_argument("--dry-run", action="store_true",
+                    help="exercise the distributed harness without a GPU (CPU tests)")
+    args = ap.parse_args()
+
+    import numpy as np
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    dist = None
+    if world > 1 or os.environ.get("MASTER_ADDR"):
+        import torch.distributed as tdist
+        import torch
+        backend = "gloo" if (args.dry_run or not torch.cuda.is_available()) else "nccl"
+        tdist.init_process_group(backend=backend)
+        dist = tdist
+        rank = tdist.get_rank()
+        world = tdist.get_world_size()
+        if backend == "nccl":
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+
+    if args.dry_run:
+        lib = para = None
+    else:
+        lib = native_setup()
+        para = lib.abpoa_init_para()
+        lib.abpoa_post_set_para(para)
+        import abpoa_amd
+
+    rng = np.random.default_rng(4242 + 1000 * rank)
+
+    def one_step(step_idx):
+        if args.dry_run:
+            time.sleep(0.01)
+            return
+        sets = gen_sets(rng, args.sets_per_step, args.depth, args.qlen)
+        run_step(lib, para, sets, args.threads)
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+        if not args.dry_run:
+            try:
+                import torch
+                if torch.cuda.is_available():
+                    torch.cuda.synchronize()
+            except Exception:
+                pass
+
+    for wstep in range(args.warmup):
+        one_step(-1 - wstep)
+
+    if not args.dry_run:
+        import abpoa_amd
+        abpoa_amd.reset_stats()
+    barrier_sync()
+    t0 = time.monotonic()
+    for k in range(args.steps):
+        one_step(k)
+    barrier_sync()
+    elapsed = time.monotonic() - t0
+
+    # max over ranks
+    if dist is not None:
+        import torch
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        total_sets = args.steps * args.sets_per_step * world
+        value = total_sets / elapsed
+        if args.dry_run:
+            line = {"metric": "read-sets/sec, 50x10 kbp global convex-gap POA (DRY RUN)",
+                    "value": value, "unit": "sets/s", "n_gpus": world,
+                    "steps": args.steps, "warmup": args.warmup,
+                    "ms_per_step": elapsed / args.steps * 1e3,
+                    "higher_is_better": True, "scaling": "weak", "dry_run": True}
+            print(json.dumps(line), flush=True)
+            return
+        import abpoa_amd
+        cells, kns, launches = abpoa_amd.get_stats()
+        gcells_s = cells / elapsed / 1e9
+        ach = (cells * ALG_BYTES_PER_CELL) / max(kns, 1)  # bytes/ns == GB/s
+        roofline = {"bound": "hbm", "achieved": round(ach, 1), "peak": HBM_PEAK_GBPS,
+                    "unit": "GB/s", "frac": round(ach / HBM_PEAK_GBPS, 4),
+                    "traffic": None}
+        cpu = cpu_baseline_leg(args.depth, args.qlen) if world == 1 else None
+        line = {
+            "metric": "read-sets/sec, 50x10 kbp global convex-gap POA",
+            "value": round(value, 4), "unit": "sets/s",
+            "n_gpus": world, "steps": args.steps, "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 1),
+            "higher_is_better": True, "scaling": "weak", "vs_baseline": None,
+            "dtype": "int16", "data": "synthetic",
+            "config": {"workload": WORKLOAD, "sets_per_step": args.sets_per_step,
+                       "depth": args.depth, "qlen": args.qlen,
+                       "gcells_per_s": round(gcells_s, 3),
+                       "kernel_s": round(kns / 1e9, 3), "launches": launches,
+                       "parallelism": "dp%d independent read-set shards (no data-path collective)" % world},
+            "roofline": roofline,
+            "cpu_baseline": cpu,
+        }
+        print(json.dumps(line), flush=True)
+
+
+if __name__ == "__main__":
+    main()
