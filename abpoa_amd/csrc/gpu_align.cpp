@@ -21,6 +21,7 @@
 #include <cstring>
 #include <vector>
 #include <cmath>
+#include <thread>
 #include "abpoa_amd.h"
 #include "gpu_core.h"
 
@@ -68,6 +69,19 @@ struct HostBuf {
     uint8_t *at(size_t off) { return v.data() + off; }
 };
 
+struct PinnedBuf {
+    uint8_t *p = nullptr;
+    size_t cap = 0;
+    void ensure(size_t n) {
+        if (n <= cap) return;
+        size_t want = cap ? cap : 1 << 20;
+        while (want < n) want <<= 1;
+        if (p) HIP_CHECK(hipHostFree(p));
+        HIP_CHECK(hipHostMalloc((void**)&p, want));
+        cap = want;
+    }
+};
+
 struct GpuCtx {
     bool init = false;
     hipStream_t stream;
@@ -75,6 +89,8 @@ struct GpuCtx {
     DevBuf slab;        /* all per-job inputs + metadata */
     DevBuf arena;       /* DP plane arena */
     DevBuf jobs, results, cigars;
+    PinnedBuf stage;    /* pinned H2D staging for the whole batch */
+    std::vector<HostBuf> jb_bufs;   /* per-job pack buffers (parallel pack) */
     HostBuf hb;
     std::vector<abamd_gpu_job_t> hjobs;
     std::vector<abamd_gpu_res_t> hres;
@@ -104,13 +120,15 @@ struct JobPack {
     size_t o_query, o_base, o_nodeid, o_preoff, o_preidx, o_preps, o_outoff, o_outidx, o_remain;
     size_t o_ml, o_mr, o_beg, o_end, o_roff;
     int64_t arena_off;      /* cells */
+    size_t slab_base;       /* this job's base offset in the device slab */
     int cigar_off;          /* entries */
     int n_rows, qlen;
     int n_pre, n_out;
 };
 
-/* Build the compact-row job into ctx.hb; returns per-job arena demand (cells). */
-static int64_t pack_job(GpuCtx &C, JobPack &P, abpoa_t *ab, abpoa_para_t *abpt,
+/* Build the compact-row job into its own host buffer; returns per-job arena
+ * demand (cells). Thread-safe: touches only hb and P. */
+static int64_t pack_job(HostBuf &hb, JobPack &P, abpoa_t *ab, abpoa_para_t *abpt,
                         int beg_node_id, int end_node_id, uint8_t *query, int qlen) {
     abpoa_graph_t *g = ab->abg;
     int beg_index = g->node_id_to_index[beg_node_id];
@@ -134,18 +152,18 @@ static int64_t pack_job(GpuCtx &C, JobPack &P, abpoa_t *ab, abpoa_para_t *abpt,
 
     P.n_rows = n_rows; P.qlen = qlen;
 
-    size_t o_query = C.hb.alloc(qlen);
-    size_t o_base = C.hb.alloc(n_rows);
-    size_t o_nodeid = C.hb.alloc((size_t)n_rows * 4);
-    size_t o_preoff = C.hb.alloc((size_t)(n_rows + 1) * 4);
-    size_t o_outoff = C.hb.alloc((size_t)(n_rows + 1) * 4);
-    size_t o_remain = C.hb.alloc((size_t)n_rows * 4);
-    memcpy(C.hb.at(o_query), query, qlen);
-    uint8_t *base = C.hb.at(o_base);
-    int *nodeid = (int*)C.hb.at(o_nodeid);
-    int *preoff = (int*)C.hb.at(o_preoff);
-    int *outoff = (int*)C.hb.at(o_outoff);
-    int *remain = (int*)C.hb.at(o_remain);
+    size_t o_query = hb.alloc(qlen);
+    size_t o_base = hb.alloc(n_rows);
+    size_t o_nodeid = hb.alloc((size_t)n_rows * 4);
+    size_t o_preoff = hb.alloc((size_t)(n_rows + 1) * 4);
+    size_t o_outoff = hb.alloc((size_t)(n_rows + 1) * 4);
+    size_t o_remain = hb.alloc((size_t)n_rows * 4);
+    memcpy(hb.at(o_query), query, qlen);
+    uint8_t *base = hb.at(o_base);
+    int *nodeid = (int*)hb.at(o_nodeid);
+    int *preoff = (int*)hb.at(o_preoff);
+    int *outoff = (int*)hb.at(o_outoff);
+    int *remain = (int*)hb.at(o_remain);
 
     int n_pre = 0, n_out = 0;
     for (int i = 0, r = 0; i < span; ++i) {
@@ -171,16 +189,16 @@ static int64_t pack_job(GpuCtx &C, JobPack &P, abpoa_t *ab, abpoa_para_t *abpt,
     }
     preoff[n_rows] = n_pre; outoff[n_rows] = n_out;
 
-    size_t o_preidx = C.hb.alloc((size_t)(n_pre > 0 ? n_pre : 1) * 4);
-    size_t o_preps = C.hb.alloc((size_t)(n_pre > 0 ? n_pre : 1) * 4);
-    size_t o_outidx = C.hb.alloc((size_t)(n_out > 0 ? n_out : 1) * 4);
-    int *preidx = (int*)C.hb.at(o_preidx);
-    int *preps = (int*)C.hb.at(o_preps);
-    int *outidx = (int*)C.hb.at(o_outidx);
+    size_t o_preidx = hb.alloc((size_t)(n_pre > 0 ? n_pre : 1) * 4);
+    size_t o_preps = hb.alloc((size_t)(n_pre > 0 ? n_pre : 1) * 4);
+    size_t o_outidx = hb.alloc((size_t)(n_out > 0 ? n_out : 1) * 4);
+    int *preidx = (int*)hb.at(o_preidx);
+    int *preps = (int*)hb.at(o_preps);
+    int *outidx = (int*)hb.at(o_outidx);
     /* re-read offsets (hb.alloc may have resized the vector) */
-    base = C.hb.at(o_base); nodeid = (int*)C.hb.at(o_nodeid);
-    preoff = (int*)C.hb.at(o_preoff); outoff = (int*)C.hb.at(o_outoff);
-    remain = (int*)C.hb.at(o_remain);
+    base = hb.at(o_base); nodeid = (int*)hb.at(o_nodeid);
+    preoff = (int*)hb.at(o_preoff); outoff = (int*)hb.at(o_outoff);
+    remain = (int*)hb.at(o_remain);
 
     n_pre = 0; n_out = 0;
     for (int i = 0, r = 0; i < span; ++i) {
@@ -218,11 +236,11 @@ static int64_t pack_job(GpuCtx &C, JobPack &P, abpoa_t *ab, abpoa_para_t *abpt,
     }
 
     /* working arrays (device-initialized) */
-    size_t o_ml = C.hb.alloc((size_t)n_rows * 4);
-    size_t o_mr = C.hb.alloc((size_t)n_rows * 4);
-    size_t o_beg = C.hb.alloc((size_t)n_rows * 4);
-    size_t o_end = C.hb.alloc((size_t)n_rows * 4);
-    size_t o_roff = C.hb.alloc((size_t)n_rows * 8);
+    size_t o_ml = hb.alloc((size_t)n_rows * 4);
+    size_t o_mr = hb.alloc((size_t)n_rows * 4);
+    size_t o_beg = hb.alloc((size_t)n_rows * 4);
+    size_t o_end = hb.alloc((size_t)n_rows * 4);
+    size_t o_roff = hb.alloc((size_t)n_rows * 8);
 
     P.o_query = o_query; P.o_base = o_base; P.o_nodeid = o_nodeid;
     P.o_preoff = o_preoff; P.o_preidx = o_preidx; P.o_preps = o_preps;
@@ -301,26 +319,69 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
         exit(EXIT_FAILURE);
     }
 
-    C.hb.reset();
     C.hjobs.resize(n_jobs);
     C.hres.resize(n_jobs);
+    if ((int)C.jb_bufs.size() < n_jobs) C.jb_bufs.resize(n_jobs);
     std::vector<JobPack> packs(n_jobs);
     std::vector<int64_t> arena_est(n_jobs);
+    std::vector<int> bits_v(n_jobs, 16);
 
-    /* pack matrix once (shared paras) */
-    size_t o_mat = C.hb.alloc((size_t)abpt->m * abpt->m * 4);
-    memcpy(C.hb.at(o_mat), abpt->mat, (size_t)abpt->m * abpt->m * 4);
-
+    /* parallel pack: each job into its own (reused) host buffer */
+    {
+        int nthr = (int)std::thread::hardware_concurrency();
+        if (nthr < 1) nthr = 1;
+        if (nthr > 16) nthr = 16;
+        if (nthr > n_jobs) nthr = n_jobs;
+        std::atomic<int> next{0};
+        auto worker = [&]() {
+            for (;;) {
+                int i = next.fetch_add(1);
+                if (i >= n_jobs) break;
+                BatchJob &B = batch[i];
+                C.jb_bufs[i].reset();
+                arena_est[i] = pack_job(C.jb_bufs[i], packs[i], B.ab, B.abpt,
+                                        B.beg_node_id, B.end_node_id, B.query, B.qlen);
+                int inf_min;
+                int span = B.ab->abg->node_id_to_index[B.end_node_id] - B.ab->abg->node_id_to_index[B.beg_node_id] + 1;
+                pick_width(B.abpt, B.qlen, span, &bits_v[i], &inf_min);
+                packs[i].jb.inf_min = inf_min;
+                packs[i].cigar_off = 0;
+            }
+        };
+        std::vector<std::thread> ts;
+        for (int t = 1; t < nthr; ++t) ts.emplace_back(worker);
+        worker();
+        for (auto &t : ts) t.join();
+    }
     int bits_max = 16;
+    for (int i = 0; i < n_jobs; ++i) if (bits_v[i] > bits_max) bits_max = bits_v[i];
+
+    /* assemble the pinned staging slab: [mat][job 0][job 1]... */
+    const size_t mat_bytes = (size_t)abpt->m * abpt->m * 4;
+    size_t o_mat = 0;
+    size_t total = (mat_bytes + 255) & ~(size_t)255;
     for (int i = 0; i < n_jobs; ++i) {
-        BatchJob &B = batch[i];
-        arena_est[i] = pack_job(C, packs[i], B.ab, B.abpt, B.beg_node_id, B.end_node_id, B.query, B.qlen);
-        int bits, inf_min;
-        int span = B.ab->abg->node_id_to_index[B.end_node_id] - B.ab->abg->node_id_to_index[B.beg_node_id] + 1;
-        pick_width(B.abpt, B.qlen, span, &bits, &inf_min);
-        packs[i].jb.inf_min = inf_min;
-        if (bits > bits_max) bits_max = bits;
-        packs[i].cigar_off = 0; /* filled below */
+        packs[i].slab_base = total;
+        total += (C.jb_bufs[i].used + 255) & ~(size_t)255;
+    }
+    C.stage.ensure(total);
+    memcpy(C.stage.p + o_mat, abpt->mat, mat_bytes);
+    {
+        int nthr = (int)std::thread::hardware_concurrency();
+        if (nthr < 1) nthr = 1;
+        if (nthr > 16) nthr = 16;
+        std::atomic<int> next{0};
+        auto worker = [&]() {
+            for (;;) {
+                int i = next.fetch_add(1);
+                if (i >= n_jobs) break;
+                memcpy(C.stage.p + packs[i].slab_base, C.jb_bufs[i].v.data(), C.jb_bufs[i].used);
+            }
+        };
+        std::vector<std::thread> ts;
+        for (int t = 1; t < nthr; ++t) ts.emplace_back(worker);
+        worker();
+        for (auto &t : ts) t.join();
     }
     /* a mixed batch runs at the widest type; widths are identical across jobs
      * of one workload in practice (same scoring paras, similar qlen) */
@@ -361,12 +422,13 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
         C.cigars.ensure((size_t)cig_total * 8);
 
         /* upload slab + jobs */
-        C.slab.ensure(C.hb.used);
-        HIP_CHECK(hipMemcpyAsync(C.slab.p, C.hb.v.data(), C.hb.used, hipMemcpyHostToDevice, C.stream));
-        uint8_t *S = (uint8_t*)C.slab.p;
+        C.slab.ensure(total);
+        HIP_CHECK(hipMemcpyAsync(C.slab.p, C.stage.p, total, hipMemcpyHostToDevice, C.stream));
+        uint8_t *S0 = (uint8_t*)C.slab.p;
         for (int i = 0; i < n_jobs; ++i) {
             JobPack &P = packs[i];
             abamd_gpu_job_t &jb = P.jb;
+            uint8_t *S = S0 + P.slab_base;
             jb.query = S + P.o_query;
             jb.row_base = S + P.o_base;
             jb.row_node_id = (int*)(S + P.o_nodeid);
@@ -381,7 +443,7 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
             jb.dp_beg = (int*)(S + P.o_beg);
             jb.dp_end = (int*)(S + P.o_end);
             jb.row_off = (int64_t*)(S + P.o_roff);
-            jb.mat = (int*)(S + o_mat);
+            jb.mat = (int*)(S0 + o_mat);
             jb.arena = (uint8_t*)C.arena.p + (size_t)P.arena_off * 5 * ssz;
             jb.arena_cap = arena_est[i];
             jb.cigar = (uint64_t*)C.cigars.p + P.cigar_off;
