@@ -27,6 +27,8 @@
 #include "gpu_core.h"
 
 #define WAVE 64
+#define JOBS_PER_BLOCK 4   /* 256-thread workgroups: 4 waves = 4 independent jobs,
+                              doubling resident waves/CU vs 64-thread blocks */
 
 template <typename S> struct ScoreLim;
 template <> struct ScoreLim<int16_t> { static constexpr int maxv = 32767; };
@@ -53,22 +55,27 @@ __device__ static int dev_push_cigar(uint64_t *cig, int *n_c, int cap, int op, i
 /* The DP + backtrack kernel (global alignment, convex gaps).          */
 /* ------------------------------------------------------------------ */
 template <typename S>
-__global__ __launch_bounds__(WAVE)
+__global__ __launch_bounds__(WAVE * JOBS_PER_BLOCK, 7)
 void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                       abamd_gpu_res_t *__restrict__ results, int n_jobs) {
-    const int jid = blockIdx.x;
-    if (jid >= n_jobs) return;
-    const abamd_gpu_job_t jb = jobs[jid];
-    abamd_gpu_res_t *res = &results[jid];
-    const int lane = threadIdx.x;
+    const int wid = threadIdx.x / WAVE;
+    const int jid = blockIdx.x * JOBS_PER_BLOCK + wid;
+    const int lane = threadIdx.x % WAVE;
 
     __shared__ int mat_lds[27 * 27];
-    const int m = jb.m;
-    for (int i = lane; i < m * m; i += WAVE) mat_lds[i] = jb.mat[i];
-    /* single wave: no barrier needed; LDS writes are visible after lgkmcnt */
-    __builtin_amdgcn_s_waitcnt(0); /* drain lds+vm counters conservatively */
+    /* every job of a batch shares one scoring matrix (gpu_align.cpp) */
+    {
+        const int m0 = jobs[0].m;
+        const int *mat0 = jobs[0].mat;
+        for (int i = threadIdx.x; i < m0 * m0; i += WAVE * JOBS_PER_BLOCK)
+            mat_lds[i] = mat0[i];
+    }
+    __syncthreads();
+    if (jid >= n_jobs) return;
+    const abamd_gpu_job_t &jb = jobs[jid];
+    abamd_gpu_res_t *res = &results[jid];
 
-    const int qlen = jb.qlen, n_rows = jb.n_rows, w = jb.w;
+    const int qlen = jb.qlen, n_rows = jb.n_rows, w = jb.w, m = jb.m;
     const S inf_min = (S)jb.inf_min;
     const S o1 = (S)jb.o1, o2 = (S)jb.o2, e1 = (S)jb.e1, e2 = (S)jb.e2;
     const S oe1 = (S)jb.oe1, oe2 = (S)jb.oe2;
@@ -413,11 +420,13 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
 
 extern "C" void abamd_launch_cg_i16(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
                                     int n_jobs, void *stream) {
-    hipLaunchKernelGGL((cg_global_kernel<int16_t>), dim3(n_jobs), dim3(WAVE), 0,
+    int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
+    hipLaunchKernelGGL((cg_global_kernel<int16_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
                        (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
 }
 extern "C" void abamd_launch_cg_i32(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
                                     int n_jobs, void *stream) {
-    hipLaunchKernelGGL((cg_global_kernel<int32_t>), dim3(n_jobs), dim3(WAVE), 0,
+    int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
+    hipLaunchKernelGGL((cg_global_kernel<int32_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
                        (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
 }

@@ -141,17 +141,20 @@ def main():
     import numpy as np
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
+    # One process per GPU: pin the device BEFORE any HIP init. torch's bundled
+    # ROCm runtime must never initialize in this process (it conflicts with
+    # the system ROCm the native library links), so rank coordination uses
+    # gloo and device work is synchronized inside the library
+    # (hipStreamSynchronize at every batch end).
+    if "HIP_VISIBLE_DEVICES" not in os.environ and "LOCAL_RANK" in os.environ:
+        os.environ["HIP_VISIBLE_DEVICES"] = os.environ["LOCAL_RANK"]
     dist = None
     if world > 1 or os.environ.get("MASTER_ADDR"):
         import torch.distributed as tdist
-        import torch
-        backend = "gloo" if (args.dry_run or not torch.cuda.is_available()) else "nccl"
-        tdist.init_process_group(backend=backend)
+        tdist.init_process_group(backend="gloo")
         dist = tdist
         rank = tdist.get_rank()
         world = tdist.get_world_size()
-        if backend == "nccl":
-            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
 
     if args.dry_run:
         lib = para = None
@@ -171,15 +174,10 @@ def main():
         run_step(lib, para, sets, args.threads)
 
     def barrier_sync():
+        # device work is already drained: the native driver ends every batch
+        # with hipStreamSynchronize on its own stream
         if dist is not None:
             dist.barrier()
-        if not args.dry_run:
-            try:
-                import torch
-                if torch.cuda.is_available():
-                    torch.cuda.synchronize()
-            except Exception:
-                pass
 
     for wstep in range(args.warmup):
         one_step(-1 - wstep)
