@@ -54,3 +54,51 @@ def test_product_fails_loudly_without_aligner(cputest_bin):
                        stdout=subprocess.PIPE, stderr=subprocess.PIPE)
     assert p.returncode != 0
     assert b"no GPU aligner" in p.stderr or b"CPU-only TEST build" in p.stderr
+
+
+MODE_CASES = [
+    ["-O", "4", "-E", "2"],          # affine
+    ["-O", "0", "-E", "2"],          # linear
+    ["-m1"],                          # local (convex)
+    ["-m2"],                          # extension (convex)
+    ["-O", "4", "-E", "2", "-m1"],   # local affine
+    ["-O", "0", "-E", "2", "-m2"],   # extension linear
+]
+
+
+@pytest.mark.parametrize("opts", MODE_CASES, ids=lambda o: "".join(o))
+def test_modes_vs_reference(cputest_bin, oracle_env, ref_bin, tmp_path, opts):
+    fa = tmp_path / "s.fa"
+    subprocess.run(["python3", os.path.join(ROOT, "tests", "make_synth.py"), str(fa),
+                    "--seed", "7", "--len", "1200", "--depth", "25"],
+                   check=True, stderr=subprocess.DEVNULL)
+    ref = run_stdout([ref_bin, str(fa)] + opts)
+    got = run_stdout([cputest_bin, str(fa)] + opts, env=oracle_env)
+    assert got == ref
+
+
+def test_aa_mode_vs_reference(cputest_bin, oracle_env, ref_bin, tmp_path):
+    """Amino-acid alphabet (m=27) with BLOSUM62, global affine (configs[3])."""
+    import random
+    random.seed(5)
+    aa = "ARNDCQEGHILKMFPSTWYV"
+    refseq = "".join(random.choice(aa) for _ in range(400))
+    reads = []
+    for _ in range(10):
+        s = []
+        for ch in refseq:
+            r = random.random()
+            if r < 0.04: s.append(random.choice(aa))
+            elif r < 0.07: pass
+            elif r < 0.09: s.extend((ch, random.choice(aa)))
+            else: s.append(ch)
+        reads.append("".join(s))
+    fa = tmp_path / "aa.fa"
+    with open(fa, "w") as f:
+        for i, r in enumerate(reads):
+            f.write(">r%d\n%s\n" % (i, r))
+    mtx = os.path.join(GOLDEN, "BLOSUM62.mtx")
+    for opts in (["-c", "-t", mtx, "-O", "4", "-E", "2"], ["-c", "-t", mtx]):
+        ref = run_stdout([ref_bin, str(fa)] + opts)
+        got = run_stdout([cputest_bin, str(fa)] + opts, env=oracle_env)
+        assert got == ref
