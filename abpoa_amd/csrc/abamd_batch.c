@@ -13,6 +13,9 @@
 #include "abpoa_amd.h"
 #include "abamd_util.h"
 
+void abamd_timing_report(const char *tag);
+static double g_fold_s, g_cons_s, g_collect_s;
+
 /* from gpu_align.cpp */
 typedef struct {
     abpoa_t *ab;
@@ -160,18 +163,27 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
             done += take;
         }
         /* fold on host threads */
+        double tf0 = abamd_realtime();
         fold_work_t fw = { sets, n_sets, abpt, r, 0, PTHREAD_MUTEX_INITIALIZER };
         int nt = n_host_threads;
         for (i = 0; i < nt; ++i) pthread_create(&tids[i], NULL, fold_worker, &fw);
         for (i = 0; i < nt; ++i) pthread_join(tids[i], NULL);
+        g_fold_s += abamd_realtime() - tf0;
     }
 
     /* consensus on host threads, then emit callbacks in order */
+    double tc0 = abamd_realtime();
     cons_work_t cw = { sets, n_sets, 0, abpt, cb, user, PTHREAD_MUTEX_INITIALIZER };
     for (i = 0; i < n_host_threads; ++i) pthread_create(&tids[i], NULL, cons_worker, &cw);
     for (i = 0; i < n_host_threads; ++i) pthread_join(tids[i], NULL);
+    g_cons_s += abamd_realtime() - tc0;
     if (cb) for (i = 0; i < n_sets; ++i) cb(i, sets[i].ab->abc, user);
 
+    if (getenv("ABPOA_AMD_TIMING")) {
+        fprintf(stderr, "[abamd timing batch] fold %.2fs cons %.2fs (threads %d)\n",
+                g_fold_s, g_cons_s, n_host_threads);
+        abamd_timing_report("batch");
+    }
     for (i = 0; i < n_sets; ++i) {
         free(sets[i].weight_buf);
         abpoa_free(sets[i].ab);

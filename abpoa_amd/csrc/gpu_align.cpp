@@ -21,6 +21,7 @@
 #include <cstring>
 #include <vector>
 #include <cmath>
+#include <ctime>
 #include <thread>
 #include "abpoa_amd.h"
 #include "gpu_core.h"
@@ -29,14 +30,31 @@
     fprintf(stderr, "[abpoa_amd] HIP error %s at %s:%d: %s\n", hipGetErrorName(_e), __FILE__, __LINE__, hipGetErrorString(_e)); \
     exit(EXIT_FAILURE); } } while (0)
 
-static std::atomic<uint64_t> g_dp_cells{0}, g_kernel_ns{0}, g_launches{0};
+static std::atomic<uint64_t> g_dp_cells{0}, g_kernel_ns{0}, g_launches{0}, g_alg_bytes{0};
+static std::atomic<uint64_t> g_pack_ns{0}, g_stage_ns{0}, g_gpu_ns{0}, g_unpack_ns{0};
+static inline uint64_t now_ns() {
+    struct timespec ts; clock_gettime(CLOCK_MONOTONIC, &ts);
+    return (uint64_t)ts.tv_sec * 1000000000ull + ts.tv_nsec;
+}
+extern "C" void abamd_timing_report(const char *tag) {
+    if (!getenv("ABPOA_AMD_TIMING")) return;
+    fprintf(stderr, "[abamd timing %s] pack %.2fs stage %.2fs gpu(upload->sync) %.2fs unpack %.2fs kernel %.2fs launches %llu\n",
+            tag, g_pack_ns/1e9, g_stage_ns/1e9, g_gpu_ns/1e9, g_unpack_ns/1e9, g_kernel_ns/1e9,
+            (unsigned long long)g_launches.load());
+}
 
 extern "C" void abpoa_amd_get_stats(uint64_t *dp_cells, uint64_t *kernel_ns, uint64_t *n_launches) {
     if (dp_cells) *dp_cells = g_dp_cells.load();
     if (kernel_ns) *kernel_ns = g_kernel_ns.load();
     if (n_launches) *n_launches = g_launches.load();
 }
-extern "C" void abpoa_amd_reset_stats(void) { g_dp_cells = 0; g_kernel_ns = 0; g_launches = 0; }
+extern "C" void abpoa_amd_get_stats2(uint64_t *alg_bytes) {
+    if (alg_bytes) *alg_bytes = g_alg_bytes.load();
+}
+extern "C" void abpoa_amd_reset_stats(void) {
+    g_dp_cells = 0; g_kernel_ns = 0; g_launches = 0; g_alg_bytes = 0;
+    g_pack_ns = 0; g_stage_ns = 0; g_gpu_ns = 0; g_unpack_ns = 0;
+}
 
 /* ---------------- device context (per thread) ---------------- */
 
@@ -118,7 +136,7 @@ thread_local GpuCtx g_ctx;
 struct JobPack {
     abamd_gpu_job_t jb;     /* device pointers filled after upload */
     size_t o_query, o_base, o_nodeid, o_preoff, o_preidx, o_preps, o_outoff, o_outidx, o_remain;
-    size_t o_ml, o_mr, o_beg, o_end, o_roff;
+    size_t o_ml, o_mr, o_meta;
     int64_t arena_off;      /* cells */
     size_t slab_base;       /* this job's base offset in the device slab */
     int cigar_off;          /* entries */
@@ -238,14 +256,12 @@ static int64_t pack_job(HostBuf &hb, JobPack &P, abpoa_t *ab, abpoa_para_t *abpt
     /* working arrays (device-initialized) */
     size_t o_ml = hb.alloc((size_t)n_rows * 4);
     size_t o_mr = hb.alloc((size_t)n_rows * 4);
-    size_t o_beg = hb.alloc((size_t)n_rows * 4);
-    size_t o_end = hb.alloc((size_t)n_rows * 4);
-    size_t o_roff = hb.alloc((size_t)n_rows * 8);
+    size_t o_meta = hb.alloc((size_t)n_rows * sizeof(abamd_row_meta_t));
 
     P.o_query = o_query; P.o_base = o_base; P.o_nodeid = o_nodeid;
     P.o_preoff = o_preoff; P.o_preidx = o_preidx; P.o_preps = o_preps;
     P.o_outoff = o_outoff; P.o_outidx = o_outidx; P.o_remain = o_remain;
-    P.o_ml = o_ml; P.o_mr = o_mr; P.o_beg = o_beg; P.o_end = o_end; P.o_roff = o_roff;
+    P.o_ml = o_ml; P.o_mr = o_mr; P.o_meta = o_meta;
     P.n_pre = n_pre; P.n_out = n_out;
 
     abamd_gpu_job_t &jb = P.jb;
@@ -319,6 +335,7 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
         exit(EXIT_FAILURE);
     }
 
+    uint64_t t_pack0 = now_ns();
     C.hjobs.resize(n_jobs);
     C.hres.resize(n_jobs);
     if ((int)C.jb_bufs.size() < n_jobs) C.jb_bufs.resize(n_jobs);
@@ -355,6 +372,8 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
     }
     int bits_max = 16;
     for (int i = 0; i < n_jobs; ++i) if (bits_v[i] > bits_max) bits_max = bits_v[i];
+    g_pack_ns += now_ns() - t_pack0;
+    uint64_t t_stage0 = now_ns();
 
     /* assemble the pinned staging slab: [mat][job 0][job 1]... */
     const size_t mat_bytes = (size_t)abpt->m * abpt->m * 4;
@@ -403,6 +422,8 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
     }
     const size_t ssz = bits == 16 ? 2 : 4;
 
+    g_stage_ns += now_ns() - t_stage0;
+    uint64_t t_gpu0 = now_ns();
     int attempt = 0;
     for (;;) {
         /* arena layout */
@@ -440,9 +461,7 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
             jb.max_remain = (int*)(S + P.o_remain);
             jb.max_left = (int*)(S + P.o_ml);
             jb.max_right = (int*)(S + P.o_mr);
-            jb.dp_beg = (int*)(S + P.o_beg);
-            jb.dp_end = (int*)(S + P.o_end);
-            jb.row_off = (int64_t*)(S + P.o_roff);
+            jb.row_meta = S + P.o_meta;
             jb.mat = (int*)(S0 + o_mat);
             jb.arena = (uint8_t*)C.arena.p + (size_t)P.arena_off * 5 * ssz;
             jb.arena_cap = arena_est[i];
@@ -480,6 +499,8 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
         }
     }
 
+    g_gpu_ns += now_ns() - t_gpu0;
+    uint64_t t_unpack0 = now_ns();
     /* unpack results */
     C.hcig.clear();
     for (int i = 0; i < n_jobs; ++i) {
@@ -489,6 +510,7 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
             exit(EXIT_FAILURE);
         }
         g_dp_cells += (uint64_t)R.cells;
+        g_alg_bytes += (uint64_t)R.cells * 5 * ssz;
         abpoa_res_t *res = batch[i].res;
         res->best_score = R.best_score;
         if (batch[i].abpt->ret_cigar && R.n_cigar > 0) {
@@ -507,6 +529,7 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
         res->node_s = R.node_s; res->node_e = R.node_e;
         res->query_s = R.query_s; res->query_e = R.query_e;
     }
+    g_unpack_ns += now_ns() - t_unpack0;
     return 0;
 }
 

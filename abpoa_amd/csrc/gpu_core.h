@@ -23,6 +23,12 @@
 extern "C" {
 #endif
 
+/* per-row metadata written by the kernel as the band is discovered */
+typedef struct {
+    int32_t beg, end;
+    int64_t off;
+} abamd_row_meta_t;
+
 enum {
     ABAMD_JOB_OK = 0,
     ABAMD_JOB_ARENA_OVERFLOW = 1,
@@ -44,9 +50,8 @@ typedef struct {
     const int *max_remain;       /* [n_rows] remaining-path metadata */
     int *max_left;               /* [n_rows] adaptive band state (mutated) */
     int *max_right;
-    int *dp_beg;                 /* [n_rows] band bounds (device computed) */
-    int *dp_end;
-    int64_t *row_off;            /* [n_rows] arena offset of the row band */
+    void *row_meta;              /* [n_rows] abamd_row_meta_t: band bounds +
+                                    arena offset, one 16-byte load per row */
     void *arena;                 /* score_t[arena_cap * 5] */
     int64_t arena_cap;           /* capacity in cells per plane */
     uint64_t *cigar;             /* packed graph-CIGAR output (backtrack order) */

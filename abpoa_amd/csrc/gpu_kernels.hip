@@ -1,38 +1,42 @@
 /* CDNA4 (gfx950) aligner core: adaptive-banded sequence-to-graph DP.
  *
- * One wavefront (64 lanes) per alignment job. The wave sweeps the graph's
- * topologically-sorted rows; within a row the band is processed in 64-cell
- * chunks held in registers:
- *   - M/E gather from predecessor rows (coalesced 2- or 4-byte loads from the
- *     banded HBM arena),
+ * One wavefront (64 lanes) per alignment job, 4 jobs per 256-thread
+ * workgroup. The wave sweeps the graph's topologically-sorted rows; within a
+ * row the band is processed in 64-cell register chunks:
+ *   - M/E gather from predecessor rows. The common predecessor (in-degree is
+ *     ~1.05) is the immediately preceding row, whose H/E planes are kept in a
+ *     double-buffered LDS cache (ABAMD_BMAX cells) — a ~50-cycle LDS read
+ *     replaces a ~900-cycle HBM round-trip on the row-to-row dependent chain.
+ *     Other predecessors (and bands wider than the cache) read the banded
+ *     HBM arena with coalesced 2/4-byte loads.
  *   - the F (insertion) recurrence is a wavefront max-plus log-scan
- *     (__shfl_up, 6 steps) with a sequential carry between chunks,
- *   - the five planes (H,E1,E2,F1,F2) stream back to the arena with coalesced
- *     stores (they are re-read once by the in-kernel backtrack),
- *   - the row argmax (adaptive band steering) is a wavefront reduction.
- * Integer max-plus throughout: MFMA does not apply; the kernel is designed to
- * be HBM-bound on the 10 B/cell plane traffic.
+ *     (__shfl_up, 6 doubling steps) with a sequential carry between chunks,
+ *   - the five planes (H,E1,E2,F1,F2) stream to the HBM arena with coalesced
+ *     stores (re-read once by the in-kernel backtrack),
+ *   - the row argmax (adaptive band steering) is a wavefront reduction,
+ *   - per-row band metadata is one 16-byte abamd_row_meta_t load.
+ * Integer max-plus throughout: MFMA does not apply; the target bound is the
+ * 10 B/cell HBM plane traffic.
  *
- * Numerics: results are bit-identical to oracle/ref_core.c (and therefore to
- * the reference x86 build): int16 arithmetic wraps (non-saturating), achieved
- * by truncating to the score type after every add/sub. Recurrence and
- * backtrack follow abpoa_align_simd.c:935-1074 / :309-458; band formulas
+ * Numerics are bit-identical to oracle/ref_core.c (and the reference x86
+ * build): int16 arithmetic wraps (non-saturating _mm*_add_epi16 semantics),
+ * reproduced by truncating to the score type after every add/sub. Recurrence
+ * and backtrack follow abpoa_align_simd.c:935-1074 / :309-458; band formulas
  * abpoa_align.h:34-35.
  *
  * This file covers global alignment with convex gaps (the north-star path),
- * int16 with int32 overflow rescore. Local/extension and affine/linear gap
- * variants are host-dispatched and added next.
+ * int16 with int32 overflow rescore. Affine/linear gap and local/extension
+ * variants are host-dispatched and tracked in DESIGN.md §7.
  */
 #include <hip/hip_runtime.h>
 #include "gpu_core.h"
 
 #define WAVE 64
-#define JOBS_PER_BLOCK 4   /* 256-thread workgroups: 4 waves = 4 independent jobs,
-                              doubling resident waves/CU vs 64-thread blocks */
-
-template <typename S> struct ScoreLim;
-template <> struct ScoreLim<int16_t> { static constexpr int maxv = 32767; };
-template <> struct ScoreLim<int32_t> { static constexpr int maxv = 2147483647; };
+#define JOBS_PER_BLOCK 4
+/* LDS previous-row cache width (cells). Default bands are ~230-450 cells
+ * (w = wb + wf*qlen = 110 at the north-star shape); wider rows fall back to
+ * arena reads. */
+#define ABAMD_BMAX 512
 
 template <typename S>
 __device__ __forceinline__ S smax(S a, S b) { return a > b ? a : b; }
@@ -63,8 +67,10 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
     const int lane = threadIdx.x % WAVE;
 
     __shared__ int mat_lds[27 * 27];
-    /* every job of a batch shares one scoring matrix (gpu_align.cpp) */
+    /* per-job double-buffered previous-row cache: H,E1,E2 planes */
+    __shared__ S prev_lds[JOBS_PER_BLOCK][2][3 * ABAMD_BMAX];
     {
+        /* every job of a batch shares one scoring matrix (gpu_align.cpp) */
         const int m0 = jobs[0].m;
         const int *mat0 = jobs[0].mat;
         for (int i = threadIdx.x; i < m0 * m0; i += WAVE * JOBS_PER_BLOCK)
@@ -74,11 +80,13 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
     if (jid >= n_jobs) return;
     const abamd_gpu_job_t &jb = jobs[jid];
     abamd_gpu_res_t *res = &results[jid];
+    abamd_row_meta_t *__restrict__ meta = (abamd_row_meta_t*)jb.row_meta;
 
     const int qlen = jb.qlen, n_rows = jb.n_rows, w = jb.w, m = jb.m;
     const S inf_min = (S)jb.inf_min;
     const S o1 = (S)jb.o1, o2 = (S)jb.o2, e1 = (S)jb.e1, e2 = (S)jb.e2;
     const S oe1 = (S)jb.oe1, oe2 = (S)jb.oe2;
+    (void)o1; (void)o2;
     const int end_remain = jb.max_remain[n_rows - 1];
     S *arena = (S*)jb.arena;
     const uint8_t *__restrict__ query = jb.query;
@@ -99,6 +107,10 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         jb.max_left[o] = 1; jb.max_right[o] = 1;
     }
 
+    /* LDS cache state: which buffer holds the previous row's planes */
+    int buf_cur = 0;
+    int prev_ok = 0, prev_row = -1, prev_beg = 0, prev_end = -1;
+
     /* ---- first row (simd_abpoa_cg_first_dp) ---- */
     int64_t used;
     {
@@ -108,22 +120,31 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
             int t = jb.max_right[0] > qlen - mr ? jb.max_right[0] : qlen - mr;
             end0 = (qlen < t + w) ? qlen : t + w;
         } else end0 = qlen;
-        if (lane == 0) { jb.dp_beg[0] = 0; jb.dp_end[0] = end0; jb.row_off[0] = 0; }
+        if (lane == 0) { meta[0].beg = 0; meta[0].end = end0; meta[0].off = 0; }
         int64_t bw = end0 + 1;
         used = bw;
         S *H = arena, *E1 = arena + bw, *E2 = arena + 2 * bw, *F1 = arena + 3 * bw, *F2 = arena + 4 * bw;
+        S *c = &prev_lds[wid][buf_cur][0];
+        const int fits = end0 + 1 <= ABAMD_BMAX;
         for (int j = lane; j <= end0; j += WAVE) {
+            S hv, e1v2, e2v2;
             if (j == 0) {
-                H[0] = 0; E1[0] = (S)(0 - oe1); E2[0] = (S)(0 - oe2);
+                hv = 0; e1v2 = (S)(0 - oe1); e2v2 = (S)(0 - oe2);
                 F1[0] = inf_min; F2[0] = inf_min;
             } else {
                 S f1 = (S)(-(jb.o1 + jb.e1 * j));
                 S f2 = (S)(-(jb.o2 + jb.e2 * j));
                 F1[j] = f1; F2[j] = f2;
-                H[j] = smax(f1, f2);
-                E1[j] = inf_min; E2[j] = inf_min;
+                hv = smax(f1, f2);
+                e1v2 = inf_min; e2v2 = inf_min;
+            }
+            H[j] = hv; E1[j] = e1v2; E2[j] = e2v2;
+            if (fits) {
+                c[j] = hv; c[ABAMD_BMAX + j] = e1v2; c[2 * ABAMD_BMAX + j] = e2v2;
             }
         }
+        if (fits) { prev_ok = 1; prev_row = 0; prev_beg = 0; prev_end = end0; }
+        buf_cur ^= 1;
     }
 
     /* ---- main row loop ---- */
@@ -140,7 +161,7 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 end = hi + w; if (end > qlen) end = qlen;
                 int min_pre_beg = 0x7fffffff;
                 for (int k = pk0; k < pk1; ++k) {
-                    int pb = jb.dp_beg[jb.pre_idx[k]];
+                    int pb = meta[jb.pre_idx[k]].beg;
                     if (pb < min_pre_beg) min_pre_beg = pb;
                 }
                 if (beg < min_pre_beg) beg = min_pre_beg;
@@ -149,11 +170,14 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         const int64_t bw = end - beg + 1;
         if (used + bw > jb.arena_cap) { if (lane == 0) res->status = ABAMD_JOB_ARENA_OVERFLOW; return; }
         const int64_t off = used;
-        if (lane == 0) { jb.dp_beg[r] = beg; jb.dp_end[r] = end; jb.row_off[r] = off; }
+        if (lane == 0) { meta[r].beg = beg; meta[r].end = end; meta[r].off = off; }
         used += bw;
         S *H = arena + off * 5, *E1r = H + bw, *E2r = E1r + bw, *F1r = E2r + bw, *F2r = F1r + bw;
         const uint8_t base = jb.row_base[r];
         const int *mrow = &mat_lds[base * m];
+        const int cache_fits = bw <= ABAMD_BMAX;
+        S *cw = &prev_lds[wid][buf_cur][0];          /* written for the next row */
+        const S *cr = &prev_lds[wid][buf_cur ^ 1][0]; /* previous row's planes */
 
         S carry_h = inf_min, f1c = inf_min, f2c = inf_min;
         S lmax = inf_min; int lleft = -1, lright = -1;
@@ -165,19 +189,34 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
             for (int k = pk0; k < pk1; ++k) {
                 const int p = jb.pre_idx[k];
                 const S ps = (S)jb.pre_ps[k];
-                const int pbeg = jb.dp_beg[p], pend = jb.dp_end[p];
-                const int64_t pbw = pend - pbeg + 1;
-                const S *__restrict__ pH = arena + jb.row_off[p] * 5;
+                if (prev_ok && p == prev_row) {
+                    if (act) { /* LDS fast path */
+                        if (j - 1 >= prev_beg && j - 1 <= prev_end) {
+                            S v = (S)(cr[j - 1 - prev_beg] + ps);
+                            if (v > h) h = v;
+                        }
+                        if (j >= prev_beg && j <= prev_end) {
+                            S v1 = (S)(cr[ABAMD_BMAX + j - prev_beg] + ps);
+                            S v2 = (S)(cr[2 * ABAMD_BMAX + j - prev_beg] + ps);
+                            if (v1 > e1v) e1v = v1;
+                            if (v2 > e2v) e2v = v2;
+                        }
+                    }
+                    continue;
+                }
+                const abamd_row_meta_t pm = meta[p];
+                const int64_t pbw = pm.end - pm.beg + 1;
+                const S *__restrict__ pH = arena + pm.off * 5;
                 const S *__restrict__ pE1 = pH + pbw;
                 const S *__restrict__ pE2 = pE1 + pbw;
                 if (act) {
-                    if (j - 1 >= pbeg && j - 1 <= pend) {
-                        S v = (S)(pH[j - 1 - pbeg] + ps);
+                    if (j - 1 >= pm.beg && j - 1 <= pm.end) {
+                        S v = (S)(pH[j - 1 - pm.beg] + ps);
                         if (v > h) h = v;
                     }
-                    if (j >= pbeg && j <= pend) {
-                        S v1 = (S)(pE1[j - pbeg] + ps);
-                        S v2 = (S)(pE2[j - pbeg] + ps);
+                    if (j >= pm.beg && j <= pm.end) {
+                        S v1 = (S)(pE1[j - pm.beg] + ps);
+                        S v2 = (S)(pE2[j - pm.beg] + ps);
                         if (v1 > e1v) e1v = v1;
                         if (v2 > e2v) e2v = v2;
                     }
@@ -221,10 +260,19 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
             if (act) {
                 H[j - beg] = hf; E1r[j - beg] = e1n; E2r[j - beg] = e2n;
                 F1r[j - beg] = f1; F2r[j - beg] = f2;
+                if (cache_fits) {
+                    cw[j - beg] = hf;
+                    cw[ABAMD_BMAX + j - beg] = e1n;
+                    cw[2 * ABAMD_BMAX + j - beg] = e2n;
+                }
                 if (hf > lmax) { lmax = hf; lleft = j; lright = j; }
                 else if (hf == lmax) { lright = j; }
             }
         }
+
+        if (cache_fits) { prev_ok = 1; prev_row = r; prev_beg = beg; prev_end = end; }
+        else prev_ok = 0;
+        buf_cur ^= 1;
 
         /* row argmax reduce + adaptive band push (simd_abpoa_max_in_row / ada_max_i) */
         if (jb.banded) {
@@ -258,12 +306,10 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
     int best_i = 0, best_j = 0;
     for (int k = jb.pre_off[n_rows - 1]; k < jb.pre_off[n_rows]; ++k) {
         const int p = jb.pre_idx[k];
-        int e = jb.dp_end[p] < qlen ? jb.dp_end[p] : qlen;
-        const int pbeg = jb.dp_beg[p];
-        const int64_t pbw = jb.dp_end[p] - pbeg + 1;
-        const S *pH = arena + jb.row_off[p] * 5;
-        int32_t sc = (e >= pbeg) ? (int32_t)pH[e - pbeg] : jb.inf_min;
-        (void)pbw;
+        const abamd_row_meta_t pm = meta[p];
+        int e = pm.end < qlen ? pm.end : qlen;
+        const S *pH = arena + pm.off * 5;
+        int32_t sc = (e >= pm.beg) ? (int32_t)pH[e - pm.beg] : jb.inf_min;
         if (sc > best_score) { best_score = sc; best_i = p; best_j = e; }
     }
     res->best_score = best_score;
@@ -280,9 +326,10 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         int id = jb.row_node_id[bi];
         if (best_j < qlen) dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, qlen - best_j, -1, qlen - 1, &status);
         while (bi > 0 && bj > 0 && status == ABAMD_JOB_OK) {
-            const int rb = jb.dp_beg[bi], re = jb.dp_end[bi];
+            const abamd_row_meta_t bm = meta[bi];
+            const int rb = bm.beg, re = bm.end;
             const int64_t bw = re - rb + 1;
-            const S *H = arena + jb.row_off[bi] * 5;
+            const S *H = arena + bm.off * 5;
             const S *E1r = H + bw, *E2r = E1r + bw, *F1r = E2r + bw, *F2r = F1r + bw;
             const S Hj = (bj >= rb && bj <= re) ? H[bj - rb] : inf_min;
             const S Hjm1 = (bj - 1 >= rb && bj - 1 <= re) ? H[bj - 1 - rb] : inf_min;
@@ -301,9 +348,10 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 for (int k = pq0; k < pq1; ++k) {
                     const int p = jb.pre_idx[k];
                     const S ps = (S)jb.pre_ps[k];
-                    if (bj - 1 < jb.dp_beg[p] || bj - 1 > jb.dp_end[p]) continue;
-                    const S *pH = arena + jb.row_off[p] * 5;
-                    if ((S)(pH[bj - 1 - jb.dp_beg[p]] + s + ps) == Hj) {
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
+                    const S *pH = arena + pm.off * 5;
+                    if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
                         dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
                         bi = p; --bj; id = jb.row_node_id[bi]; hit = 1;
                         cur_op = 0x1f; ++n_aln; n_matched += is_match;
@@ -315,10 +363,11 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 for (int k = pq0; k < pq1; ++k) {
                     const int p = jb.pre_idx[k];
                     const S ps = (S)jb.pre_ps[k];
-                    if (bj < jb.dp_beg[p] || bj > jb.dp_end[p]) continue;
-                    const int poffc = bj - jb.dp_beg[p];
-                    const int64_t pbw = jb.dp_end[p] - jb.dp_beg[p] + 1;
-                    const S *pH = arena + jb.row_off[p] * 5;
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj < pm.beg || bj > pm.end) continue;
+                    const int poffc = bj - pm.beg;
+                    const int64_t pbw = pm.end - pm.beg + 1;
+                    const S *pH = arena + pm.off * 5;
                     const S *pE1 = pH + pbw, *pE2 = pE1 + pbw;
                     if (cur_op & 0x2) {
                         if (cur_op & 0x1) {
@@ -394,9 +443,10 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 for (int k = pq0; k < pq1; ++k) {
                     const int p = jb.pre_idx[k];
                     const S ps = (S)jb.pre_ps[k];
-                    if (bj - 1 < jb.dp_beg[p] || bj - 1 > jb.dp_end[p]) continue;
-                    const S *pH = arena + jb.row_off[p] * 5;
-                    if ((S)(pH[bj - 1 - jb.dp_beg[p]] + s + ps) == Hj) {
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
+                    const S *pH = arena + pm.off * 5;
+                    if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
                         dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
                         bi = p; --bj; id = jb.row_node_id[bi]; hit = 1;
                         cur_op = 0x1f; ++n_aln; n_matched += is_match;

@@ -224,8 +224,12 @@ def main():
             return
         import abpoa_amd
         cells, kns, launches = abpoa_amd.get_stats()
+        import ctypes as _ct
+        _ab = _ct.c_uint64()
+        abpoa_amd.lib().abpoa_amd_get_stats2(_ct.byref(_ab))
+        alg_bytes = _ab.value or cells * ALG_BYTES_PER_CELL
         gcells_s = cells / elapsed / 1e9
-        ach = (cells * ALG_BYTES_PER_CELL) / max(kns, 1)  # bytes/ns == GB/s
+        ach = alg_bytes / max(kns, 1)  # bytes/ns == GB/s
         roofline = {"bound": "hbm", "achieved": round(ach, 1), "peak": HBM_PEAK_GBPS,
                     "unit": "GB/s", "frac": round(ach / HBM_PEAK_GBPS, 4),
                     "traffic": None}
