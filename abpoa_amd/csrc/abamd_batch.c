@@ -155,7 +155,14 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
             while (done + take < n_jobs) {
                 abamd_batch_job_t *J = &jobs[done + take];
                 int w = abpt->wb < 0 ? J->qlen : abpt->wb + (int)(abpt->wf * J->qlen);
-                double est = (double)(J->ab->abg->node_n) * (2.0 * w + 160.0) * 5 * 2; /* int16 bytes */
+                int gn = J->ab->abg->node_n, qlen = J->qlen;
+                /* score width this job will run at (abpoa_align_simd.c:1293-1302) */
+                long max_score = (long)qlen * abpt->max_mat;
+                long alt = (long)(qlen > gn ? qlen : gn) * abpt->gap_ext1 + abpt->gap_open1;
+                if (alt > max_score) max_score = alt;
+                int oe1 = abpt->gap_open1 + abpt->gap_ext1, oe2 = abpt->gap_open2 + abpt->gap_ext2;
+                int ssz = (max_score <= 32767 - abpt->min_mis - oe1 - oe2) ? 2 : 4;
+                double est = (double)gn * (2.0 * w + 160.0) * 5 * ssz;
                 if (take > 0 && acc + est > budget_bytes) break;
                 acc += est; ++take;
             }

@@ -126,6 +126,12 @@ struct GpuCtx {
         HIP_CHECK(hipStreamCreate(&stream));
         HIP_CHECK(hipEventCreate(&ev0));
         HIP_CHECK(hipEventCreate(&ev1));
+        /* reserve the plane arena at the full batch memory budget once:
+         * re-growing a 100+ GB hipMalloc every round costs seconds */
+        double mem_gb = 48.0;
+        const char *s = getenv("ABPOA_AMD_MEM_GB");
+        if (s && *s) mem_gb = atof(s);
+        arena.ensure((size_t)(mem_gb * 1e9));
         init = true;
     }
 };
@@ -347,7 +353,7 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
     {
         int nthr = (int)std::thread::hardware_concurrency();
         if (nthr < 1) nthr = 1;
-        if (nthr > 16) nthr = 16;
+        if (nthr > 64) nthr = 64;
         if (nthr > n_jobs) nthr = n_jobs;
         std::atomic<int> next{0};
         auto worker = [&]() {
@@ -388,7 +394,7 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
     {
         int nthr = (int)std::thread::hardware_concurrency();
         if (nthr < 1) nthr = 1;
-        if (nthr > 16) nthr = 16;
+        if (nthr > 64) nthr = 64;
         std::atomic<int> next{0};
         auto worker = [&]() {
             for (;;) {

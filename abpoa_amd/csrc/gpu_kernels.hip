@@ -5,7 +5,7 @@
  * row the band is processed in 64-cell register chunks:
  *   - M/E gather from predecessor rows. The common predecessor (in-degree is
  *     ~1.05) is the immediately preceding row, whose H/E planes are kept in a
- *     double-buffered LDS cache (ABAMD_BMAX cells) — a ~50-cycle LDS read
+ *     double-buffered LDS cache (BMAX cells) — a ~50-cycle LDS read
  *     replaces a ~900-cycle HBM round-trip on the row-to-row dependent chain.
  *     Other predecessors (and bands wider than the cache) read the banded
  *     HBM arena with coalesced 2/4-byte loads.
@@ -37,6 +37,9 @@
  * (w = wb + wf*qlen = 110 at the north-star shape); wider rows fall back to
  * arena reads. */
 #define ABAMD_BMAX 512
+/* int32 rounds halve the cache width to keep LDS within 4 resident blocks/CU */
+template <typename S> struct BmaxOf { static constexpr int v = ABAMD_BMAX; };
+template <> struct BmaxOf<int32_t> { static constexpr int v = 384; };
 
 template <typename S>
 __device__ __forceinline__ S smax(S a, S b) { return a > b ? a : b; }
@@ -68,7 +71,8 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
 
     __shared__ int mat_lds[27 * 27];
     /* per-job double-buffered previous-row cache: H,E1,E2 planes */
-    __shared__ S prev_lds[JOBS_PER_BLOCK][2][3 * ABAMD_BMAX];
+    constexpr int BMAX = BmaxOf<S>::v;
+    __shared__ S prev_lds[JOBS_PER_BLOCK][2][3 * BMAX];
     {
         /* every job of a batch shares one scoring matrix (gpu_align.cpp) */
         const int m0 = jobs[0].m;
@@ -125,7 +129,7 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         used = bw;
         S *H = arena, *E1 = arena + bw, *E2 = arena + 2 * bw, *F1 = arena + 3 * bw, *F2 = arena + 4 * bw;
         S *c = &prev_lds[wid][buf_cur][0];
-        const int fits = end0 + 1 <= ABAMD_BMAX;
+        const int fits = end0 + 1 <= BMAX;
         for (int j = lane; j <= end0; j += WAVE) {
             S hv, e1v2, e2v2;
             if (j == 0) {
@@ -140,7 +144,7 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
             }
             H[j] = hv; E1[j] = e1v2; E2[j] = e2v2;
             if (fits) {
-                c[j] = hv; c[ABAMD_BMAX + j] = e1v2; c[2 * ABAMD_BMAX + j] = e2v2;
+                c[j] = hv; c[BMAX + j] = e1v2; c[2 * BMAX + j] = e2v2;
             }
         }
         if (fits) { prev_ok = 1; prev_row = 0; prev_beg = 0; prev_end = end0; }
@@ -175,7 +179,7 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         S *H = arena + off * 5, *E1r = H + bw, *E2r = E1r + bw, *F1r = E2r + bw, *F2r = F1r + bw;
         const uint8_t base = jb.row_base[r];
         const int *mrow = &mat_lds[base * m];
-        const int cache_fits = bw <= ABAMD_BMAX;
+        const int cache_fits = bw <= BMAX;
         S *cw = &prev_lds[wid][buf_cur][0];          /* written for the next row */
         const S *cr = &prev_lds[wid][buf_cur ^ 1][0]; /* previous row's planes */
 
@@ -196,8 +200,8 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                             if (v > h) h = v;
                         }
                         if (j >= prev_beg && j <= prev_end) {
-                            S v1 = (S)(cr[ABAMD_BMAX + j - prev_beg] + ps);
-                            S v2 = (S)(cr[2 * ABAMD_BMAX + j - prev_beg] + ps);
+                            S v1 = (S)(cr[BMAX + j - prev_beg] + ps);
+                            S v2 = (S)(cr[2 * BMAX + j - prev_beg] + ps);
                             if (v1 > e1v) e1v = v1;
                             if (v2 > e2v) e2v = v2;
                         }
@@ -262,8 +266,8 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 F1r[j - beg] = f1; F2r[j - beg] = f2;
                 if (cache_fits) {
                     cw[j - beg] = hf;
-                    cw[ABAMD_BMAX + j - beg] = e1n;
-                    cw[2 * ABAMD_BMAX + j - beg] = e2n;
+                    cw[BMAX + j - beg] = e1n;
+                    cw[2 * BMAX + j - beg] = e2n;
                 }
                 if (hf > lmax) { lmax = hf; lleft = j; lright = j; }
                 else if (hf == lmax) { lright = j; }
