@@ -162,7 +162,8 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
                 if (alt > max_score) max_score = alt;
                 int oe1 = abpt->gap_open1 + abpt->gap_ext1, oe2 = abpt->gap_open2 + abpt->gap_ext2;
                 int ssz = (max_score <= 32767 - abpt->min_mis - oe1 - oe2) ? 2 : 4;
-                double est = (double)gn * (2.0 * w + 160.0) * 5 * ssz;
+                int planes = abpt->gap_mode == ABPOA_CONVEX_GAP ? 5 : abpt->gap_mode == ABPOA_AFFINE_GAP ? 3 : 1;
+                double est = (double)gn * (2.0 * w + 160.0) * planes * ssz;
                 if (take > 0 && acc + est > budget_bytes) break;
                 acc += est; ++take;
             }

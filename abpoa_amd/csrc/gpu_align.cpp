@@ -334,12 +334,14 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
     GpuCtx &C = g_ctx;
     C.ensure_init();
     abpoa_para_t *abpt = batch[0].abpt;
-    if (abpt->gap_mode != ABPOA_CONVEX_GAP || abpt->align_mode != ABPOA_GLOBAL_MODE) {
-        fprintf(stderr, "[abpoa_amd] GPU core: gap_mode %d / align_mode %d not implemented yet "
-                        "(convex-gap global alignment only in this build)\n",
-                abpt->gap_mode, abpt->align_mode);
+    if (abpt->align_mode != ABPOA_GLOBAL_MODE) {
+        fprintf(stderr, "[abpoa_amd] GPU core: align_mode %d not implemented yet "
+                        "(global alignment only in this build; local/extension pending)\n",
+                abpt->align_mode);
         exit(EXIT_FAILURE);
     }
+    const int planes = abpt->gap_mode == ABPOA_CONVEX_GAP ? 5
+                     : abpt->gap_mode == ABPOA_AFFINE_GAP ? 3 : 1;
 
     uint64_t t_pack0 = now_ns();
     C.hjobs.resize(n_jobs);
@@ -438,7 +440,7 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
             packs[i].arena_off = arena_cells;
             arena_cells += arena_est[i];
         }
-        C.arena.ensure((size_t)arena_cells * 5 * ssz);
+        C.arena.ensure((size_t)arena_cells * planes * ssz);
 
         /* cigar buffers */
         int cig_total = 0;
@@ -469,7 +471,7 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
             jb.max_right = (int*)(S + P.o_mr);
             jb.row_meta = S + P.o_meta;
             jb.mat = (int*)(S0 + o_mat);
-            jb.arena = (uint8_t*)C.arena.p + (size_t)P.arena_off * 5 * ssz;
+            jb.arena = (uint8_t*)C.arena.p + (size_t)P.arena_off * planes * ssz;
             jb.arena_cap = arena_est[i];
             jb.cigar = (uint64_t*)C.cigars.p + P.cigar_off;
             jb.cigar_cap = 2 * P.qlen + 1024;
@@ -480,10 +482,16 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
         HIP_CHECK(hipMemcpyAsync(C.jobs.p, C.hjobs.data(), (size_t)n_jobs * sizeof(abamd_gpu_job_t), hipMemcpyHostToDevice, C.stream));
 
         HIP_CHECK(hipEventRecord(C.ev0, C.stream));
-        if (bits == 16)
-            abamd_launch_cg_i16((abamd_gpu_job_t*)C.jobs.p, (abamd_gpu_res_t*)C.results.p, n_jobs, C.stream);
-        else
-            abamd_launch_cg_i32((abamd_gpu_job_t*)C.jobs.p, (abamd_gpu_res_t*)C.results.p, n_jobs, C.stream);
+        {
+            abamd_gpu_job_t *J = (abamd_gpu_job_t*)C.jobs.p;
+            abamd_gpu_res_t *R = (abamd_gpu_res_t*)C.results.p;
+            if (abpt->gap_mode == ABPOA_CONVEX_GAP)
+                bits == 16 ? abamd_launch_cg_i16(J, R, n_jobs, C.stream) : abamd_launch_cg_i32(J, R, n_jobs, C.stream);
+            else if (abpt->gap_mode == ABPOA_AFFINE_GAP)
+                bits == 16 ? abamd_launch_ag_i16(J, R, n_jobs, C.stream) : abamd_launch_ag_i32(J, R, n_jobs, C.stream);
+            else
+                bits == 16 ? abamd_launch_lg_i16(J, R, n_jobs, C.stream) : abamd_launch_lg_i32(J, R, n_jobs, C.stream);
+        }
         HIP_CHECK(hipGetLastError());
         HIP_CHECK(hipEventRecord(C.ev1, C.stream));
         HIP_CHECK(hipMemcpyAsync(C.hres.data(), C.results.p, (size_t)n_jobs * sizeof(abamd_gpu_res_t), hipMemcpyDeviceToHost, C.stream));
@@ -516,7 +524,7 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
             exit(EXIT_FAILURE);
         }
         g_dp_cells += (uint64_t)R.cells;
-        g_alg_bytes += (uint64_t)R.cells * 5 * ssz;
+        g_alg_bytes += (uint64_t)R.cells * planes * ssz;
         abpoa_res_t *res = batch[i].res;
         res->best_score = R.best_score;
         if (batch[i].abpt->ret_cigar && R.n_cigar > 0) {

@@ -86,6 +86,14 @@ void abamd_launch_cg_i16(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_r
                          int n_jobs, void *stream);
 void abamd_launch_cg_i32(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
                          int n_jobs, void *stream);
+void abamd_launch_ag_i16(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
+                         int n_jobs, void *stream);
+void abamd_launch_ag_i32(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
+                         int n_jobs, void *stream);
+void abamd_launch_lg_i16(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
+                         int n_jobs, void *stream);
+void abamd_launch_lg_i32(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
+                         int n_jobs, void *stream);
 
 #ifdef __cplusplus
 }

@@ -484,3 +484,663 @@ extern "C" void abamd_launch_cg_i32(const abamd_gpu_job_t *dev_jobs, abamd_gpu_r
     hipLaunchKernelGGL((cg_global_kernel<int32_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
                        (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
 }
+
+/* ------------------------------------------------------------------ */
+/* Affine-gap global kernel: 3 planes (H,E1,F1).                       */
+/* Reference core simd_abpoa_ag_dp (abpoa_align_simd.c:817-933):       */
+/* the F chain feeds from H BEFORE the E fold, and the stored E is     */
+/* inf_min when the insertion won the cell (SIMDSetIfEqual, :930).     */
+/* Backtrack: simd_abpoa_ag_backtrack (:196-307).                      */
+/* ------------------------------------------------------------------ */
+template <typename S>
+__global__ __launch_bounds__(WAVE * JOBS_PER_BLOCK, 7)
+void ag_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
+                      abamd_gpu_res_t *__restrict__ results, int n_jobs) {
+    const int wid = threadIdx.x / WAVE;
+    const int jid = blockIdx.x * JOBS_PER_BLOCK + wid;
+    const int lane = threadIdx.x % WAVE;
+
+    __shared__ int mat_lds[27 * 27];
+    constexpr int BMAX = BmaxOf<S>::v;
+    /* cache H and E1 of the previous row */
+    __shared__ S prev_lds[JOBS_PER_BLOCK][2][2 * BMAX];
+    {
+        const int m0 = jobs[0].m;
+        const int *mat0 = jobs[0].mat;
+        for (int i = threadIdx.x; i < m0 * m0; i += WAVE * JOBS_PER_BLOCK)
+            mat_lds[i] = mat0[i];
+    }
+    __syncthreads();
+    if (jid >= n_jobs) return;
+    const abamd_gpu_job_t &jb = jobs[jid];
+    abamd_gpu_res_t *res = &results[jid];
+    abamd_row_meta_t *__restrict__ meta = (abamd_row_meta_t*)jb.row_meta;
+
+    const int qlen = jb.qlen, n_rows = jb.n_rows, w = jb.w, m = jb.m;
+    const S inf_min = (S)jb.inf_min;
+    const S e1 = (S)jb.e1, oe1 = (S)jb.oe1;
+    const int end_remain = jb.max_remain[n_rows - 1];
+    S *arena = (S*)jb.arena;
+    const uint8_t *__restrict__ query = jb.query;
+
+    if (lane == 0) { res->status = ABAMD_JOB_OK; res->n_cigar = 0; }
+
+    for (int i = lane; i < n_rows; i += WAVE) {
+        jb.max_left[i] = jb.node_n_init;
+        jb.max_right[i] = 0;
+    }
+    if (lane == 0) { jb.max_left[0] = 0; jb.max_right[0] = 0; }
+    for (int k = jb.out_off[0] + lane; k < jb.out_off[1]; k += WAVE) {
+        int o = jb.out_idx[k];
+        jb.max_left[o] = 1; jb.max_right[o] = 1;
+    }
+
+    int buf_cur = 0;
+    int prev_ok = 0, prev_row = -1, prev_beg = 0, prev_end = -1;
+
+    /* first row (simd_abpoa_ag_first_dp, abpoa_align_simd.c:651-667) */
+    int64_t used;
+    {
+        int mr = jb.max_remain[0] - end_remain - 1;
+        int end0;
+        if (jb.banded) {
+            int t = jb.max_right[0] > qlen - mr ? jb.max_right[0] : qlen - mr;
+            end0 = (qlen < t + w) ? qlen : t + w;
+        } else end0 = qlen;
+        if (lane == 0) { meta[0].beg = 0; meta[0].end = end0; meta[0].off = 0; }
+        int64_t bw = end0 + 1;
+        used = bw;
+        S *H = arena, *E1 = arena + bw, *F1 = arena + 2 * bw;
+        S *c = &prev_lds[wid][buf_cur][0];
+        const int fits = end0 + 1 <= BMAX;
+        for (int j = lane; j <= end0; j += WAVE) {
+            S hv, e1v2;
+            if (j == 0) {
+                hv = 0; e1v2 = (S)(0 - oe1);
+                F1[0] = inf_min;
+            } else {
+                S f1 = (S)(-(jb.o1 + jb.e1 * j));
+                F1[j] = f1;
+                hv = f1; e1v2 = inf_min;
+            }
+            H[j] = hv; E1[j] = e1v2;
+            if (fits) { c[j] = hv; c[BMAX + j] = e1v2; }
+        }
+        if (fits) { prev_ok = 1; prev_row = 0; prev_beg = 0; prev_end = end0; }
+        buf_cur ^= 1;
+    }
+
+    for (int r = 1; r < n_rows - 1; ++r) {
+        const int pk0 = jb.pre_off[r], pk1 = jb.pre_off[r + 1];
+        int beg, end;
+        {
+            int mr = jb.max_remain[r] - end_remain - 1;
+            if (jb.banded) {
+                int ml = jb.max_left[r], mrr = jb.max_right[r];
+                int lo = ml < qlen - mr ? ml : qlen - mr;
+                beg = lo - w; if (beg < 0) beg = 0;
+                int hi = mrr > qlen - mr ? mrr : qlen - mr;
+                end = hi + w; if (end > qlen) end = qlen;
+                int min_pre_beg = 0x7fffffff;
+                for (int k = pk0; k < pk1; ++k) {
+                    int pb = meta[jb.pre_idx[k]].beg;
+                    if (pb < min_pre_beg) min_pre_beg = pb;
+                }
+                if (beg < min_pre_beg) beg = min_pre_beg;
+            } else { beg = 0; end = qlen; }
+        }
+        const int64_t bw = end - beg + 1;
+        if (used + bw > jb.arena_cap) { if (lane == 0) res->status = ABAMD_JOB_ARENA_OVERFLOW; return; }
+        const int64_t off = used;
+        if (lane == 0) { meta[r].beg = beg; meta[r].end = end; meta[r].off = off; }
+        used += bw;
+        S *H = arena + off * 3, *E1r = H + bw, *F1r = E1r + bw;
+        const uint8_t base = jb.row_base[r];
+        const int *mrow = &mat_lds[base * m];
+        const int cache_fits = bw <= BMAX;
+        S *cw = &prev_lds[wid][buf_cur][0];
+        const S *cr = &prev_lds[wid][buf_cur ^ 1][0];
+
+        S carry_hm = inf_min, f1c = inf_min;
+        S lmax = inf_min; int lleft = -1, lright = -1;
+
+        for (int cs = beg; cs <= end; cs += WAVE) {
+            const int j = cs + lane;
+            const bool act = j <= end;
+            S h = inf_min, e1v = inf_min;
+            for (int k = pk0; k < pk1; ++k) {
+                const int p = jb.pre_idx[k];
+                const S ps = (S)jb.pre_ps[k];
+                if (prev_ok && p == prev_row) {
+                    if (act) {
+                        if (j - 1 >= prev_beg && j - 1 <= prev_end) {
+                            S v = (S)(cr[j - 1 - prev_beg] + ps);
+                            if (v > h) h = v;
+                        }
+                        if (j >= prev_beg && j <= prev_end) {
+                            S v1 = (S)(cr[BMAX + j - prev_beg] + ps);
+                            if (v1 > e1v) e1v = v1;
+                        }
+                    }
+                    continue;
+                }
+                const abamd_row_meta_t pm = meta[p];
+                const int64_t pbw = pm.end - pm.beg + 1;
+                const S *__restrict__ pH = arena + pm.off * 3;
+                const S *__restrict__ pE1 = pH + pbw;
+                if (act) {
+                    if (j - 1 >= pm.beg && j - 1 <= pm.end) {
+                        S v = (S)(pH[j - 1 - pm.beg] + ps);
+                        if (v > h) h = v;
+                    }
+                    if (j >= pm.beg && j <= pm.end) {
+                        S v1 = (S)(pE1[j - pm.beg] + ps);
+                        if (v1 > e1v) e1v = v1;
+                    }
+                }
+            }
+            const S q = (S)((j == 0 || !act) ? 0 : mrow[query[j - 1]]);
+            S hm = (S)(h + q);   /* M+q: the F chain feeds from this, pre-E */
+            if (!act) hm = inf_min;
+
+            S hmshift = (S)__shfl_up((int)hm, 1);
+            S c1;
+            if (lane == 0) {
+                if (cs == beg) c1 = (S)(inf_min - oe1);
+                else c1 = smax((S)(carry_hm - oe1), (S)(f1c - e1));
+            } else c1 = (S)(hmshift - oe1);
+            S f1 = c1;
+            #pragma unroll
+            for (int sft = 1; sft < WAVE; sft <<= 1) {
+                S t1 = (S)__shfl_up((int)f1, sft);
+                if (lane >= sft) f1 = smax(f1, (S)(t1 - (S)(sft * jb.e1)));
+            }
+            carry_hm = (S)__shfl((int)hm, WAVE - 1);
+            f1c = (S)__shfl((int)f1, WAVE - 1);
+
+            S tmp = smax(hm, e1v);
+            S hf = smax(tmp, f1);
+            S e1n = (hf == tmp) ? smax((S)(e1v - e1), (S)(hf - oe1)) : inf_min;
+            if (act) {
+                H[j - beg] = hf; E1r[j - beg] = e1n; F1r[j - beg] = f1;
+                if (cache_fits) { cw[j - beg] = hf; cw[BMAX + j - beg] = e1n; }
+                if (hf > lmax) { lmax = hf; lleft = j; lright = j; }
+                else if (hf == lmax) { lright = j; }
+            }
+        }
+
+        if (cache_fits) { prev_ok = 1; prev_row = r; prev_beg = beg; prev_end = end; }
+        else prev_ok = 0;
+        buf_cur ^= 1;
+
+        if (jb.banded) {
+            int mv = (int)lmax;
+            #pragma unroll
+            for (int sft = 32; sft >= 1; sft >>= 1) {
+                int o = __shfl_xor(mv, sft);
+                if (o > mv) mv = o;
+            }
+            int ll = ((int)lmax == mv && lleft >= 0) ? lleft : 0x7fffffff;
+            int rr = ((int)lmax == mv && lright >= 0) ? lright : -1;
+            #pragma unroll
+            for (int sft = 32; sft >= 1; sft >>= 1) {
+                int lo = __shfl_xor(ll, sft); if (lo < ll) ll = lo;
+                int ro = __shfl_xor(rr, sft); if (ro > rr) rr = ro;
+            }
+            for (int k = jb.out_off[r] + lane; k < jb.out_off[r + 1]; k += WAVE) {
+                int o = jb.out_idx[k];
+                if (rr + 1 > jb.max_right[o]) jb.max_right[o] = rr + 1;
+                if (ll + 1 < jb.max_left[o]) jb.max_left[o] = ll + 1;
+            }
+        }
+    }
+
+    if (lane == 0) res->cells = used;
+    if (lane != 0) return;
+
+    int32_t best_score = jb.inf_min;
+    int best_i = 0, best_j = 0;
+    for (int k = jb.pre_off[n_rows - 1]; k < jb.pre_off[n_rows]; ++k) {
+        const int p = jb.pre_idx[k];
+        const abamd_row_meta_t pm = meta[p];
+        int e = pm.end < qlen ? pm.end : qlen;
+        const S *pH = arena + pm.off * 3;
+        int32_t sc = (e >= pm.beg) ? (int32_t)pH[e - pm.beg] : jb.inf_min;
+        if (sc > best_score) { best_score = sc; best_i = p; best_j = e; }
+    }
+    res->best_score = best_score;
+    res->best_i = best_i; res->best_j = best_j;
+    if (!jb.ret_cigar) return;
+
+    { /* simd_abpoa_ag_backtrack (:196-307) */
+        int bi = best_i, bj = best_j, start_i = best_i, start_j = best_j;
+        int cur_op = 0x1f, n_c = 0, status = ABAMD_JOB_OK;
+        int look_end = jb.put_gap_at_end, put_right = jb.put_gap_on_right;
+        int n_aln = 0, n_matched = 0;
+        uint64_t *cig = jb.cigar;
+        int id = jb.row_node_id[bi];
+        if (best_j < qlen) dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, qlen - best_j, -1, qlen - 1, &status);
+        while (bi > 0 && bj > 0 && status == ABAMD_JOB_OK) {
+            const abamd_row_meta_t bm = meta[bi];
+            const int rb = bm.beg, re = bm.end;
+            const int64_t bw = re - rb + 1;
+            const S *H = arena + bm.off * 3;
+            const S *E1r = H + bw, *F1r = E1r + bw;
+            const S Hj = (bj >= rb && bj <= re) ? H[bj - rb] : inf_min;
+            const S Hjm1 = (bj - 1 >= rb && bj - 1 <= re) ? H[bj - 1 - rb] : inf_min;
+            const S E1j = (bj >= rb && bj <= re) ? E1r[bj - rb] : inf_min;
+            const S F1j = (bj >= rb && bj <= re) ? F1r[bj - rb] : inf_min;
+            const S F1jm1 = (bj - 1 >= rb && bj - 1 <= re) ? F1r[bj - 1 - rb] : inf_min;
+            start_i = bi; start_j = bj;
+            const int pq0 = jb.pre_off[bi], pq1 = jb.pre_off[bi + 1];
+            const S s = (S)mat_lds[m * jb.row_base[bi] + query[bj - 1]];
+            const int is_match = jb.row_base[bi] == query[bj - 1];
+            int hit = 0;
+            if (put_right == 0 && look_end == 0 && (cur_op & 0x1)) {
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
+                    const S *pH = arena + pm.off * 3;
+                    if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
+                        cur_op = 0x1f; hit = 1;
+                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
+                        bi = p; --bj; id = jb.row_node_id[bi];
+                        ++n_aln; n_matched += is_match;
+                        break;
+                    }
+                }
+            }
+            if (!hit && (cur_op & 0x2)) { /* deletion */
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj < pm.beg || bj > pm.end) continue;
+                    const int poffc = bj - pm.beg;
+                    const int64_t pbw = pm.end - pm.beg + 1;
+                    const S *pH = arena + pm.off * 3;
+                    const S *pE1 = pH + pbw;
+                    if (cur_op & 0x1) {
+                        if (Hj == (S)(pE1[poffc] + ps)) {
+                            cur_op = ((S)(pH[poffc] - oe1) == pE1[poffc]) ? (0x1 | 0x18) : 0x2;
+                            hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                            bi = p; id = jb.row_node_id[bi];
+                            if (look_end) look_end = 0;
+                            break;
+                        }
+                    } else {
+                        if (E1j == (S)(pE1[poffc] - e1 + ps)) {
+                            cur_op = ((S)(pH[poffc] - oe1) == pE1[poffc]) ? (0x1 | 0x18) : 0x2;
+                            hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                            bi = p; id = jb.row_node_id[bi];
+                            if (look_end) look_end = 0;
+                            break;
+                        }
+                    }
+                }
+            }
+            if (!hit && (cur_op & 0x18)) { /* insertion */
+                if (cur_op & 0x1) {
+                    if (Hj == F1j) {
+                        if ((S)(Hjm1 - oe1) == F1j) { cur_op = 0x1 | 0x6; hit = 1; }
+                        else if ((S)(F1jm1 - e1) == F1j) { cur_op = 0x8; hit = 1; }
+                    }
+                } else {
+                    if ((S)(Hjm1 - oe1) == F1j) { cur_op = 0x1 | 0x6; hit = 1; }
+                    else if ((S)(F1jm1 - e1) == F1j) { cur_op = 0x8; hit = 1; }
+                }
+                if (hit) {
+                    dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, 1, id, bj - 1, &status);
+                    --bj;
+                    if (look_end) look_end = 0;
+                    ++n_aln;
+                }
+            }
+            if (!hit && (cur_op & 0x1)) {
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
+                    const S *pH = arena + pm.off * 3;
+                    if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
+                        cur_op = 0x1f; hit = 1;
+                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
+                        bi = p; --bj; id = jb.row_node_id[bi];
+                        ++n_aln; n_matched += is_match;
+                        look_end = 0;
+                        break;
+                    }
+                }
+            }
+            if (!hit) { status = ABAMD_JOB_BT_DEAD_END; break; }
+        }
+        if (status == ABAMD_JOB_OK && bj > 0)
+            dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, bj, -1, bj - 1, &status);
+        res->status = status;
+        res->n_cigar = n_c;
+        res->n_aln_bases = n_aln;
+        res->n_matched_bases = n_matched;
+        res->node_e = jb.row_node_id[best_i]; res->query_e = best_j - 1;
+        res->node_s = jb.row_node_id[start_i]; res->query_s = start_j - 1;
+    }
+}
+
+/* ------------------------------------------------------------------ */
+/* Linear-gap global kernel: one H plane.                              */
+/* Reference core simd_abpoa_lg_dp (abpoa_align_simd.c:727-815):       */
+/* H = max over preds of (preH[j-1]+q, preH[j]-e1), then an in-row     */
+/* max-plus insertion scan directly on H.                              */
+/* Backtrack: simd_abpoa_lg_backtrack (:116-194).                      */
+/* ------------------------------------------------------------------ */
+template <typename S>
+__global__ __launch_bounds__(WAVE * JOBS_PER_BLOCK, 7)
+void lg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
+                      abamd_gpu_res_t *__restrict__ results, int n_jobs) {
+    const int wid = threadIdx.x / WAVE;
+    const int jid = blockIdx.x * JOBS_PER_BLOCK + wid;
+    const int lane = threadIdx.x % WAVE;
+
+    __shared__ int mat_lds[27 * 27];
+    constexpr int BMAX = BmaxOf<S>::v;
+    __shared__ S prev_lds[JOBS_PER_BLOCK][2][BMAX];
+    {
+        const int m0 = jobs[0].m;
+        const int *mat0 = jobs[0].mat;
+        for (int i = threadIdx.x; i < m0 * m0; i += WAVE * JOBS_PER_BLOCK)
+            mat_lds[i] = mat0[i];
+    }
+    __syncthreads();
+    if (jid >= n_jobs) return;
+    const abamd_gpu_job_t &jb = jobs[jid];
+    abamd_gpu_res_t *res = &results[jid];
+    abamd_row_meta_t *__restrict__ meta = (abamd_row_meta_t*)jb.row_meta;
+
+    const int qlen = jb.qlen, n_rows = jb.n_rows, w = jb.w, m = jb.m;
+    const S inf_min = (S)jb.inf_min;
+    const S e1 = (S)jb.e1;
+    const int end_remain = jb.max_remain[n_rows - 1];
+    S *arena = (S*)jb.arena;
+    const uint8_t *__restrict__ query = jb.query;
+
+    if (lane == 0) { res->status = ABAMD_JOB_OK; res->n_cigar = 0; }
+
+    for (int i = lane; i < n_rows; i += WAVE) {
+        jb.max_left[i] = jb.node_n_init;
+        jb.max_right[i] = 0;
+    }
+    if (lane == 0) { jb.max_left[0] = 0; jb.max_right[0] = 0; }
+    for (int k = jb.out_off[0] + lane; k < jb.out_off[1]; k += WAVE) {
+        int o = jb.out_idx[k];
+        jb.max_left[o] = 1; jb.max_right[o] = 1;
+    }
+
+    int buf_cur = 0;
+    int prev_ok = 0, prev_row = -1, prev_beg = 0, prev_end = -1;
+
+    /* first row (simd_abpoa_lg_first_dp, abpoa_align_simd.c:635-649) */
+    int64_t used;
+    {
+        int mr = jb.max_remain[0] - end_remain - 1;
+        int end0;
+        if (jb.banded) {
+            int t = jb.max_right[0] > qlen - mr ? jb.max_right[0] : qlen - mr;
+            end0 = (qlen < t + w) ? qlen : t + w;
+        } else end0 = qlen;
+        if (lane == 0) { meta[0].beg = 0; meta[0].end = end0; meta[0].off = 0; }
+        used = end0 + 1;
+        S *H = arena;
+        S *c = &prev_lds[wid][buf_cur][0];
+        const int fits = end0 + 1 <= BMAX;
+        for (int j = lane; j <= end0; j += WAVE) {
+            S hv = (S)(-jb.e1 * j);
+            H[j] = hv;
+            if (fits) c[j] = hv;
+        }
+        if (fits) { prev_ok = 1; prev_row = 0; prev_beg = 0; prev_end = end0; }
+        buf_cur ^= 1;
+    }
+
+    for (int r = 1; r < n_rows - 1; ++r) {
+        const int pk0 = jb.pre_off[r], pk1 = jb.pre_off[r + 1];
+        int beg, end;
+        {
+            int mr = jb.max_remain[r] - end_remain - 1;
+            if (jb.banded) {
+                int ml = jb.max_left[r], mrr = jb.max_right[r];
+                int lo = ml < qlen - mr ? ml : qlen - mr;
+                beg = lo - w; if (beg < 0) beg = 0;
+                int hi = mrr > qlen - mr ? mrr : qlen - mr;
+                end = hi + w; if (end > qlen) end = qlen;
+                int min_pre_beg = 0x7fffffff;
+                for (int k = pk0; k < pk1; ++k) {
+                    int pb = meta[jb.pre_idx[k]].beg;
+                    if (pb < min_pre_beg) min_pre_beg = pb;
+                }
+                if (beg < min_pre_beg) beg = min_pre_beg;
+            } else { beg = 0; end = qlen; }
+        }
+        const int64_t bw = end - beg + 1;
+        if (used + bw > jb.arena_cap) { if (lane == 0) res->status = ABAMD_JOB_ARENA_OVERFLOW; return; }
+        const int64_t off = used;
+        if (lane == 0) { meta[r].beg = beg; meta[r].end = end; meta[r].off = off; }
+        used += bw;
+        S *H = arena + off;
+        const uint8_t base = jb.row_base[r];
+        const int *mrow = &mat_lds[base * m];
+        const int cache_fits = bw <= BMAX;
+        S *cw = &prev_lds[wid][buf_cur][0];
+        const S *cr = &prev_lds[wid][buf_cur ^ 1][0];
+
+        S carry_h = inf_min;
+        S lmax = inf_min; int lleft = -1, lright = -1;
+
+        for (int cs = beg; cs <= end; cs += WAVE) {
+            const int j = cs + lane;
+            const bool act = j <= end;
+            const S q = (S)((j == 0 || !act) ? 0 : mrow[query[j - 1]]);
+            S h = inf_min;
+            for (int k = pk0; k < pk1; ++k) {
+                const int p = jb.pre_idx[k];
+                const S ps = (S)jb.pre_ps[k];
+                if (prev_ok && p == prev_row) {
+                    if (act) {
+                        if (j - 1 >= prev_beg && j - 1 <= prev_end) {
+                            S v = (S)(cr[j - 1 - prev_beg] + ps + q);
+                            if (v > h) h = v;
+                        }
+                        if (j >= prev_beg && j <= prev_end) {
+                            S v = (S)(cr[j - prev_beg] + ps - e1);
+                            if (v > h) h = v;
+                        }
+                    }
+                    continue;
+                }
+                const abamd_row_meta_t pm = meta[p];
+                const S *__restrict__ pH = arena + pm.off;
+                if (act) {
+                    if (j - 1 >= pm.beg && j - 1 <= pm.end) {
+                        S v = (S)(pH[j - 1 - pm.beg] + ps + q);
+                        if (v > h) h = v;
+                    }
+                    if (j >= pm.beg && j <= pm.end) {
+                        S v = (S)(pH[j - pm.beg] + ps - e1);
+                        if (v > h) h = v;
+                    }
+                }
+            }
+            if (!act) h = inf_min;
+            /* in-row insertion scan on H with inter-chunk carry */
+            S hs = (S)__shfl_up((int)h, 1);
+            if (lane == 0) {
+                if (cs != beg) h = smax(h, (S)(carry_h - e1));
+            } else h = smax(h, (S)(hs - e1));
+            #pragma unroll
+            for (int sft = 1; sft < WAVE; sft <<= 1) {
+                S t = (S)__shfl_up((int)h, sft);
+                if (lane >= sft) h = smax(h, (S)(t - (S)(sft * jb.e1)));
+            }
+            carry_h = (S)__shfl((int)h, WAVE - 1);
+            if (act) {
+                H[j - beg] = h;
+                if (cache_fits) cw[j - beg] = h;
+                if (h > lmax) { lmax = h; lleft = j; lright = j; }
+                else if (h == lmax) { lright = j; }
+            }
+        }
+
+        if (cache_fits) { prev_ok = 1; prev_row = r; prev_beg = beg; prev_end = end; }
+        else prev_ok = 0;
+        buf_cur ^= 1;
+
+        if (jb.banded) {
+            int mv = (int)lmax;
+            #pragma unroll
+            for (int sft = 32; sft >= 1; sft >>= 1) {
+                int o = __shfl_xor(mv, sft);
+                if (o > mv) mv = o;
+            }
+            int ll = ((int)lmax == mv && lleft >= 0) ? lleft : 0x7fffffff;
+            int rr = ((int)lmax == mv && lright >= 0) ? lright : -1;
+            #pragma unroll
+            for (int sft = 32; sft >= 1; sft >>= 1) {
+                int lo = __shfl_xor(ll, sft); if (lo < ll) ll = lo;
+                int ro = __shfl_xor(rr, sft); if (ro > rr) rr = ro;
+            }
+            for (int k = jb.out_off[r] + lane; k < jb.out_off[r + 1]; k += WAVE) {
+                int o = jb.out_idx[k];
+                if (rr + 1 > jb.max_right[o]) jb.max_right[o] = rr + 1;
+                if (ll + 1 < jb.max_left[o]) jb.max_left[o] = ll + 1;
+            }
+        }
+    }
+
+    if (lane == 0) res->cells = used;
+    if (lane != 0) return;
+
+    int32_t best_score = jb.inf_min;
+    int best_i = 0, best_j = 0;
+    for (int k = jb.pre_off[n_rows - 1]; k < jb.pre_off[n_rows]; ++k) {
+        const int p = jb.pre_idx[k];
+        const abamd_row_meta_t pm = meta[p];
+        int e = pm.end < qlen ? pm.end : qlen;
+        const S *pH = arena + pm.off;
+        int32_t sc = (e >= pm.beg) ? (int32_t)pH[e - pm.beg] : jb.inf_min;
+        if (sc > best_score) { best_score = sc; best_i = p; best_j = e; }
+    }
+    res->best_score = best_score;
+    res->best_i = best_i; res->best_j = best_j;
+    if (!jb.ret_cigar) return;
+
+    { /* simd_abpoa_lg_backtrack (:116-194) */
+        int bi = best_i, bj = best_j, start_i = best_i, start_j = best_j;
+        int n_c = 0, status = ABAMD_JOB_OK;
+        int look_end = jb.put_gap_at_end, put_right = jb.put_gap_on_right;
+        int n_aln = 0, n_matched = 0;
+        uint64_t *cig = jb.cigar;
+        int id = jb.row_node_id[bi];
+        if (best_j < qlen) dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, qlen - best_j, -1, qlen - 1, &status);
+        while (bi > 0 && bj > 0 && status == ABAMD_JOB_OK) {
+            const abamd_row_meta_t bm = meta[bi];
+            const int rb = bm.beg, re = bm.end;
+            const S *H = arena + bm.off;
+            const S Hj = (bj >= rb && bj <= re) ? H[bj - rb] : inf_min;
+            const S Hjm1 = (bj - 1 >= rb && bj - 1 <= re) ? H[bj - 1 - rb] : inf_min;
+            start_i = bi; start_j = bj;
+            const int pq0 = jb.pre_off[bi], pq1 = jb.pre_off[bi + 1];
+            const S s = (S)mat_lds[m * jb.row_base[bi] + query[bj - 1]];
+            const int is_match = jb.row_base[bi] == query[bj - 1];
+            int hit = 0;
+            if (put_right == 0 && look_end == 0) {
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
+                    const S *pH = arena + pm.off;
+                    if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
+                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
+                        bi = p; --bj; id = jb.row_node_id[bi]; hit = 1;
+                        ++n_aln; n_matched += is_match;
+                        break;
+                    }
+                }
+            }
+            if (!hit) { /* deletion */
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj < pm.beg || bj > pm.end) continue;
+                    const S *pH = arena + pm.off;
+                    if ((S)(pH[bj - pm.beg] - e1 + ps) == Hj) {
+                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                        bi = p; id = jb.row_node_id[bi]; hit = 1;
+                        if (look_end) look_end = 0;
+                        break;
+                    }
+                }
+            }
+            if (!hit) { /* insertion */
+                if ((S)(Hjm1 - e1) == Hj) {
+                    dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, 1, id, bj - 1, &status);
+                    --bj;
+                    if (look_end) look_end = 0;
+                    hit = 1; ++n_aln;
+                }
+            }
+            if (!hit) {
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
+                    const S *pH = arena + pm.off;
+                    if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
+                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
+                        bi = p; --bj; id = jb.row_node_id[bi]; hit = 1;
+                        ++n_aln; n_matched += is_match;
+                        look_end = 0;
+                        break;
+                    }
+                }
+            }
+            if (!hit) { status = ABAMD_JOB_BT_DEAD_END; break; }
+        }
+        if (status == ABAMD_JOB_OK && bj > 0)
+            dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, bj, -1, bj - 1, &status);
+        res->status = status;
+        res->n_cigar = n_c;
+        res->n_aln_bases = n_aln;
+        res->n_matched_bases = n_matched;
+        res->node_e = jb.row_node_id[best_i]; res->query_e = best_j - 1;
+        res->node_s = jb.row_node_id[start_i]; res->query_s = start_j - 1;
+    }
+}
+
+extern "C" void abamd_launch_ag_i16(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
+                                    int n_jobs, void *stream) {
+    int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
+    hipLaunchKernelGGL((ag_global_kernel<int16_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
+                       (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
+}
+extern "C" void abamd_launch_ag_i32(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
+                                    int n_jobs, void *stream) {
+    int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
+    hipLaunchKernelGGL((ag_global_kernel<int32_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
+                       (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
+}
+extern "C" void abamd_launch_lg_i16(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
+                                    int n_jobs, void *stream) {
+    int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
+    hipLaunchKernelGGL((lg_global_kernel<int16_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
+                       (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
+}
+extern "C" void abamd_launch_lg_i32(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
+                                    int n_jobs, void *stream) {
+    int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
+    hipLaunchKernelGGL((lg_global_kernel<int32_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
+                       (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
+}

@@ -67,3 +67,44 @@ def test_gpu_10kbp_set(gpu_bin, cputest_bin, oracle_env, tmp_path):
     gpu = run_stdout([gpu_bin, str(fa)])
     cpu = run_stdout([cputest_bin, str(fa)], env=oracle_env)
     assert gpu == cpu
+
+
+@pytest.mark.parametrize("opts", [["-O", "4", "-E", "2"], ["-O", "0", "-E", "2"]],
+                         ids=["affine", "linear"])
+def test_gpu_gap_modes_vs_oracle(gpu_bin, cputest_bin, oracle_env, tmp_path, opts):
+    """Affine and linear gap GPU kernels vs the CPU oracle (consensus + RC-MSA)."""
+    fa = tmp_path / "s.fa"
+    subprocess.run(["python3", os.path.join(ROOT, "tests", "make_synth.py"), str(fa),
+                    "--seed", "21", "--len", "1500", "--depth", "30"],
+                   check=True, stderr=subprocess.DEVNULL)
+    for extra in ([], ["-r1"]):
+        gpu = run_stdout([gpu_bin, str(fa)] + opts + extra)
+        cpu = run_stdout([cputest_bin, str(fa)] + opts + extra, env=oracle_env)
+        assert gpu == cpu, "GPU/oracle divergence opts=%r extra=%r" % (opts, extra)
+
+
+def test_gpu_aa_mode_vs_oracle(gpu_bin, cputest_bin, oracle_env, tmp_path):
+    """Amino-acid alphabet (m=27, BLOSUM62) on the GPU vs the oracle."""
+    import random
+    random.seed(6)
+    aa = "ARNDCQEGHILKMFPSTWYV"
+    refseq = "".join(random.choice(aa) for _ in range(600))
+    reads = []
+    for _ in range(12):
+        s = []
+        for ch in refseq:
+            r = random.random()
+            if r < 0.04: s.append(random.choice(aa))
+            elif r < 0.07: pass
+            elif r < 0.09: s.extend((ch, random.choice(aa)))
+            else: s.append(ch)
+        reads.append("".join(s))
+    fa = tmp_path / "aa.fa"
+    with open(fa, "w") as f:
+        for i, r in enumerate(reads):
+            f.write(">r%d\n%s\n" % (i, r))
+    mtx = os.path.join(GOLDEN, "BLOSUM62.mtx")
+    for opts in (["-c", "-t", mtx], ["-c", "-t", mtx, "-O", "4", "-E", "2"]):
+        gpu = run_stdout([gpu_bin, str(fa)] + opts)
+        cpu = run_stdout([cputest_bin, str(fa)] + opts, env=oracle_env)
+        assert gpu == cpu
