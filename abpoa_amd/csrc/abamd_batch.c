@@ -26,6 +26,9 @@ typedef struct {
     abpoa_res_t *res;
 } abamd_batch_job_t;
 int abamd_gpu_align_batch(abamd_batch_job_t *batch, int n_jobs);
+int abamd_gpu_batch_prepare(abamd_batch_job_t *batch, int n_jobs, int slot);
+int abamd_gpu_batch_launch(int slot);
+int abamd_gpu_batch_finish_slot(int slot);
 
 typedef struct {
     abpoa_t *ab;
@@ -127,57 +130,104 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
     }
     const double budget_bytes = mem_gb * 1e9;
 
-    for (r = 0; r < max_reads; ++r) {
-        /* collect this round's alignment jobs */
-        int n_jobs = 0;
-        for (i = 0; i < n_sets; ++i) {
-            sets[i].active = 0;
-            if (r >= sets[i].n_seqs) continue;
-            sets[i].active = 1;
-            memset(&sets[i].res, 0, sizeof(abpoa_res_t));
-            if (sets[i].ab->abg->node_n <= 2) continue; /* first read folds directly */
-            if (sets[i].ab->abg->is_topological_sorted == 0)
-                abpoa_topological_sort(sets[i].ab->abg, abpt);
-            abamd_batch_job_t *J = &jobs[n_jobs];
-            J->ab = sets[i].ab; J->abpt = abpt;
-            J->beg_node_id = ABPOA_SRC_NODE_ID; J->end_node_id = ABPOA_SINK_NODE_ID;
-            J->query = (uint8_t*)sets[i].seqs[r];
-            J->qlen = sets[i].seq_lens[r];
-            J->res = &sets[i].res;
-            job_set[n_jobs] = i;
-            ++n_jobs;
-        }
-        /* launch in memory-bounded chunks */
-        int done = 0;
-        while (done < n_jobs) {
-            double acc = 0.0;
-            int take = 0;
-            while (done + take < n_jobs) {
-                abamd_batch_job_t *J = &jobs[done + take];
-                int w = abpt->wb < 0 ? J->qlen : abpt->wb + (int)(abpt->wf * J->qlen);
-                int gn = J->ab->abg->node_n, qlen = J->qlen;
-                /* score width this job will run at (abpoa_align_simd.c:1293-1302) */
-                long max_score = (long)qlen * abpt->max_mat;
-                long alt = (long)(qlen > gn ? qlen : gn) * abpt->gap_ext1 + abpt->gap_open1;
-                if (alt > max_score) max_score = alt;
-                int oe1 = abpt->gap_open1 + abpt->gap_ext1, oe2 = abpt->gap_open2 + abpt->gap_ext2;
-                int ssz = (max_score <= 32767 - abpt->min_mis - oe1 - oe2) ? 2 : 4;
-                int planes = abpt->gap_mode == ABPOA_CONVEX_GAP ? 5 : abpt->gap_mode == ABPOA_AFFINE_GAP ? 3 : 1;
-                double est = (double)gn * (2.0 * w + 160.0) * planes * ssz;
-                if (take > 0 && acc + est > budget_bytes) break;
-                acc += est; ++take;
-            }
-            abamd_gpu_align_batch(&jobs[done], take);
-            done += take;
-        }
-        /* fold on host threads */
+    /* ---- round 0: first reads thread straight into their graphs ---- */
+    {
         double tf0 = abamd_realtime();
-        fold_work_t fw = { sets, n_sets, abpt, r, 0, PTHREAD_MUTEX_INITIALIZER };
-        int nt = n_host_threads;
-        for (i = 0; i < nt; ++i) pthread_create(&tids[i], NULL, fold_worker, &fw);
-        for (i = 0; i < nt; ++i) pthread_join(tids[i], NULL);
+        for (i = 0; i < n_sets; ++i) {
+            sets[i].active = sets[i].n_seqs > 0;
+            memset(&sets[i].res, 0, sizeof(abpoa_res_t));
+        }
+        fold_work_t fw = { sets, n_sets, abpt, 0, 0, PTHREAD_MUTEX_INITIALIZER };
+        for (i = 0; i < n_host_threads; ++i) pthread_create(&tids[i], NULL, fold_worker, &fw);
+        for (i = 0; i < n_host_threads; ++i) pthread_join(tids[i], NULL);
         g_fold_s += abamd_realtime() - tf0;
     }
+
+    /* ---- rounds 1..max: software pipeline over set groups -------------
+     * Group count is sized so one group-round fits the device memory
+     * budget at the LAST (largest) round; with >= 2 groups, the host fold
+     * and the next pack run while the other group's kernel executes. */
+    int n_groups = 2;
+    {
+        int w0 = abpt->wb < 0 ? 20000 : abpt->wb + (int)(abpt->wf * 20000);
+        double worst_gn = 20000.0 + 1.1 * (double)max_reads * 1000.0; /* coarse upper bound */
+        double per_job = worst_gn * (2.0 * w0 + 160.0) * 5 * 4;
+        int fit = (int)(budget_bytes / (per_job > 1 ? per_job : 1));
+        if (fit < 1) fit = 1;
+        n_groups = (n_sets + fit - 1) / fit;
+        if (n_groups < 2) n_groups = 2;
+        if (n_groups > n_sets) n_groups = n_sets > 0 ? n_sets : 1;
+    }
+    int *grp_of = (int*)abamd_malloc((size_t)n_sets * sizeof(int));
+    for (i = 0; i < n_sets; ++i) grp_of[i] = i % n_groups;
+
+    /* item = (round r >= 1, group g); g-major within a round */
+    long n_items = (long)(max_reads - 1) * n_groups;
+    abamd_batch_job_t *slot_jobs[2];
+    slot_jobs[0] = (abamd_batch_job_t*)abamd_malloc((size_t)n_sets * sizeof(abamd_batch_job_t));
+    slot_jobs[1] = (abamd_batch_job_t*)abamd_malloc((size_t)n_sets * sizeof(abamd_batch_job_t));
+    int slot_nj[2] = {0, 0};
+
+    #define ITEM_R(it) (1 + (int)((it) / n_groups))
+    #define ITEM_G(it) ((int)((it) % n_groups))
+
+    /* build the job list for one (round, group) item into a slot */
+    #define BUILD_ITEM(it, slot) do { \
+        int _r = ITEM_R(it), _g = ITEM_G(it), _nj = 0; \
+        for (i = 0; i < n_sets; ++i) { \
+            if (grp_of[i] != _g || _r >= sets[i].n_seqs) continue; \
+            memset(&sets[i].res, 0, sizeof(abpoa_res_t)); \
+            if (sets[i].ab->abg->node_n <= 2) continue; \
+            if (sets[i].ab->abg->is_topological_sorted == 0) \
+                abpoa_topological_sort(sets[i].ab->abg, abpt); \
+            abamd_batch_job_t *J = &slot_jobs[slot][_nj]; \
+            J->ab = sets[i].ab; J->abpt = abpt; \
+            J->beg_node_id = ABPOA_SRC_NODE_ID; J->end_node_id = ABPOA_SINK_NODE_ID; \
+            J->query = (uint8_t*)sets[i].seqs[_r]; \
+            J->qlen = sets[i].seq_lens[_r]; \
+            J->res = &sets[i].res; \
+            ++_nj; \
+        } \
+        slot_nj[slot] = _nj; \
+    } while (0)
+
+    /* fold one (round, group) item on the host thread pool */
+    #define FOLD_ITEM(it) do { \
+        double _tf0 = abamd_realtime(); \
+        int _r = ITEM_R(it), _g = ITEM_G(it); \
+        for (i = 0; i < n_sets; ++i) \
+            sets[i].active = (grp_of[i] == _g && _r < sets[i].n_seqs); \
+        fold_work_t fw = { sets, n_sets, abpt, _r, 0, PTHREAD_MUTEX_INITIALIZER }; \
+        int _nt = n_host_threads; \
+        for (i = 0; i < _nt; ++i) pthread_create(&tids[i], NULL, fold_worker, &fw); \
+        for (i = 0; i < _nt; ++i) pthread_join(tids[i], NULL); \
+        g_fold_s += abamd_realtime() - _tf0; \
+    } while (0)
+
+    if (n_items > 0) {
+        int slot = 0;
+        BUILD_ITEM(0, slot);
+        abamd_gpu_batch_prepare(slot_jobs[slot], slot_nj[slot], slot);
+        abamd_gpu_batch_launch(slot);
+        for (long it = 0; it < n_items; ++it) {
+            int nslot = slot ^ 1;
+            if (it + 1 < n_items) {
+                /* overlapped with item `it`'s kernel: fold the previous item,
+                 * then pack the next one (its group folded >= 1 item ago) */
+                if (it >= 1) FOLD_ITEM(it - 1);
+                BUILD_ITEM(it + 1, nslot);
+                abamd_gpu_batch_prepare(slot_jobs[nslot], slot_nj[nslot], nslot);
+                abamd_gpu_batch_finish_slot(slot);
+                abamd_gpu_batch_launch(nslot);
+            } else {
+                if (it >= 1) FOLD_ITEM(it - 1);
+                abamd_gpu_batch_finish_slot(slot);
+            }
+            slot = nslot;
+        }
+        FOLD_ITEM(n_items - 1);
+    }
+    free(grp_of); free(slot_jobs[0]); free(slot_jobs[1]);
 
     /* consensus on host threads, then emit callbacks in order */
     double tc0 = abamd_realtime();

@@ -103,7 +103,7 @@ struct PinnedBuf {
 struct GpuCtx {
     bool init = false;
     hipStream_t stream;
-    hipEvent_t ev0, ev1;
+    hipEvent_t ev0, ev1, ev_h2d;
     DevBuf slab;        /* all per-job inputs + metadata */
     DevBuf arena;       /* DP plane arena */
     DevBuf jobs, results, cigars;
@@ -126,6 +126,7 @@ struct GpuCtx {
         HIP_CHECK(hipStreamCreate(&stream));
         HIP_CHECK(hipEventCreate(&ev0));
         HIP_CHECK(hipEventCreate(&ev1));
+        HIP_CHECK(hipEventCreate(&ev_h2d));
         /* reserve the plane arena at the full batch memory budget once:
          * re-growing a 100+ GB hipMalloc every round costs seconds */
         double mem_gb = 48.0;
@@ -325,12 +326,44 @@ static void pick_width(abpoa_para_t *abpt, int qlen, int gn, int *bits, int *inf
 
 } // namespace
 
-/* Align a batch of independent jobs in one kernel launch. Returns 0.
- * All jobs must share gap_mode/align_mode (they do: one abpoa_para_t per run). */
+/* Pipelined batch interface: begin() packs, uploads and launches without
+ * waiting; finish() synchronizes and unpacks. The caller overlaps host-side
+ * graph folds of one set group with the kernel of the other. begin/finish
+ * pairs must not nest. abamd_gpu_align_batch() = begin + finish. */
 extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs);
+extern "C" int abamd_gpu_batch_prepare(BatchJob *batch, int n_jobs, int slot);
+extern "C" int abamd_gpu_batch_launch(int slot);
+extern "C" int abamd_gpu_batch_finish_slot(int slot);
+
+namespace {
+struct PendingBatch {
+    BatchJob *batch = nullptr;
+    int n_jobs = 0;
+    std::vector<JobPack> packs;
+    std::vector<int64_t> arena_est;
+    std::vector<abamd_gpu_job_t> hjobs;   /* per-slot: an async D2H/H2D may be
+    in flight for one slot while the other slot repacks */
+    std::vector<abamd_gpu_res_t> hres;
+    int bits = 16;
+    int planes = 5;
+    size_t total = 0;
+    bool active = false;
+};
+thread_local PendingBatch g_slots[2];
+}
+
+static int batch_launch(GpuCtx &C, PendingBatch &P); /* fwd */
 
 extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
-    if (n_jobs <= 0) return 0;
+    int r = abamd_gpu_batch_prepare(batch, n_jobs, 0);
+    if (r) return r;
+    r = abamd_gpu_batch_launch(0);
+    if (r) return r;
+    return abamd_gpu_batch_finish_slot(0);
+}
+
+extern "C" int abamd_gpu_batch_prepare(BatchJob *batch, int n_jobs, int slot) {
+    if (n_jobs <= 0) { g_slots[slot].active = false; g_slots[slot].n_jobs = 0; return 0; }
     GpuCtx &C = g_ctx;
     C.ensure_init();
     abpoa_para_t *abpt = batch[0].abpt;
@@ -344,8 +377,6 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
                      : abpt->gap_mode == ABPOA_AFFINE_GAP ? 3 : 1;
 
     uint64_t t_pack0 = now_ns();
-    C.hjobs.resize(n_jobs);
-    C.hres.resize(n_jobs);
     if ((int)C.jb_bufs.size() < n_jobs) C.jb_bufs.resize(n_jobs);
     std::vector<JobPack> packs(n_jobs);
     std::vector<int64_t> arena_est(n_jobs);
@@ -428,12 +459,46 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
             packs[i].jb.inf_min = im + 512 * ext_max;
         }
     }
-    const size_t ssz = bits == 16 ? 2 : 4;
-
+    PendingBatch &PB = g_slots[slot];
+    PB.batch = batch;
+    PB.n_jobs = n_jobs;
+    PB.packs = std::move(packs);
+    PB.arena_est = std::move(arena_est);
+    PB.bits = bits;
+    PB.planes = planes;
+    PB.total = total;
+    PB.hjobs.resize(n_jobs);
+    PB.hres.resize(n_jobs);
+    PB.active = true;
     g_stage_ns += now_ns() - t_stage0;
+    return 0;
+}
+
+extern "C" int abamd_gpu_batch_launch(int slot) {
+    GpuCtx &C = g_ctx;
+    PendingBatch &PB = g_slots[slot];
+    if (!PB.active) return 0;
     uint64_t t_gpu0 = now_ns();
-    int attempt = 0;
-    for (;;) {
+    int rr = batch_launch(C, PB);
+    /* wait for the H2D copies only: the caller may repack the pinned staging
+     * buffer for the other slot while this kernel runs */
+    HIP_CHECK(hipEventSynchronize(C.ev_h2d));
+    g_gpu_ns += now_ns() - t_gpu0;
+    return rr;
+}
+
+/* pack+upload happen in begin(); this issues the H2D + kernel (async) */
+static int batch_launch(GpuCtx &C, PendingBatch &PB) {
+    const int n_jobs = PB.n_jobs;
+    std::vector<JobPack> &packs = PB.packs;
+    std::vector<int64_t> &arena_est = PB.arena_est;
+    abpoa_para_t *abpt = PB.batch[0].abpt;
+    const int bits = PB.bits;
+    const int planes = PB.planes;
+    const size_t total = PB.total;
+    const size_t ssz = bits == 16 ? 2 : 4;
+    (void)abpt;
+    {
         /* arena layout */
         int64_t arena_cells = 0;
         for (int i = 0; i < n_jobs; ++i) {
@@ -470,16 +535,17 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
             jb.max_left = (int*)(S + P.o_ml);
             jb.max_right = (int*)(S + P.o_mr);
             jb.row_meta = S + P.o_meta;
-            jb.mat = (int*)(S0 + o_mat);
+            jb.mat = (int*)(S0 + 0); /* matrix is at slab offset 0 */
             jb.arena = (uint8_t*)C.arena.p + (size_t)P.arena_off * planes * ssz;
             jb.arena_cap = arena_est[i];
             jb.cigar = (uint64_t*)C.cigars.p + P.cigar_off;
             jb.cigar_cap = 2 * P.qlen + 1024;
-            C.hjobs[i] = jb;
+            PB.hjobs[i] = jb;
         }
         C.jobs.ensure((size_t)n_jobs * sizeof(abamd_gpu_job_t));
         C.results.ensure((size_t)n_jobs * sizeof(abamd_gpu_res_t));
-        HIP_CHECK(hipMemcpyAsync(C.jobs.p, C.hjobs.data(), (size_t)n_jobs * sizeof(abamd_gpu_job_t), hipMemcpyHostToDevice, C.stream));
+        HIP_CHECK(hipMemcpyAsync(C.jobs.p, PB.hjobs.data(), (size_t)n_jobs * sizeof(abamd_gpu_job_t), hipMemcpyHostToDevice, C.stream));
+        HIP_CHECK(hipEventRecord(C.ev_h2d, C.stream));
 
         HIP_CHECK(hipEventRecord(C.ev0, C.stream));
         {
@@ -494,31 +560,52 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
         }
         HIP_CHECK(hipGetLastError());
         HIP_CHECK(hipEventRecord(C.ev1, C.stream));
-        HIP_CHECK(hipMemcpyAsync(C.hres.data(), C.results.p, (size_t)n_jobs * sizeof(abamd_gpu_res_t), hipMemcpyDeviceToHost, C.stream));
-        HIP_CHECK(hipStreamSynchronize(C.stream));
+        HIP_CHECK(hipMemcpyAsync(PB.hres.data(), C.results.p, (size_t)n_jobs * sizeof(abamd_gpu_res_t), hipMemcpyDeviceToHost, C.stream));
+    }
+    return 0;
+}
 
+extern "C" int abamd_gpu_batch_finish_slot(int slot) {
+    GpuCtx &C = g_ctx;
+    PendingBatch &PB = g_slots[slot];
+    if (!PB.active) return 0;
+    const int n_jobs = PB.n_jobs;
+    BatchJob *batch = PB.batch;
+    std::vector<JobPack> &packs = PB.packs;
+    const size_t ssz = PB.bits == 16 ? 2 : 4;
+    const int planes = PB.planes;
+    uint64_t t_gpu0 = now_ns();
+    HIP_CHECK(hipStreamSynchronize(C.stream));
+    {
         float ms = 0.f;
         HIP_CHECK(hipEventElapsedTime(&ms, C.ev0, C.ev1));
         g_kernel_ns += (uint64_t)(ms * 1e6);
         g_launches += 1;
-
-        /* retry on arena overflow with doubled reservations */
+    }
+    /* rare: a job's adaptive band outgrew its reservation; relaunch the whole
+     * pending batch synchronously with doubled reservations */
+    for (int attempt = 0;; ++attempt) {
         bool overflow = false;
         for (int i = 0; i < n_jobs; ++i)
-            if (C.hres[i].status == ABAMD_JOB_ARENA_OVERFLOW) { arena_est[i] *= 2; overflow = true; }
+            if (PB.hres[i].status == ABAMD_JOB_ARENA_OVERFLOW) { PB.arena_est[i] *= 2; overflow = true; }
         if (!overflow) break;
-        if (++attempt > 8) {
+        if (attempt > 8) {
             fprintf(stderr, "[abpoa_amd] arena overflow persists after %d retries\n", attempt);
             exit(EXIT_FAILURE);
         }
+        batch_launch(C, PB);
+        HIP_CHECK(hipStreamSynchronize(C.stream));
+        float ms = 0.f;
+        HIP_CHECK(hipEventElapsedTime(&ms, C.ev0, C.ev1));
+        g_kernel_ns += (uint64_t)(ms * 1e6);
+        g_launches += 1;
     }
-
     g_gpu_ns += now_ns() - t_gpu0;
     uint64_t t_unpack0 = now_ns();
     /* unpack results */
     C.hcig.clear();
     for (int i = 0; i < n_jobs; ++i) {
-        abamd_gpu_res_t &R = C.hres[i];
+        abamd_gpu_res_t &R = PB.hres[i];
         if (R.status != ABAMD_JOB_OK) {
             fprintf(stderr, "[abpoa_amd] GPU job %d failed with status %d\n", i, R.status);
             exit(EXIT_FAILURE);
@@ -544,6 +631,7 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
         res->query_s = R.query_s; res->query_e = R.query_e;
     }
     g_unpack_ns += now_ns() - t_unpack0;
+    PB.active = false;
     return 0;
 }
 
