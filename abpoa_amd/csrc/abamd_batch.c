@@ -147,17 +147,11 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
      * Group count is sized so one group-round fits the device memory
      * budget at the LAST (largest) round; with >= 2 groups, the host fold
      * and the next pack run while the other group's kernel executes. */
-    int n_groups = 2;
-    {
-        int w0 = abpt->wb < 0 ? 20000 : abpt->wb + (int)(abpt->wf * 20000);
-        double worst_gn = 20000.0 + 1.1 * (double)max_reads * 1000.0; /* coarse upper bound */
-        double per_job = worst_gn * (2.0 * w0 + 160.0) * 5 * 4;
-        int fit = (int)(budget_bytes / (per_job > 1 ? per_job : 1));
-        if (fit < 1) fit = 1;
-        n_groups = (n_sets + fit - 1) / fit;
-        if (n_groups < 2) n_groups = 2;
-        if (n_groups > n_sets) n_groups = n_sets > 0 ? n_sets : 1;
-    }
+    /* two groups: every fold/pack overlaps the other group's kernel. Launch
+     * wall time is dominated by the per-row dependence chain, so FEWER,
+     * larger launches always win; if a group-round's arena demand exceeds
+     * the budget it falls back to sequential sub-chunks for that item. */
+    int n_groups = n_sets >= 2 ? 2 : 1;
     int *grp_of = (int*)abamd_malloc((size_t)n_sets * sizeof(int));
     for (i = 0; i < n_sets; ++i) grp_of[i] = i % n_groups;
 
@@ -171,9 +165,25 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
     #define ITEM_R(it) (1 + (int)((it) / n_groups))
     #define ITEM_G(it) ((int)((it) % n_groups))
 
+    double slot_est[2] = {0, 0};
+    int slot_big[2] = {0, 0};
+
+    /* arena bytes one job will demand (matches gpu_align.cpp's reservation) */
+    #define JOB_EST(J) ({ \
+        int _w = abpt->wb < 0 ? (J)->qlen : abpt->wb + (int)(abpt->wf * (J)->qlen); \
+        int _gn = (J)->ab->abg->node_n, _q = (J)->qlen; \
+        long _ms = (long)_q * abpt->max_mat; \
+        long _alt = (long)(_q > _gn ? _q : _gn) * abpt->gap_ext1 + abpt->gap_open1; \
+        if (_alt > _ms) _ms = _alt; \
+        int _oe1 = abpt->gap_open1 + abpt->gap_ext1, _oe2 = abpt->gap_open2 + abpt->gap_ext2; \
+        int _ssz = (_ms <= 32767 - abpt->min_mis - _oe1 - _oe2) ? 2 : 4; \
+        int _pl = abpt->gap_mode == ABPOA_CONVEX_GAP ? 5 : abpt->gap_mode == ABPOA_AFFINE_GAP ? 3 : 1; \
+        (double)_gn * (2.0 * _w + 160.0) * _pl * _ssz; })
+
     /* build the job list for one (round, group) item into a slot */
     #define BUILD_ITEM(it, slot) do { \
         int _r = ITEM_R(it), _g = ITEM_G(it), _nj = 0; \
+        double _est = 0; \
         for (i = 0; i < n_sets; ++i) { \
             if (grp_of[i] != _g || _r >= sets[i].n_seqs) continue; \
             memset(&sets[i].res, 0, sizeof(abpoa_res_t)); \
@@ -186,9 +196,25 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
             J->query = (uint8_t*)sets[i].seqs[_r]; \
             J->qlen = sets[i].seq_lens[_r]; \
             J->res = &sets[i].res; \
+            _est += JOB_EST(J); \
             ++_nj; \
         } \
-        slot_nj[slot] = _nj; \
+        slot_nj[slot] = _nj; slot_est[slot] = _est; slot_big[slot] = _est > budget_bytes; \
+    } while (0)
+
+    /* oversized item: memory-bounded sequential chunks (not pipelined) */
+    #define RUN_BIG_ITEM(slot) do { \
+        int _done = 0; \
+        while (_done < slot_nj[slot]) { \
+            double _acc = 0; int _take = 0; \
+            while (_done + _take < slot_nj[slot]) { \
+                double _e = JOB_EST(&slot_jobs[slot][_done + _take]); \
+                if (_take > 0 && _acc + _e > budget_bytes) break; \
+                _acc += _e; ++_take; \
+            } \
+            abamd_gpu_align_batch(&slot_jobs[slot][_done], _take); \
+            _done += _take; \
+        } \
     } while (0)
 
     /* fold one (round, group) item on the host thread pool */
@@ -207,8 +233,11 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
     if (n_items > 0) {
         int slot = 0;
         BUILD_ITEM(0, slot);
-        abamd_gpu_batch_prepare(slot_jobs[slot], slot_nj[slot], slot);
-        abamd_gpu_batch_launch(slot);
+        if (slot_big[slot]) RUN_BIG_ITEM(slot);
+        else {
+            abamd_gpu_batch_prepare(slot_jobs[slot], slot_nj[slot], slot);
+            abamd_gpu_batch_launch(slot);
+        }
         for (long it = 0; it < n_items; ++it) {
             int nslot = slot ^ 1;
             if (it + 1 < n_items) {
@@ -216,12 +245,14 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
                  * then pack the next one (its group folded >= 1 item ago) */
                 if (it >= 1) FOLD_ITEM(it - 1);
                 BUILD_ITEM(it + 1, nslot);
-                abamd_gpu_batch_prepare(slot_jobs[nslot], slot_nj[nslot], nslot);
-                abamd_gpu_batch_finish_slot(slot);
-                abamd_gpu_batch_launch(nslot);
+                if (!slot_big[nslot])
+                    abamd_gpu_batch_prepare(slot_jobs[nslot], slot_nj[nslot], nslot);
+                if (!slot_big[slot]) abamd_gpu_batch_finish_slot(slot);
+                if (slot_big[nslot]) RUN_BIG_ITEM(nslot);
+                else abamd_gpu_batch_launch(nslot);
             } else {
                 if (it >= 1) FOLD_ITEM(it - 1);
-                abamd_gpu_batch_finish_slot(slot);
+                if (!slot_big[slot]) abamd_gpu_batch_finish_slot(slot);
             }
             slot = nslot;
         }
