@@ -560,7 +560,8 @@ static int batch_launch(GpuCtx &C, PendingBatch &PB) {
         }
         HIP_CHECK(hipGetLastError());
         HIP_CHECK(hipEventRecord(C.ev1, C.stream));
-        HIP_CHECK(hipMemcpyAsync(PB.hres.data(), C.results.p, (size_t)n_jobs * sizeof(abamd_gpu_res_t), hipMemcpyDeviceToHost, C.stream));
+        /* results D2H happens in finish(): an async copy into pageable host
+         * memory would silently synchronize and stall the pipeline */
     }
     return 0;
 }
@@ -576,6 +577,7 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
     const int planes = PB.planes;
     uint64_t t_gpu0 = now_ns();
     HIP_CHECK(hipStreamSynchronize(C.stream));
+    HIP_CHECK(hipMemcpy(PB.hres.data(), C.results.p, (size_t)n_jobs * sizeof(abamd_gpu_res_t), hipMemcpyDeviceToHost));
     {
         float ms = 0.f;
         HIP_CHECK(hipEventElapsedTime(&ms, C.ev0, C.ev1));
@@ -595,6 +597,7 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
         }
         batch_launch(C, PB);
         HIP_CHECK(hipStreamSynchronize(C.stream));
+        HIP_CHECK(hipMemcpy(PB.hres.data(), C.results.p, (size_t)n_jobs * sizeof(abamd_gpu_res_t), hipMemcpyDeviceToHost));
         float ms = 0.f;
         HIP_CHECK(hipEventElapsedTime(&ms, C.ev0, C.ev1));
         g_kernel_ns += (uint64_t)(ms * 1e6);
