@@ -44,6 +44,34 @@ template <> struct BmaxOf<int32_t> { static constexpr int v = 384; };
 template <typename S>
 __device__ __forceinline__ S smax(S a, S b) { return a > b ? a : b; }
 
+/* Wave-wide inclusive max-plus scan with decay `e` per lane step:
+ *   out[l] = max_{k<=l} (in[k] - (l-k)*e)
+ * Intra-row (16-lane) steps use DPP row_shr (VALU latency) instead of
+ * ds_bpermute; only the two cross-row steps pay the LDS-unit latency.
+ * Arithmetic wraps at the score width after every op (reference int16
+ * semantics). `neutral` fills lanes with no source. */
+template <typename S>
+__device__ __forceinline__ S scan_maxplus(S f, int e, S neutral, int lane) {
+    (void)neutral;
+    /* lanes with no in-row source keep their own value (old = f), exactly
+     * like the guarded shuffle formulation — a synthetic floor could exceed
+     * the reference's stored out-of-band values and break bit-parity */
+    int cand;
+    cand = __builtin_amdgcn_update_dpp((int)f, (int)f, 0x111, 0xf, 0xf, false);
+    if ((lane & 15) >= 1) f = smax(f, (S)((S)cand - (S)(1 * e)));
+    cand = __builtin_amdgcn_update_dpp((int)f, (int)f, 0x112, 0xf, 0xf, false);
+    if ((lane & 15) >= 2) f = smax(f, (S)((S)cand - (S)(2 * e)));
+    cand = __builtin_amdgcn_update_dpp((int)f, (int)f, 0x114, 0xf, 0xf, false);
+    if ((lane & 15) >= 4) f = smax(f, (S)((S)cand - (S)(4 * e)));
+    cand = __builtin_amdgcn_update_dpp((int)f, (int)f, 0x118, 0xf, 0xf, false);
+    if ((lane & 15) >= 8) f = smax(f, (S)((S)cand - (S)(8 * e)));
+    S t = (S)__shfl_up((int)f, 16);
+    if (lane >= 16) f = smax(f, (S)(t - (S)(16 * e)));
+    t = (S)__shfl_up((int)f, 32);
+    if (lane >= 32) f = smax(f, (S)(t - (S)(32 * e)));
+    return f;
+}
+
 /* device push_cigar, matching abpoa_align.h:54-73 (run-length merge for I/S/H) */
 __device__ static int dev_push_cigar(uint64_t *cig, int *n_c, int cap, int op, int len,
                                      int node_id, int query_id, int *status) {
@@ -165,7 +193,8 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 end = hi + w; if (end > qlen) end = qlen;
                 int min_pre_beg = 0x7fffffff;
                 for (int k = pk0; k < pk1; ++k) {
-                    int pb = meta[jb.pre_idx[k]].beg;
+                    const int pidx = jb.pre_idx[k];
+                    int pb = (prev_ok && pidx == prev_row) ? prev_beg : meta[pidx].beg;
                     if (pb < min_pre_beg) min_pre_beg = pb;
                 }
                 if (beg < min_pre_beg) beg = min_pre_beg;
@@ -244,19 +273,11 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 c1 = (S)(hshift - oe1);
                 c2 = (S)(hshift - oe2);
             }
-            S f1 = c1, f2 = c2;
-            #pragma unroll
-            for (int sft = 1; sft < WAVE; sft <<= 1) {
-                S t1 = (S)__shfl_up((int)f1, sft);
-                S t2 = (S)__shfl_up((int)f2, sft);
-                if (lane >= sft) {
-                    f1 = smax(f1, (S)(t1 - (S)(sft * jb.e1)));
-                    f2 = smax(f2, (S)(t2 - (S)(sft * jb.e2)));
-                }
-            }
-            carry_h = (S)__shfl((int)hpre, WAVE - 1);
-            f1c = (S)__shfl((int)f1, WAVE - 1);
-            f2c = (S)__shfl((int)f2, WAVE - 1);
+            S f1 = scan_maxplus(c1, jb.e1, inf_min, lane);
+            S f2 = scan_maxplus(c2, jb.e2, inf_min, lane);
+            carry_h = (S)__builtin_amdgcn_readlane((int)hpre, WAVE - 1);
+            f1c = (S)__builtin_amdgcn_readlane((int)f1, WAVE - 1);
+            f2c = (S)__builtin_amdgcn_readlane((int)f2, WAVE - 1);
 
             S hf = smax(hpre, smax(f1, f2));
             S e1n = smax((S)(e1v - e1), (S)(hf - oe1));
@@ -583,7 +604,8 @@ void ag_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 end = hi + w; if (end > qlen) end = qlen;
                 int min_pre_beg = 0x7fffffff;
                 for (int k = pk0; k < pk1; ++k) {
-                    int pb = meta[jb.pre_idx[k]].beg;
+                    const int pidx = jb.pre_idx[k];
+                    int pb = (prev_ok && pidx == prev_row) ? prev_beg : meta[pidx].beg;
                     if (pb < min_pre_beg) min_pre_beg = pb;
                 }
                 if (beg < min_pre_beg) beg = min_pre_beg;
@@ -649,14 +671,9 @@ void ag_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 if (cs == beg) c1 = (S)(inf_min - oe1);
                 else c1 = smax((S)(carry_hm - oe1), (S)(f1c - e1));
             } else c1 = (S)(hmshift - oe1);
-            S f1 = c1;
-            #pragma unroll
-            for (int sft = 1; sft < WAVE; sft <<= 1) {
-                S t1 = (S)__shfl_up((int)f1, sft);
-                if (lane >= sft) f1 = smax(f1, (S)(t1 - (S)(sft * jb.e1)));
-            }
-            carry_hm = (S)__shfl((int)hm, WAVE - 1);
-            f1c = (S)__shfl((int)f1, WAVE - 1);
+            S f1 = scan_maxplus(c1, jb.e1, inf_min, lane);
+            carry_hm = (S)__builtin_amdgcn_readlane((int)hm, WAVE - 1);
+            f1c = (S)__builtin_amdgcn_readlane((int)f1, WAVE - 1);
 
             S tmp = smax(hm, e1v);
             S hf = smax(tmp, f1);
@@ -916,7 +933,8 @@ void lg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 end = hi + w; if (end > qlen) end = qlen;
                 int min_pre_beg = 0x7fffffff;
                 for (int k = pk0; k < pk1; ++k) {
-                    int pb = meta[jb.pre_idx[k]].beg;
+                    const int pidx = jb.pre_idx[k];
+                    int pb = (prev_ok && pidx == prev_row) ? prev_beg : meta[pidx].beg;
                     if (pb < min_pre_beg) min_pre_beg = pb;
                 }
                 if (beg < min_pre_beg) beg = min_pre_beg;
@@ -977,12 +995,8 @@ void lg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
             if (lane == 0) {
                 if (cs != beg) h = smax(h, (S)(carry_h - e1));
             } else h = smax(h, (S)(hs - e1));
-            #pragma unroll
-            for (int sft = 1; sft < WAVE; sft <<= 1) {
-                S t = (S)__shfl_up((int)h, sft);
-                if (lane >= sft) h = smax(h, (S)(t - (S)(sft * jb.e1)));
-            }
-            carry_h = (S)__shfl((int)h, WAVE - 1);
+            h = scan_maxplus(h, jb.e1, inf_min, lane);
+            carry_h = (S)__builtin_amdgcn_readlane((int)h, WAVE - 1);
             if (act) {
                 H[j - beg] = h;
                 if (cache_fits) cw[j - beg] = h;
