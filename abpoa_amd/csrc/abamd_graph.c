@@ -239,17 +239,36 @@ static int find_aligned_with_base(abpoa_graph_t *g, int node_id, uint8_t base) {
 
 /* ---------------- topological machinery ---------------- */
 
-/* simple int FIFO */
+/* simple int FIFO over a thread-local reusable buffer: the topological
+ * machinery runs once per read fold, and per-call malloc traffic serializes
+ * the 200+-thread fold pool on the allocator */
 typedef struct { int *a; int head, tail, cap; } ab_queue_t;
+static __thread int *tls_q_buf = NULL;
+static __thread int tls_q_cap = 0;
+static __thread int *tls_deg_buf = NULL;
+static __thread int tls_deg_cap = 0;
+
+static int *tls_degree(int n) {
+    if (n > tls_deg_cap) {
+        tls_deg_buf = (int*)abamd_realloc(tls_deg_buf, (size_t)(n * 2) * sizeof(int));
+        tls_deg_cap = n * 2;
+    }
+    return tls_deg_buf;
+}
 static void q_init(ab_queue_t *q, int cap_hint) {
-    q->cap = cap_hint > 16 ? cap_hint : 16;
-    q->a = (int*)abamd_malloc((size_t)q->cap * sizeof(int));
+    int want = cap_hint > 16 ? cap_hint : 16;
+    if (want > tls_q_cap) {
+        tls_q_buf = (int*)abamd_realloc(tls_q_buf, (size_t)(want * 2) * sizeof(int));
+        tls_q_cap = want * 2;
+    }
+    q->a = tls_q_buf; q->cap = tls_q_cap;
     q->head = q->tail = 0;
 }
 static void q_push(ab_queue_t *q, int v) {
     if (q->tail == q->cap) {
         q->cap <<= 1;
         q->a = (int*)abamd_realloc(q->a, (size_t)q->cap * sizeof(int));
+        tls_q_buf = q->a; tls_q_cap = q->cap;
     }
     q->a[q->tail++] = v;
 }
@@ -257,21 +276,21 @@ static int q_pop(ab_queue_t *q, int *v) {
     if (q->head == q->tail) return 0;
     *v = q->a[q->head++]; return 1;
 }
-static void q_free(ab_queue_t *q) { free(q->a); }
+static void q_free(ab_queue_t *q) { (void)q; /* thread-local, reused */ }
 
 /* Kahn BFS from src: assign dense topo indices; an aligned-node group enters
  * the queue together once every member's in-degree is exhausted
  * (abpoa_graph.c:221-266). */
 void abpoa_BFS_set_node_index(abpoa_graph_t *g, int src_id, int sink_id) {
     int i, j, cur, index = 0;
-    int *in_deg = (int*)abamd_malloc((size_t)g->node_n * sizeof(int));
+    int *in_deg = tls_degree(g->node_n);
     for (i = 0; i < g->node_n; ++i) in_deg[i] = g->node[i].in_edge_n;
     ab_queue_t q; q_init(&q, g->node_n);
     q_push(&q, src_id);
     while (q_pop(&q, &cur)) {
         g->index_to_node_id[index] = cur;
         g->node_id_to_index[cur] = index++;
-        if (cur == sink_id) { q_free(&q); free(in_deg); return; }
+        if (cur == sink_id) { q_free(&q); return; }
         for (i = 0; i < g->node[cur].out_edge_n; ++i) {
             int out = g->node[cur].out_id[i];
             if (--in_deg[out] == 0) {
@@ -292,7 +311,7 @@ void abpoa_BFS_set_node_index(abpoa_graph_t *g, int src_id, int sink_id) {
  * used by the adaptive band (abpoa_graph.c:268-309) */
 void abpoa_BFS_set_node_remain(abpoa_graph_t *g, int src_id, int sink_id) {
     int i, cur;
-    int *out_deg = (int*)abamd_malloc((size_t)g->node_n * sizeof(int));
+    int *out_deg = tls_degree(g->node_n);
     for (i = 0; i < g->node_n; ++i) {
         out_deg[i] = g->node[i].out_edge_n;
         g->node_id_to_max_remain[i] = 0;
@@ -311,7 +330,7 @@ void abpoa_BFS_set_node_remain(abpoa_graph_t *g, int src_id, int sink_id) {
             }
             g->node_id_to_max_remain[cur] = g->node_id_to_max_remain[max_id] + 1;
         }
-        if (cur == src_id) { q_free(&q); free(out_deg); return; }
+        if (cur == src_id) { q_free(&q); return; }
         for (i = 0; i < g->node[cur].in_edge_n; ++i) {
             int in = g->node[cur].in_id[i];
             if (--out_deg[in] == 0) q_push(&q, in);

@@ -160,8 +160,13 @@ static int64_t pack_job(HostBuf &hb, JobPack &P, abpoa_t *ab, abpoa_para_t *abpt
     int end_index = g->node_id_to_index[end_node_id];
     int span = end_index - beg_index + 1;
 
-    /* reachability map (abpoa_align_simd.c:1259-1269) */
-    std::vector<uint8_t> imap(g->node_n, 0);
+    /* reachability map (abpoa_align_simd.c:1259-1269); thread-local scratch
+     * (pack runs on a 64-thread pool: per-call allocation serializes on the
+     * allocator) */
+    static thread_local std::vector<uint8_t> imap_s;
+    static thread_local std::vector<int> idx2row_s;
+    imap_s.assign(g->node_n, 0);
+    std::vector<uint8_t> &imap = imap_s;
     imap[beg_index] = imap[end_index] = 1;
     for (int i = beg_index; i < end_index - 1; ++i) {
         if (!imap[i]) continue;
@@ -170,7 +175,8 @@ static int64_t pack_job(HostBuf &hb, JobPack &P, abpoa_t *ab, abpoa_para_t *abpt
             imap[g->node_id_to_index[g->node[nid].out_id[j]]] = 1;
     }
     /* compact row numbering */
-    std::vector<int> idx2row(span, -1);
+    idx2row_s.assign(span, -1);
+    std::vector<int> &idx2row = idx2row_s;
     int n_rows = 0;
     for (int i = 0; i < span; ++i)
         if (imap[beg_index + i]) idx2row[i] = n_rows++;
