@@ -65,10 +65,16 @@ __device__ __forceinline__ S scan_maxplus(S f, int e, S neutral, int lane) {
     if ((lane & 15) >= 4) f = smax(f, (S)((S)cand - (S)(4 * e)));
     cand = __builtin_amdgcn_update_dpp((int)f, (int)f, 0x118, 0xf, 0xf, false);
     if ((lane & 15) >= 8) f = smax(f, (S)((S)cand - (S)(8 * e)));
-    S t = (S)__shfl_up((int)f, 16);
-    if (lane >= 16) f = smax(f, (S)(t - (S)(16 * e)));
-    t = (S)__shfl_up((int)f, 32);
-    if (lane >= 32) f = smax(f, (S)(t - (S)(32 * e)));
+    /* cross-row propagation: each row's full prefix ends at its last lane;
+     * broadcast it (readlane, SALU) and decay by the per-lane distance.
+     * r after each step covers all rows up to that boundary, so coverage is
+     * complete (a plain +16/+32 jump after a row-local scan is NOT). */
+    S r = (S)__builtin_amdgcn_readlane((int)f, 15);
+    if (lane >= 16) f = smax(f, (S)(r - (S)((lane - 15) * e)));
+    r = (S)__builtin_amdgcn_readlane((int)f, 31);
+    if (lane >= 32) f = smax(f, (S)(r - (S)((lane - 31) * e)));
+    r = (S)__builtin_amdgcn_readlane((int)f, 47);
+    if (lane >= 48) f = smax(f, (S)(r - (S)((lane - 47) * e)));
     return f;
 }
 
