@@ -373,12 +373,6 @@ extern "C" int abamd_gpu_batch_prepare(BatchJob *batch, int n_jobs, int slot) {
     GpuCtx &C = g_ctx;
     C.ensure_init();
     abpoa_para_t *abpt = batch[0].abpt;
-    if (abpt->align_mode != ABPOA_GLOBAL_MODE) {
-        fprintf(stderr, "[abpoa_amd] GPU core: align_mode %d not implemented yet "
-                        "(global alignment only in this build; local/extension pending)\n",
-                abpt->align_mode);
-        exit(EXIT_FAILURE);
-    }
     const int planes = abpt->gap_mode == ABPOA_CONVEX_GAP ? 5
                      : abpt->gap_mode == ABPOA_AFFINE_GAP ? 3 : 1;
 

@@ -125,6 +125,10 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
     const S o1 = (S)jb.o1, o2 = (S)jb.o2, e1 = (S)jb.e1, e2 = (S)jb.e2;
     const S oe1 = (S)jb.oe1, oe2 = (S)jb.oe2;
     (void)o1; (void)o2;
+    const int local_mode = jb.align_mode == 1, extend_mode = jb.align_mode == 2;
+    int32_t run_best = jb.inf_min;      /* local/extend running best */
+    int run_best_i = 0, run_best_j = 0, run_best_remain = jb.max_remain[0];
+    int zdropped = 0;
     const int end_remain = jb.max_remain[n_rows - 1];
     S *arena = (S*)jb.arena;
     const uint8_t *__restrict__ query = jb.query;
@@ -166,7 +170,10 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         const int fits = end0 + 1 <= BMAX;
         for (int j = lane; j <= end0; j += WAVE) {
             S hv, e1v2, e2v2;
-            if (j == 0) {
+            if (local_mode) {
+                hv = 0; e1v2 = 0; e2v2 = 0;
+                F1[j] = 0; F2[j] = 0;
+            } else if (j == 0) {
                 hv = 0; e1v2 = (S)(0 - oe1); e2v2 = (S)(0 - oe2);
                 F1[0] = inf_min; F2[0] = inf_min;
             } else {
@@ -230,6 +237,7 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 const S ps = (S)jb.pre_ps[k];
                 if (prev_ok && p == prev_row) {
                     if (act) { /* LDS fast path */
+                        if (local_mode && j == 0) { if (ps > h) h = ps; }
                         if (j - 1 >= prev_beg && j - 1 <= prev_end) {
                             S v = (S)(cr[j - 1 - prev_beg] + ps);
                             if (v > h) h = v;
@@ -249,6 +257,7 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 const S *__restrict__ pE1 = pH + pbw;
                 const S *__restrict__ pE2 = pE1 + pbw;
                 if (act) {
+                    if (local_mode && j == 0) { if (ps > h) h = ps; }
                     if (j - 1 >= pm.beg && j - 1 <= pm.end) {
                         S v = (S)(pH[j - 1 - pm.beg] + ps);
                         if (v > h) h = v;
@@ -286,8 +295,10 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
             f2c = (S)__builtin_amdgcn_readlane((int)f2, WAVE - 1);
 
             S hf = smax(hpre, smax(f1, f2));
+            if (local_mode) hf = smax(hf, (S)0);
             S e1n = smax((S)(e1v - e1), (S)(hf - oe1));
             S e2n = smax((S)(e2v - e2), (S)(hf - oe2));
+            if (local_mode) { e1n = smax(e1n, (S)0); e2n = smax(e2n, (S)0); }
             if (act) {
                 H[j - beg] = hf; E1r[j - beg] = e1n; E2r[j - beg] = e2n;
                 F1r[j - beg] = f1; F2r[j - beg] = f2;
@@ -305,8 +316,9 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         else prev_ok = 0;
         buf_cur ^= 1;
 
-        /* row argmax reduce + adaptive band push (simd_abpoa_max_in_row / ada_max_i) */
-        if (jb.banded) {
+        /* row argmax reduce + adaptive band push (simd_abpoa_max_in_row /
+         * ada_max_i); local/extend also track the running best here */
+        if (jb.banded || local_mode || extend_mode) {
             int mv = (int)lmax;
             #pragma unroll
             for (int sft = 32; sft >= 1; sft >>= 1) {
@@ -320,28 +332,46 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 int lo = __shfl_xor(ll, sft); if (lo < ll) ll = lo;
                 int ro = __shfl_xor(rr, sft); if (ro > rr) rr = ro;
             }
-            for (int k = jb.out_off[r] + lane; k < jb.out_off[r + 1]; k += WAVE) {
-                int o = jb.out_idx[k];
-                if (rr + 1 > jb.max_right[o]) jb.max_right[o] = rr + 1;
-                if (ll + 1 < jb.max_left[o]) jb.max_left[o] = ll + 1;
+            if (local_mode) {
+                if (mv > run_best) { run_best = mv; run_best_i = r; run_best_j = ll; }
+            } else if (extend_mode) {
+                if (mv > run_best) {
+                    run_best = mv; run_best_i = r; run_best_j = rr;
+                    run_best_remain = jb.max_remain[r];
+                } else if (jb.zdrop > 0) {
+                    int delta = run_best_remain - jb.max_remain[r];
+                    int dd = delta - (rr - run_best_j); if (dd < 0) dd = -dd;
+                    if (run_best - mv > jb.zdrop + jb.e1 * dd) zdropped = 1;
+                }
             }
+            if (!zdropped && jb.banded) {
+                for (int k = jb.out_off[r] + lane; k < jb.out_off[r + 1]; k += WAVE) {
+                    int o = jb.out_idx[k];
+                    if (rr + 1 > jb.max_right[o]) jb.max_right[o] = rr + 1;
+                    if (ll + 1 < jb.max_left[o]) jb.max_left[o] = ll + 1;
+                }
+            }
+            if (zdropped) break;
         }
     }
 
     if (lane == 0) res->cells = used;
 
-    /* ---- final best over the end row's predecessors + backtrack (lane 0) ---- */
+    /* ---- final best + backtrack (lane 0) ---- */
     if (lane != 0) return;
 
-    int32_t best_score = jb.inf_min;
-    int best_i = 0, best_j = 0;
-    for (int k = jb.pre_off[n_rows - 1]; k < jb.pre_off[n_rows]; ++k) {
-        const int p = jb.pre_idx[k];
-        const abamd_row_meta_t pm = meta[p];
-        int e = pm.end < qlen ? pm.end : qlen;
-        const S *pH = arena + pm.off * 5;
-        int32_t sc = (e >= pm.beg) ? (int32_t)pH[e - pm.beg] : jb.inf_min;
-        if (sc > best_score) { best_score = sc; best_i = p; best_j = e; }
+    int32_t best_score = run_best;
+    int best_i = run_best_i, best_j = run_best_j;
+    if (jb.align_mode == 0) { /* global: max over the end row's predecessors */
+        best_score = jb.inf_min; best_i = 0; best_j = 0;
+        for (int k = jb.pre_off[n_rows - 1]; k < jb.pre_off[n_rows]; ++k) {
+            const int p = jb.pre_idx[k];
+            const abamd_row_meta_t pm = meta[p];
+            int e = pm.end < qlen ? pm.end : qlen;
+            const S *pH = arena + pm.off * 5;
+            int32_t sc = (e >= pm.beg) ? (int32_t)pH[e - pm.beg] : jb.inf_min;
+            if (sc > best_score) { best_score = sc; best_i = p; best_j = e; }
+        }
     }
     res->best_score = best_score;
     res->best_i = best_i; res->best_j = best_j;
@@ -370,6 +400,7 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
             const S F2j = (bj >= rb && bj <= re) ? F2r[bj - rb] : inf_min;
             const S F1jm1 = (bj - 1 >= rb && bj - 1 <= re) ? F1r[bj - 1 - rb] : inf_min;
             const S F2jm1 = (bj - 1 >= rb && bj - 1 <= re) ? F2r[bj - 1 - rb] : inf_min;
+            if (local_mode && Hj == 0) break;
             start_i = bi; start_j = bj;
             const int pq0 = jb.pre_off[bi], pq1 = jb.pre_off[bi + 1];
             const S s = (S)mat_lds[m * jb.row_base[bi] + query[bj - 1]];
@@ -546,6 +577,10 @@ void ag_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
     const int qlen = jb.qlen, n_rows = jb.n_rows, w = jb.w, m = jb.m;
     const S inf_min = (S)jb.inf_min;
     const S e1 = (S)jb.e1, oe1 = (S)jb.oe1;
+    const int local_mode = jb.align_mode == 1, extend_mode = jb.align_mode == 2;
+    int32_t run_best = jb.inf_min;
+    int run_best_i = 0, run_best_j = 0, run_best_remain = jb.max_remain[0];
+    int zdropped = 0;
     const int end_remain = jb.max_remain[n_rows - 1];
     S *arena = (S*)jb.arena;
     const uint8_t *__restrict__ query = jb.query;
@@ -582,7 +617,9 @@ void ag_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         const int fits = end0 + 1 <= BMAX;
         for (int j = lane; j <= end0; j += WAVE) {
             S hv, e1v2;
-            if (j == 0) {
+            if (local_mode) {
+                hv = 0; e1v2 = 0; F1[j] = 0;
+            } else if (j == 0) {
                 hv = 0; e1v2 = (S)(0 - oe1);
                 F1[0] = inf_min;
             } else {
@@ -641,6 +678,7 @@ void ag_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 const S ps = (S)jb.pre_ps[k];
                 if (prev_ok && p == prev_row) {
                     if (act) {
+                        if (local_mode && j == 0) { if (ps > h) h = ps; }
                         if (j - 1 >= prev_beg && j - 1 <= prev_end) {
                             S v = (S)(cr[j - 1 - prev_beg] + ps);
                             if (v > h) h = v;
@@ -657,6 +695,7 @@ void ag_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 const S *__restrict__ pH = arena + pm.off * 3;
                 const S *__restrict__ pE1 = pH + pbw;
                 if (act) {
+                    if (local_mode && j == 0) { if (ps > h) h = ps; }
                     if (j - 1 >= pm.beg && j - 1 <= pm.end) {
                         S v = (S)(pH[j - 1 - pm.beg] + ps);
                         if (v > h) h = v;
@@ -683,7 +722,9 @@ void ag_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
 
             S tmp = smax(hm, e1v);
             S hf = smax(tmp, f1);
-            S e1n = (hf == tmp) ? smax((S)(e1v - e1), (S)(hf - oe1)) : inf_min;
+            if (local_mode) hf = smax(hf, (S)0);
+            S e1n = (hf == tmp) ? smax((S)(e1v - e1), (S)(hf - oe1))
+                                : (local_mode ? (S)0 : inf_min);
             if (act) {
                 H[j - beg] = hf; E1r[j - beg] = e1n; F1r[j - beg] = f1;
                 if (cache_fits) { cw[j - beg] = hf; cw[BMAX + j - beg] = e1n; }
@@ -696,7 +737,7 @@ void ag_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         else prev_ok = 0;
         buf_cur ^= 1;
 
-        if (jb.banded) {
+        if (jb.banded || local_mode || extend_mode) {
             int mv = (int)lmax;
             #pragma unroll
             for (int sft = 32; sft >= 1; sft >>= 1) {
@@ -710,26 +751,44 @@ void ag_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 int lo = __shfl_xor(ll, sft); if (lo < ll) ll = lo;
                 int ro = __shfl_xor(rr, sft); if (ro > rr) rr = ro;
             }
-            for (int k = jb.out_off[r] + lane; k < jb.out_off[r + 1]; k += WAVE) {
-                int o = jb.out_idx[k];
-                if (rr + 1 > jb.max_right[o]) jb.max_right[o] = rr + 1;
-                if (ll + 1 < jb.max_left[o]) jb.max_left[o] = ll + 1;
+            if (local_mode) {
+                if (mv > run_best) { run_best = mv; run_best_i = r; run_best_j = ll; }
+            } else if (extend_mode) {
+                if (mv > run_best) {
+                    run_best = mv; run_best_i = r; run_best_j = rr;
+                    run_best_remain = jb.max_remain[r];
+                } else if (jb.zdrop > 0) {
+                    int delta = run_best_remain - jb.max_remain[r];
+                    int dd = delta - (rr - run_best_j); if (dd < 0) dd = -dd;
+                    if (run_best - mv > jb.zdrop + jb.e1 * dd) zdropped = 1;
+                }
             }
+            if (!zdropped && jb.banded) {
+                for (int k = jb.out_off[r] + lane; k < jb.out_off[r + 1]; k += WAVE) {
+                    int o = jb.out_idx[k];
+                    if (rr + 1 > jb.max_right[o]) jb.max_right[o] = rr + 1;
+                    if (ll + 1 < jb.max_left[o]) jb.max_left[o] = ll + 1;
+                }
+            }
+            if (zdropped) break;
         }
     }
 
     if (lane == 0) res->cells = used;
     if (lane != 0) return;
 
-    int32_t best_score = jb.inf_min;
-    int best_i = 0, best_j = 0;
-    for (int k = jb.pre_off[n_rows - 1]; k < jb.pre_off[n_rows]; ++k) {
-        const int p = jb.pre_idx[k];
-        const abamd_row_meta_t pm = meta[p];
-        int e = pm.end < qlen ? pm.end : qlen;
-        const S *pH = arena + pm.off * 3;
-        int32_t sc = (e >= pm.beg) ? (int32_t)pH[e - pm.beg] : jb.inf_min;
-        if (sc > best_score) { best_score = sc; best_i = p; best_j = e; }
+    int32_t best_score = run_best;
+    int best_i = run_best_i, best_j = run_best_j;
+    if (jb.align_mode == 0) {
+        best_score = jb.inf_min; best_i = 0; best_j = 0;
+        for (int k = jb.pre_off[n_rows - 1]; k < jb.pre_off[n_rows]; ++k) {
+            const int p = jb.pre_idx[k];
+            const abamd_row_meta_t pm = meta[p];
+            int e = pm.end < qlen ? pm.end : qlen;
+            const S *pH = arena + pm.off * 3;
+            int32_t sc = (e >= pm.beg) ? (int32_t)pH[e - pm.beg] : jb.inf_min;
+            if (sc > best_score) { best_score = sc; best_i = p; best_j = e; }
+        }
     }
     res->best_score = best_score;
     res->best_i = best_i; res->best_j = best_j;
@@ -754,6 +813,7 @@ void ag_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
             const S E1j = (bj >= rb && bj <= re) ? E1r[bj - rb] : inf_min;
             const S F1j = (bj >= rb && bj <= re) ? F1r[bj - rb] : inf_min;
             const S F1jm1 = (bj - 1 >= rb && bj - 1 <= re) ? F1r[bj - 1 - rb] : inf_min;
+            if (local_mode && Hj == 0) break;
             start_i = bi; start_j = bj;
             const int pq0 = jb.pre_off[bi], pq1 = jb.pre_off[bi + 1];
             const S s = (S)mat_lds[m * jb.row_base[bi] + query[bj - 1]];
@@ -884,6 +944,10 @@ void lg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
     const int qlen = jb.qlen, n_rows = jb.n_rows, w = jb.w, m = jb.m;
     const S inf_min = (S)jb.inf_min;
     const S e1 = (S)jb.e1;
+    const int local_mode = jb.align_mode == 1, extend_mode = jb.align_mode == 2;
+    int32_t run_best = jb.inf_min;
+    int run_best_i = 0, run_best_j = 0, run_best_remain = jb.max_remain[0];
+    int zdropped = 0;
     const int end_remain = jb.max_remain[n_rows - 1];
     S *arena = (S*)jb.arena;
     const uint8_t *__restrict__ query = jb.query;
@@ -918,7 +982,7 @@ void lg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         S *c = &prev_lds[wid][buf_cur][0];
         const int fits = end0 + 1 <= BMAX;
         for (int j = lane; j <= end0; j += WAVE) {
-            S hv = (S)(-jb.e1 * j);
+            S hv = local_mode ? (S)0 : (S)(-jb.e1 * j);
             H[j] = hv;
             if (fits) c[j] = hv;
         }
@@ -971,6 +1035,7 @@ void lg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 const S ps = (S)jb.pre_ps[k];
                 if (prev_ok && p == prev_row) {
                     if (act) {
+                        if (local_mode && j == 0) { S v = (S)(ps + q); if (v > h) h = v; }
                         if (j - 1 >= prev_beg && j - 1 <= prev_end) {
                             S v = (S)(cr[j - 1 - prev_beg] + ps + q);
                             if (v > h) h = v;
@@ -985,6 +1050,7 @@ void lg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 const abamd_row_meta_t pm = meta[p];
                 const S *__restrict__ pH = arena + pm.off;
                 if (act) {
+                    if (local_mode && j == 0) { S v = (S)(ps + q); if (v > h) h = v; }
                     if (j - 1 >= pm.beg && j - 1 <= pm.end) {
                         S v = (S)(pH[j - 1 - pm.beg] + ps + q);
                         if (v > h) h = v;
@@ -1003,6 +1069,7 @@ void lg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
             } else h = smax(h, (S)(hs - e1));
             h = scan_maxplus(h, jb.e1, inf_min, lane);
             carry_h = (S)__builtin_amdgcn_readlane((int)h, WAVE - 1);
+            if (local_mode) h = smax(h, (S)0); /* clamp AFTER the scan+carry */
             if (act) {
                 H[j - beg] = h;
                 if (cache_fits) cw[j - beg] = h;
@@ -1015,7 +1082,7 @@ void lg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         else prev_ok = 0;
         buf_cur ^= 1;
 
-        if (jb.banded) {
+        if (jb.banded || local_mode || extend_mode) {
             int mv = (int)lmax;
             #pragma unroll
             for (int sft = 32; sft >= 1; sft >>= 1) {
@@ -1029,26 +1096,44 @@ void lg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 int lo = __shfl_xor(ll, sft); if (lo < ll) ll = lo;
                 int ro = __shfl_xor(rr, sft); if (ro > rr) rr = ro;
             }
-            for (int k = jb.out_off[r] + lane; k < jb.out_off[r + 1]; k += WAVE) {
-                int o = jb.out_idx[k];
-                if (rr + 1 > jb.max_right[o]) jb.max_right[o] = rr + 1;
-                if (ll + 1 < jb.max_left[o]) jb.max_left[o] = ll + 1;
+            if (local_mode) {
+                if (mv > run_best) { run_best = mv; run_best_i = r; run_best_j = ll; }
+            } else if (extend_mode) {
+                if (mv > run_best) {
+                    run_best = mv; run_best_i = r; run_best_j = rr;
+                    run_best_remain = jb.max_remain[r];
+                } else if (jb.zdrop > 0) {
+                    int delta = run_best_remain - jb.max_remain[r];
+                    int dd = delta - (rr - run_best_j); if (dd < 0) dd = -dd;
+                    if (run_best - mv > jb.zdrop + jb.e1 * dd) zdropped = 1;
+                }
             }
+            if (!zdropped && jb.banded) {
+                for (int k = jb.out_off[r] + lane; k < jb.out_off[r + 1]; k += WAVE) {
+                    int o = jb.out_idx[k];
+                    if (rr + 1 > jb.max_right[o]) jb.max_right[o] = rr + 1;
+                    if (ll + 1 < jb.max_left[o]) jb.max_left[o] = ll + 1;
+                }
+            }
+            if (zdropped) break;
         }
     }
 
     if (lane == 0) res->cells = used;
     if (lane != 0) return;
 
-    int32_t best_score = jb.inf_min;
-    int best_i = 0, best_j = 0;
-    for (int k = jb.pre_off[n_rows - 1]; k < jb.pre_off[n_rows]; ++k) {
-        const int p = jb.pre_idx[k];
-        const abamd_row_meta_t pm = meta[p];
-        int e = pm.end < qlen ? pm.end : qlen;
-        const S *pH = arena + pm.off;
-        int32_t sc = (e >= pm.beg) ? (int32_t)pH[e - pm.beg] : jb.inf_min;
-        if (sc > best_score) { best_score = sc; best_i = p; best_j = e; }
+    int32_t best_score = run_best;
+    int best_i = run_best_i, best_j = run_best_j;
+    if (jb.align_mode == 0) {
+        best_score = jb.inf_min; best_i = 0; best_j = 0;
+        for (int k = jb.pre_off[n_rows - 1]; k < jb.pre_off[n_rows]; ++k) {
+            const int p = jb.pre_idx[k];
+            const abamd_row_meta_t pm = meta[p];
+            int e = pm.end < qlen ? pm.end : qlen;
+            const S *pH = arena + pm.off;
+            int32_t sc = (e >= pm.beg) ? (int32_t)pH[e - pm.beg] : jb.inf_min;
+            if (sc > best_score) { best_score = sc; best_i = p; best_j = e; }
+        }
     }
     res->best_score = best_score;
     res->best_i = best_i; res->best_j = best_j;
@@ -1068,6 +1153,7 @@ void lg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
             const S *H = arena + bm.off;
             const S Hj = (bj >= rb && bj <= re) ? H[bj - rb] : inf_min;
             const S Hjm1 = (bj - 1 >= rb && bj - 1 <= re) ? H[bj - 1 - rb] : inf_min;
+            if (local_mode && Hj == 0) break;
             start_i = bi; start_j = bj;
             const int pq0 = jb.pre_off[bi], pq1 = jb.pre_off[bi + 1];
             const S s = (S)mat_lds[m * jb.row_base[bi] + query[bj - 1]];

@@ -108,3 +108,18 @@ def test_gpu_aa_mode_vs_oracle(gpu_bin, cputest_bin, oracle_env, tmp_path):
         gpu = run_stdout([gpu_bin, str(fa)] + opts)
         cpu = run_stdout([cputest_bin, str(fa)] + opts, env=oracle_env)
         assert gpu == cpu
+
+
+@pytest.mark.parametrize("opts", [["-m1"], ["-m2"], ["-m1", "-O", "4", "-E", "2"],
+                                  ["-m2", "-O", "0", "-E", "2"], ["-m2", "-z", "100"]],
+                         ids=["local", "extend", "local-affine", "extend-linear", "extend-zdrop"])
+def test_gpu_align_modes_vs_oracle(gpu_bin, cputest_bin, oracle_env, tmp_path, opts):
+    """Local and extension alignment GPU kernels vs the CPU oracle."""
+    fa = tmp_path / "s.fa"
+    subprocess.run(["python3", os.path.join(ROOT, "tests", "make_synth.py"), str(fa),
+                    "--seed", "31", "--len", "1200", "--depth", "25"],
+                   check=True, stderr=subprocess.DEVNULL)
+    for extra in ([], ["-r1"]):
+        gpu = run_stdout([gpu_bin, str(fa)] + opts + extra)
+        cpu = run_stdout([cputest_bin, str(fa)] + opts + extra, env=oracle_env)
+        assert gpu == cpu, "GPU/oracle divergence opts=%r extra=%r" % (opts, extra)
