@@ -24,6 +24,8 @@ typedef struct {
     uint8_t *query;
     int qlen;
     abpoa_res_t *res;
+    int64_t est_cells_hint;
+    int64_t *cells_out;
 } abamd_batch_job_t;
 int abamd_gpu_align_batch(abamd_batch_job_t *batch, int n_jobs);
 int abamd_gpu_batch_prepare(abamd_batch_job_t *batch, int n_jobs, int slot);
@@ -38,6 +40,7 @@ typedef struct {
     abpoa_res_t res;
     int *weight_buf; int weight_cap;
     int active; /* has a job this round */
+    int64_t last_cells; /* banded cells of this set's previous alignment */
 } set_state_t;
 
 typedef struct {
@@ -168,7 +171,8 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
     double slot_est[2] = {0, 0};
     int slot_big[2] = {0, 0};
 
-    /* arena bytes one job will demand (matches gpu_align.cpp's reservation) */
+    /* arena bytes one job will demand (matches gpu_align.cpp's reservation,
+     * including the measured-cells tightening) */
     #define JOB_EST(J) ({ \
         int _w = abpt->wb < 0 ? (J)->qlen : abpt->wb + (int)(abpt->wf * (J)->qlen); \
         int _gn = (J)->ab->abg->node_n, _q = (J)->qlen; \
@@ -178,7 +182,12 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
         int _oe1 = abpt->gap_open1 + abpt->gap_ext1, _oe2 = abpt->gap_open2 + abpt->gap_ext2; \
         int _ssz = (_ms <= 32767 - abpt->min_mis - _oe1 - _oe2) ? 2 : 4; \
         int _pl = abpt->gap_mode == ABPOA_CONVEX_GAP ? 5 : abpt->gap_mode == ABPOA_AFFINE_GAP ? 3 : 1; \
-        (double)_gn * (2.0 * _w + 160.0) * _pl * _ssz; })
+        double _cells = (double)_gn * (2.0 * _w + 160.0); \
+        if ((J)->est_cells_hint > 0) { \
+            double _t = (double)(J)->est_cells_hint * 1.25 + _q; \
+            if (_t < _cells) _cells = _t; \
+        } \
+        _cells * _pl * _ssz; })
 
     /* build the job list for one (round, group) item into a slot */
     #define BUILD_ITEM(it, slot) do { \
@@ -196,6 +205,8 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
             J->query = (uint8_t*)sets[i].seqs[_r]; \
             J->qlen = sets[i].seq_lens[_r]; \
             J->res = &sets[i].res; \
+            J->est_cells_hint = sets[i].last_cells; \
+            J->cells_out = &sets[i].last_cells; \
             _est += JOB_EST(J); \
             ++_nj; \
         } \

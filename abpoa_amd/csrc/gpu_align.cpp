@@ -306,6 +306,9 @@ struct BatchJob {
     uint8_t *query;
     int qlen;
     abpoa_res_t *res;
+    int64_t est_cells_hint;  /* measured cells of this set's previous round
+                                (0 = none): tightens the arena reservation */
+    int64_t *cells_out;      /* optional: actual banded cells written back */
 };
 
 /* score-width pick (abpoa_align_simd.c:1284-1302); assumes uniform paras */
@@ -397,6 +400,11 @@ extern "C" int abamd_gpu_batch_prepare(BatchJob *batch, int n_jobs, int slot) {
                 C.jb_bufs[i].reset();
                 arena_est[i] = pack_job(C.jb_bufs[i], packs[i], B.ab, B.abpt,
                                         B.beg_node_id, B.end_node_id, B.query, B.qlen);
+                if (B.est_cells_hint > 0) {
+                    /* bands drift a few % per round; 25% headroom + overflow retry */
+                    int64_t tight = B.est_cells_hint + B.est_cells_hint / 4 + B.qlen;
+                    if (tight < arena_est[i]) arena_est[i] = tight;
+                }
                 int inf_min;
                 int span = B.ab->abg->node_id_to_index[B.end_node_id] - B.ab->abg->node_id_to_index[B.beg_node_id] + 1;
                 pick_width(B.abpt, B.qlen, span, &bits_v[i], &inf_min);
@@ -615,6 +623,7 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
         }
         g_dp_cells += (uint64_t)R.cells;
         g_alg_bytes += (uint64_t)R.cells * planes * ssz;
+        if (batch[i].cells_out) *batch[i].cells_out = R.cells;
         abpoa_res_t *res = batch[i].res;
         res->best_score = R.best_score;
         if (batch[i].abpt->ret_cigar && R.n_cigar > 0) {
@@ -644,5 +653,6 @@ extern "C" int abamd_gpu_align_sequence_to_subgraph(abpoa_t *ab, abpoa_para_t *a
     b.ab = ab; b.abpt = abpt;
     b.beg_node_id = beg_node_id; b.end_node_id = end_node_id;
     b.query = query; b.qlen = qlen; b.res = res;
+    b.est_cells_hint = 0; b.cells_out = nullptr;
     return abamd_gpu_align_batch(&b, 1);
 }

@@ -23,6 +23,7 @@ typedef struct {
     abpoa_t *ab; abpoa_para_t *abpt;
     int beg_node_id, end_node_id;
     uint8_t *query; int qlen; abpoa_res_t *res;
+    int64_t est_cells_hint; int64_t *cells_out;
 } abamd_batch_job_t;
 
 int abamd_gpu_align_batch(abamd_batch_job_t *batch, int n_jobs) {
