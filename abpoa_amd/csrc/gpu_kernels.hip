@@ -228,101 +228,6 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         S carry_h = inf_min, f1c = inf_min, f2c = inf_min;
         S lmax = inf_min; int lleft = -1, lright = -1;
 
-        /* Two-phase row: the M/E gathers of every chunk are independent, so
-         * phase 1 keeps all their LDS/HBM loads in flight at once; the only
-         * serial dependency — the F carry — runs in phase 2 as pure
-         * VALU/DPP. Rows wider than BMAX take the fused fallback loop. */
-        const int nch = (int)((bw + WAVE - 1) / WAVE);
-        if (nch <= BMAX / WAVE) {
-            S hpreA[BMAX / WAVE], e1A[BMAX / WAVE], e2A[BMAX / WAVE];
-            for (int c = 0; c < nch; ++c) {
-                const int j = beg + c * WAVE + lane;
-                const bool act = j <= end;
-                S h = inf_min, e1v = inf_min, e2v = inf_min;
-                for (int k = pk0; k < pk1; ++k) {
-                    const int p = jb.pre_idx[k];
-                    const S ps = (S)jb.pre_ps[k];
-                    if (prev_ok && p == prev_row) {
-                        if (act) {
-                            if (local_mode && j == 0) { if (ps > h) h = ps; }
-                            if (j - 1 >= prev_beg && j - 1 <= prev_end) {
-                                S v = (S)(cr[j - 1 - prev_beg] + ps);
-                                if (v > h) h = v;
-                            }
-                            if (j >= prev_beg && j <= prev_end) {
-                                S v1 = (S)(cr[BMAX + j - prev_beg] + ps);
-                                S v2 = (S)(cr[2 * BMAX + j - prev_beg] + ps);
-                                if (v1 > e1v) e1v = v1;
-                                if (v2 > e2v) e2v = v2;
-                            }
-                        }
-                        continue;
-                    }
-                    const abamd_row_meta_t pm = meta[p];
-                    const int64_t pbw = pm.end - pm.beg + 1;
-                    const S *__restrict__ pH = arena + pm.off * 5;
-                    const S *__restrict__ pE1 = pH + pbw;
-                    const S *__restrict__ pE2 = pE1 + pbw;
-                    if (act) {
-                        if (local_mode && j == 0) { if (ps > h) h = ps; }
-                        if (j - 1 >= pm.beg && j - 1 <= pm.end) {
-                            S v = (S)(pH[j - 1 - pm.beg] + ps);
-                            if (v > h) h = v;
-                        }
-                        if (j >= pm.beg && j <= pm.end) {
-                            S v1 = (S)(pE1[j - pm.beg] + ps);
-                            S v2 = (S)(pE2[j - pm.beg] + ps);
-                            if (v1 > e1v) e1v = v1;
-                            if (v2 > e2v) e2v = v2;
-                        }
-                    }
-                }
-                const S q = (S)((j == 0 || !act) ? 0 : mrow[query[j - 1]]);
-                S hpre = (S)(h + q);
-                hpre = smax(hpre, smax(e1v, e2v));
-                if (!act) hpre = inf_min;
-                hpreA[c] = hpre; e1A[c] = e1v; e2A[c] = e2v;
-            }
-            for (int c = 0; c < nch; ++c) {
-                const int cs = beg + c * WAVE;
-                const int j = cs + lane;
-                const bool act = j <= end;
-                const S hpre = hpreA[c];
-                S hshift = (S)__shfl_up((int)hpre, 1);
-                S c1, c2;
-                if (lane == 0) {
-                    if (cs == beg) { c1 = (S)(inf_min - oe1); c2 = (S)(inf_min - oe2); }
-                    else {
-                        c1 = smax((S)(carry_h - oe1), (S)(f1c - e1));
-                        c2 = smax((S)(carry_h - oe2), (S)(f2c - e2));
-                    }
-                } else {
-                    c1 = (S)(hshift - oe1);
-                    c2 = (S)(hshift - oe2);
-                }
-                S f1 = scan_maxplus(c1, jb.e1, inf_min, lane);
-                S f2 = scan_maxplus(c2, jb.e2, inf_min, lane);
-                carry_h = (S)__builtin_amdgcn_readlane((int)hpre, WAVE - 1);
-                f1c = (S)__builtin_amdgcn_readlane((int)f1, WAVE - 1);
-                f2c = (S)__builtin_amdgcn_readlane((int)f2, WAVE - 1);
-                S hf = smax(hpre, smax(f1, f2));
-                if (local_mode) hf = smax(hf, (S)0);
-                S e1n = smax((S)(e1A[c] - e1), (S)(hf - oe1));
-                S e2n = smax((S)(e2A[c] - e2), (S)(hf - oe2));
-                if (local_mode) { e1n = smax(e1n, (S)0); e2n = smax(e2n, (S)0); }
-                if (act) {
-                    H[j - beg] = hf; E1r[j - beg] = e1n; E2r[j - beg] = e2n;
-                    F1r[j - beg] = f1; F2r[j - beg] = f2;
-                    if (cache_fits) {
-                        cw[j - beg] = hf;
-                        cw[BMAX + j - beg] = e1n;
-                        cw[2 * BMAX + j - beg] = e2n;
-                    }
-                    if (hf > lmax) { lmax = hf; lleft = j; lright = j; }
-                    else if (hf == lmax) { lright = j; }
-                }
-            }
-        } else
         for (int cs = beg; cs <= end; cs += WAVE) {
             const int j = cs + lane;
             const bool act = j <= end;
