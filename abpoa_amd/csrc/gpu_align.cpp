@@ -400,7 +400,7 @@ extern "C" int abamd_gpu_batch_prepare(BatchJob *batch, int n_jobs, int slot) {
                 C.jb_bufs[i].reset();
                 arena_est[i] = pack_job(C.jb_bufs[i], packs[i], B.ab, B.abpt,
                                         B.beg_node_id, B.end_node_id, B.query, B.qlen);
-                if (B.est_cells_hint > 0) {
+                if (B.est_cells_hint > 0 && !getenv("ABPOA_AMD_NO_HINT")) {
                     /* bands drift a few % per round; 25% headroom + overflow retry */
                     int64_t tight = B.est_cells_hint + B.est_cells_hint / 4 + B.qlen;
                     if (tight < arena_est[i]) arena_est[i] = tight;
