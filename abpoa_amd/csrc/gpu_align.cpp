@@ -107,8 +107,8 @@ struct GpuCtx {
     DevBuf slab;        /* all per-job inputs + metadata */
     DevBuf arena;       /* DP plane arena */
     DevBuf jobs, results, cigars;
-    PinnedBuf stage;    /* pinned H2D staging for the whole batch */
-    std::vector<HostBuf> jb_bufs;   /* per-job pack buffers (parallel pack) */
+    PinnedBuf stage2[2];              /* per-slot pinned H2D staging */
+    std::vector<HostBuf> jb_bufs2[2]; /* per-slot per-job pack buffers */
     HostBuf hb;
     std::vector<abamd_gpu_job_t> hjobs;
     std::vector<abamd_gpu_res_t> hres;
@@ -347,6 +347,7 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot);
 namespace {
 struct PendingBatch {
     BatchJob *batch = nullptr;
+    int slot = 0;
     int n_jobs = 0;
     std::vector<JobPack> packs;
     std::vector<int64_t> arena_est;
@@ -371,7 +372,15 @@ extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
     return abamd_gpu_batch_finish_slot(0);
 }
 
+static int prepare_internal(BatchJob *batch, int n_jobs, int slot, const int64_t *min_est);
+
 extern "C" int abamd_gpu_batch_prepare(BatchJob *batch, int n_jobs, int slot) {
+    return prepare_internal(batch, n_jobs, slot, nullptr);
+}
+
+/* min_est: per-job arena floor (cells) used by the overflow retry, which must
+ * REPACK: the shared pinned stage may already hold the other slot's batch */
+static int prepare_internal(BatchJob *batch, int n_jobs, int slot, const int64_t *min_est) {
     if (n_jobs <= 0) { g_slots[slot].active = false; g_slots[slot].n_jobs = 0; return 0; }
     GpuCtx &C = g_ctx;
     C.ensure_init();
@@ -380,7 +389,8 @@ extern "C" int abamd_gpu_batch_prepare(BatchJob *batch, int n_jobs, int slot) {
                      : abpt->gap_mode == ABPOA_AFFINE_GAP ? 3 : 1;
 
     uint64_t t_pack0 = now_ns();
-    if ((int)C.jb_bufs.size() < n_jobs) C.jb_bufs.resize(n_jobs);
+    std::vector<HostBuf> &jbufs = C.jb_bufs2[slot];
+    if ((int)jbufs.size() < n_jobs) jbufs.resize(n_jobs);
     std::vector<JobPack> packs(n_jobs);
     std::vector<int64_t> arena_est(n_jobs);
     std::vector<int> bits_v(n_jobs, 16);
@@ -397,14 +407,15 @@ extern "C" int abamd_gpu_batch_prepare(BatchJob *batch, int n_jobs, int slot) {
                 int i = next.fetch_add(1);
                 if (i >= n_jobs) break;
                 BatchJob &B = batch[i];
-                C.jb_bufs[i].reset();
-                arena_est[i] = pack_job(C.jb_bufs[i], packs[i], B.ab, B.abpt,
+                jbufs[i].reset();
+                arena_est[i] = pack_job(jbufs[i], packs[i], B.ab, B.abpt,
                                         B.beg_node_id, B.end_node_id, B.query, B.qlen);
                 if (B.est_cells_hint > 0 && !getenv("ABPOA_AMD_NO_HINT")) {
                     /* bands drift a few % per round; 25% headroom + overflow retry */
                     int64_t tight = B.est_cells_hint + B.est_cells_hint / 4 + B.qlen;
                     if (tight < arena_est[i]) arena_est[i] = tight;
                 }
+                if (min_est && min_est[i] > arena_est[i]) arena_est[i] = min_est[i];
                 int inf_min;
                 int span = B.ab->abg->node_id_to_index[B.end_node_id] - B.ab->abg->node_id_to_index[B.beg_node_id] + 1;
                 pick_width(B.abpt, B.qlen, span, &bits_v[i], &inf_min);
@@ -428,10 +439,11 @@ extern "C" int abamd_gpu_batch_prepare(BatchJob *batch, int n_jobs, int slot) {
     size_t total = (mat_bytes + 255) & ~(size_t)255;
     for (int i = 0; i < n_jobs; ++i) {
         packs[i].slab_base = total;
-        total += (C.jb_bufs[i].used + 255) & ~(size_t)255;
+        total += (jbufs[i].used + 255) & ~(size_t)255;
     }
-    C.stage.ensure(total);
-    memcpy(C.stage.p + o_mat, abpt->mat, mat_bytes);
+    PinnedBuf &stage = C.stage2[slot];
+    stage.ensure(total);
+    memcpy(stage.p + o_mat, abpt->mat, mat_bytes);
     {
         int nthr = (int)std::thread::hardware_concurrency();
         if (nthr < 1) nthr = 1;
@@ -441,7 +453,7 @@ extern "C" int abamd_gpu_batch_prepare(BatchJob *batch, int n_jobs, int slot) {
             for (;;) {
                 int i = next.fetch_add(1);
                 if (i >= n_jobs) break;
-                memcpy(C.stage.p + packs[i].slab_base, C.jb_bufs[i].v.data(), C.jb_bufs[i].used);
+                memcpy(stage.p + packs[i].slab_base, jbufs[i].v.data(), jbufs[i].used);
             }
         };
         std::vector<std::thread> ts;
@@ -469,6 +481,7 @@ extern "C" int abamd_gpu_batch_prepare(BatchJob *batch, int n_jobs, int slot) {
     }
     PendingBatch &PB = g_slots[slot];
     PB.batch = batch;
+    PB.slot = slot;
     PB.n_jobs = n_jobs;
     PB.packs = std::move(packs);
     PB.arena_est = std::move(arena_est);
@@ -525,7 +538,7 @@ static int batch_launch(GpuCtx &C, PendingBatch &PB) {
 
         /* upload slab + jobs */
         C.slab.ensure(total);
-        HIP_CHECK(hipMemcpyAsync(C.slab.p, C.stage.p, total, hipMemcpyHostToDevice, C.stream));
+        HIP_CHECK(hipMemcpyAsync(C.slab.p, C.stage2[PB.slot].p, total, hipMemcpyHostToDevice, C.stream));
         uint8_t *S0 = (uint8_t*)C.slab.p;
         for (int i = 0; i < n_jobs; ++i) {
             JobPack &P = packs[i];
@@ -592,17 +605,22 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
         g_kernel_ns += (uint64_t)(ms * 1e6);
         g_launches += 1;
     }
-    /* rare: a job's adaptive band outgrew its reservation; relaunch the whole
-     * pending batch synchronously with doubled reservations */
+    /* rare: a job's adaptive band outgrew its reservation. The pinned stage
+     * (and the per-job pack buffers) may already hold the OTHER slot's batch,
+     * so the retry must fully REPACK this batch before relaunching. */
     for (int attempt = 0;; ++attempt) {
         bool overflow = false;
-        for (int i = 0; i < n_jobs; ++i)
-            if (PB.hres[i].status == ABAMD_JOB_ARENA_OVERFLOW) { PB.arena_est[i] *= 2; overflow = true; }
+        std::vector<int64_t> floor_est(n_jobs, 0);
+        for (int i = 0; i < n_jobs; ++i) {
+            floor_est[i] = PB.arena_est[i];
+            if (PB.hres[i].status == ABAMD_JOB_ARENA_OVERFLOW) { floor_est[i] *= 2; overflow = true; }
+        }
         if (!overflow) break;
         if (attempt > 8) {
             fprintf(stderr, "[abpoa_amd] arena overflow persists after %d retries\n", attempt);
             exit(EXIT_FAILURE);
         }
+        prepare_internal(batch, n_jobs, slot, floor_est.data());
         batch_launch(C, PB);
         HIP_CHECK(hipStreamSynchronize(C.stream));
         HIP_CHECK(hipMemcpy(PB.hres.data(), C.results.p, (size_t)n_jobs * sizeof(abamd_gpu_res_t), hipMemcpyDeviceToHost));
