@@ -28,6 +28,7 @@ typedef struct {
     int64_t *cells_out;
 } abamd_batch_job_t;
 int abamd_gpu_align_batch(abamd_batch_job_t *batch, int n_jobs);
+int64_t abamd_gpu_free_mem(void);
 int abamd_gpu_batch_prepare(abamd_batch_job_t *batch, int n_jobs, int slot);
 int abamd_gpu_batch_launch(int slot);
 int abamd_gpu_batch_finish_slot(int slot);
@@ -125,11 +126,16 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
     int *job_set = (int*)abamd_malloc((size_t)n_sets * sizeof(int));
     pthread_t *tids = (pthread_t*)abamd_malloc((size_t)n_host_threads * sizeof(pthread_t));
 
-    /* GPU memory budget per launch (arena is the dominant term) */
-    double mem_gb = 48.0;
+    /* GPU memory budget per launch (arena is the dominant term); default
+     * mirrors the shim's 70%-of-free-HBM reservation */
+    double mem_gb = 0.0;
     {
         const char *s = getenv("ABPOA_AMD_MEM_GB");
         if (s && *s) mem_gb = atof(s);
+    }
+    if (mem_gb <= 0.0) {
+        int64_t free_b = abamd_gpu_free_mem();
+        mem_gb = free_b > 0 ? (double)free_b * 0.70 / 1e9 : 48.0;
     }
     const double budget_bytes = mem_gb * 1e9;
 

@@ -128,10 +128,16 @@ struct GpuCtx {
         HIP_CHECK(hipEventCreate(&ev1));
         HIP_CHECK(hipEventCreate(&ev_h2d));
         /* reserve the plane arena at the full batch memory budget once:
-         * re-growing a 100+ GB hipMalloc every round costs seconds */
-        double mem_gb = 48.0;
+         * re-growing a 100+ GB hipMalloc every round costs seconds.
+         * Default budget: 70% of free HBM (ABPOA_AMD_MEM_GB overrides). */
+        double mem_gb;
         const char *s = getenv("ABPOA_AMD_MEM_GB");
         if (s && *s) mem_gb = atof(s);
+        else {
+            size_t free_b = 0, total_b = 0;
+            HIP_CHECK(hipMemGetInfo(&free_b, &total_b));
+            mem_gb = (double)free_b * 0.70 / 1e9;
+        }
         arena.ensure((size_t)(mem_gb * 1e9));
         init = true;
     }
@@ -663,6 +669,12 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
     g_unpack_ns += now_ns() - t_unpack0;
     PB.active = false;
     return 0;
+}
+
+extern "C" int64_t abamd_gpu_free_mem(void) {
+    size_t free_b = 0, total_b = 0;
+    if (hipMemGetInfo(&free_b, &total_b) != hipSuccess) return 0;
+    return (int64_t)free_b;
 }
 
 extern "C" int abamd_gpu_align_sequence_to_subgraph(abpoa_t *ab, abpoa_para_t *abpt,

@@ -48,3 +48,5 @@ int abamd_gpu_batch_launch(int slot) {
     return abamd_gpu_align_batch(stub_slot_batch[slot], stub_slot_n[slot]);
 }
 int abamd_gpu_batch_finish_slot(int slot) { (void)slot; return 0; }
+
+int64_t abamd_gpu_free_mem(void) { return 0; }

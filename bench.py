@@ -128,9 +128,9 @@ def cpu_baseline_leg(depth, qlen):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=2)
+    ap.add_argument("--steps", type=int, default=1)
     ap.add_argument("--warmup", type=int, default=1)
-    ap.add_argument("--sets-per-step", type=int, default=48)
+    ap.add_argument("--sets-per-step", type=int, default=1000)
     ap.add_argument("--depth", type=int, default=50)
     ap.add_argument("--qlen", type=int, default=10000)
     ap.add_argument("--threads", type=int, default=max(2, (os.cpu_count() or 8) - 2))
