@@ -107,8 +107,8 @@ struct GpuCtx {
     DevBuf slab;        /* all per-job inputs + metadata */
     DevBuf arena;       /* DP plane arena */
     DevBuf jobs, results, cigars;
-    PinnedBuf stage2[2];              /* per-slot pinned H2D staging */
-    std::vector<HostBuf> jb_bufs2[2]; /* per-slot per-job pack buffers */
+    PinnedBuf stage2[3];              /* per-slot pinned H2D staging (slot 2 = retry) */
+    std::vector<HostBuf> jb_bufs2[3]; /* per-slot per-job pack buffers */
     HostBuf hb;
     std::vector<abamd_gpu_job_t> hjobs;
     std::vector<abamd_gpu_res_t> hres;
@@ -365,7 +365,7 @@ struct PendingBatch {
     size_t total = 0;
     bool active = false;
 };
-thread_local PendingBatch g_slots[2];
+thread_local PendingBatch g_slots[3];
 }
 
 static int batch_launch(GpuCtx &C, PendingBatch &P); /* fwd */
@@ -611,29 +611,50 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
         g_kernel_ns += (uint64_t)(ms * 1e6);
         g_launches += 1;
     }
-    /* rare: a job's adaptive band outgrew its reservation. The pinned stage
-     * (and the per-job pack buffers) may already hold the OTHER slot's batch,
-     * so the retry must fully REPACK this batch before relaunching. */
-    for (int attempt = 0;; ++attempt) {
-        bool overflow = false;
-        std::vector<int64_t> floor_est(n_jobs, 0);
+    /* rare: a job's adaptive band outgrew its reservation. Re-run ONLY the
+     * overflowed jobs as a fresh sub-batch on the dedicated retry slot
+     * (slot 2) with doubled reservations — a whole-batch relaunch costs a
+     * full latency-bound kernel pass. The sub-batch repacks from the graphs
+     * (the pinned stage may already hold another slot's data). */
+    if (slot != 2) {
+        std::vector<BatchJob> failed;
+        std::vector<int64_t> floors;
         for (int i = 0; i < n_jobs; ++i) {
-            floor_est[i] = PB.arena_est[i];
-            if (PB.hres[i].status == ABAMD_JOB_ARENA_OVERFLOW) { floor_est[i] *= 2; overflow = true; }
+            if (PB.hres[i].status == ABAMD_JOB_ARENA_OVERFLOW) {
+                BatchJob fj = batch[i];
+                fj.est_cells_hint = 0; /* use the formula floor below instead */
+                failed.push_back(fj);
+                floors.push_back(PB.arena_est[i] * 2);
+            }
         }
-        if (!overflow) break;
-        if (attempt > 8) {
-            fprintf(stderr, "[abpoa_amd] arena overflow persists after %d retries\n", attempt);
-            exit(EXIT_FAILURE);
+        if (!failed.empty()) {
+            prepare_internal(failed.data(), (int)failed.size(), 2, floors.data());
+            int r2 = abamd_gpu_batch_launch(2);
+            if (!r2) abamd_gpu_batch_finish_slot(2);
         }
-        prepare_internal(batch, n_jobs, slot, floor_est.data());
-        batch_launch(C, PB);
-        HIP_CHECK(hipStreamSynchronize(C.stream));
-        HIP_CHECK(hipMemcpy(PB.hres.data(), C.results.p, (size_t)n_jobs * sizeof(abamd_gpu_res_t), hipMemcpyDeviceToHost));
-        float ms = 0.f;
-        HIP_CHECK(hipEventElapsedTime(&ms, C.ev0, C.ev1));
-        g_kernel_ns += (uint64_t)(ms * 1e6);
-        g_launches += 1;
+    } else {
+        /* retry slot itself overflowed: escalate reservations in place */
+        for (int attempt = 0;; ++attempt) {
+            bool overflow = false;
+            std::vector<int64_t> floor_est(n_jobs, 0);
+            for (int i = 0; i < n_jobs; ++i) {
+                floor_est[i] = PB.arena_est[i];
+                if (PB.hres[i].status == ABAMD_JOB_ARENA_OVERFLOW) { floor_est[i] *= 2; overflow = true; }
+            }
+            if (!overflow) break;
+            if (attempt > 8) {
+                fprintf(stderr, "[abpoa_amd] arena overflow persists after %d retries\n", attempt);
+                exit(EXIT_FAILURE);
+            }
+            prepare_internal(batch, n_jobs, 2, floor_est.data());
+            batch_launch(C, PB);
+            HIP_CHECK(hipStreamSynchronize(C.stream));
+            HIP_CHECK(hipMemcpy(PB.hres.data(), C.results.p, (size_t)n_jobs * sizeof(abamd_gpu_res_t), hipMemcpyDeviceToHost));
+            float ms = 0.f;
+            HIP_CHECK(hipEventElapsedTime(&ms, C.ev0, C.ev1));
+            g_kernel_ns += (uint64_t)(ms * 1e6);
+            g_launches += 1;
+        }
     }
     g_gpu_ns += now_ns() - t_gpu0;
     uint64_t t_unpack0 = now_ns();
@@ -641,6 +662,8 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
     C.hcig.clear();
     for (int i = 0; i < n_jobs; ++i) {
         abamd_gpu_res_t &R = PB.hres[i];
+        if (slot != 2 && R.status == ABAMD_JOB_ARENA_OVERFLOW)
+            continue; /* re-run and unpacked by the retry sub-batch */
         if (R.status != ABAMD_JOB_OK) {
             fprintf(stderr, "[abpoa_amd] GPU job %d failed with status %d\n", i, R.status);
             exit(EXIT_FAILURE);
