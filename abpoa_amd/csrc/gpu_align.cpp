@@ -63,10 +63,13 @@ namespace {
 struct DevBuf {
     void *p = nullptr;
     size_t cap = 0;
+    /* grow exactly (x1.5 headroom), never shrink: repeated 100+GB
+     * free/malloc cycles cost seconds, and power-of-2 rounding on a
+     * ~200 GB arena would overshoot the device */
     void ensure(size_t n) {
         if (n <= cap) return;
-        size_t want = cap ? cap : 4096;
-        while (want < n) want <<= 1;
+        size_t want = n + n / 2;
+        if (want < 4096) want = 4096;
         if (p) HIP_CHECK(hipFree(p));
         HIP_CHECK(hipMalloc(&p, want));
         cap = want;
@@ -127,18 +130,6 @@ struct GpuCtx {
         HIP_CHECK(hipEventCreate(&ev0));
         HIP_CHECK(hipEventCreate(&ev1));
         HIP_CHECK(hipEventCreate(&ev_h2d));
-        /* reserve the plane arena at the full batch memory budget once:
-         * re-growing a 100+ GB hipMalloc every round costs seconds.
-         * Default budget: 70% of free HBM (ABPOA_AMD_MEM_GB overrides). */
-        double mem_gb;
-        const char *s = getenv("ABPOA_AMD_MEM_GB");
-        if (s && *s) mem_gb = atof(s);
-        else {
-            size_t free_b = 0, total_b = 0;
-            HIP_CHECK(hipMemGetInfo(&free_b, &total_b));
-            mem_gb = (double)free_b * 0.70 / 1e9;
-        }
-        arena.ensure((size_t)(mem_gb * 1e9));
         init = true;
     }
 };
