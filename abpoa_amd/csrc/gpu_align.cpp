@@ -602,28 +602,8 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
         g_kernel_ns += (uint64_t)(ms * 1e6);
         g_launches += 1;
     }
-    /* rare: a job's adaptive band outgrew its reservation. Re-run ONLY the
-     * overflowed jobs as a fresh sub-batch on the dedicated retry slot
-     * (slot 2) with doubled reservations — a whole-batch relaunch costs a
-     * full latency-bound kernel pass. The sub-batch repacks from the graphs
-     * (the pinned stage may already hold another slot's data). */
-    if (slot != 2) {
-        std::vector<BatchJob> failed;
-        std::vector<int64_t> floors;
-        for (int i = 0; i < n_jobs; ++i) {
-            if (PB.hres[i].status == ABAMD_JOB_ARENA_OVERFLOW) {
-                BatchJob fj = batch[i];
-                fj.est_cells_hint = 0; /* use the formula floor below instead */
-                failed.push_back(fj);
-                floors.push_back(PB.arena_est[i] * 2);
-            }
-        }
-        if (!failed.empty()) {
-            prepare_internal(failed.data(), (int)failed.size(), 2, floors.data());
-            int r2 = abamd_gpu_batch_launch(2);
-            if (!r2) abamd_gpu_batch_finish_slot(2);
-        }
-    } else {
+    g_gpu_ns += now_ns() - t_gpu0;
+    if (slot == 2) {
         /* retry slot itself overflowed: escalate reservations in place */
         for (int attempt = 0;; ++attempt) {
             bool overflow = false;
@@ -647,7 +627,6 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
             g_launches += 1;
         }
     }
-    g_gpu_ns += now_ns() - t_gpu0;
     uint64_t t_unpack0 = now_ns();
     /* unpack results */
     C.hcig.clear();
@@ -681,6 +660,29 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
         res->query_s = R.query_s; res->query_e = R.query_e;
     }
     g_unpack_ns += now_ns() - t_unpack0;
+
+    /* rare: a job's adaptive band outgrew its reservation. Re-run ONLY the
+     * overflowed jobs as a fresh sub-batch on the dedicated retry slot
+     * (slot 2) with doubled reservations — a whole-batch relaunch costs a
+     * full latency-bound kernel pass. The sub-batch repacks from the graphs
+     * (the pinned stage may already hold another slot's data). */
+    if (slot != 2) {
+        std::vector<BatchJob> failed;
+        std::vector<int64_t> floors;
+        for (int i = 0; i < n_jobs; ++i) {
+            if (PB.hres[i].status == ABAMD_JOB_ARENA_OVERFLOW) {
+                BatchJob fj = batch[i];
+                fj.est_cells_hint = 0; /* use the formula floor below instead */
+                failed.push_back(fj);
+                floors.push_back(PB.arena_est[i] * 2);
+            }
+        }
+        if (!failed.empty()) {
+            prepare_internal(failed.data(), (int)failed.size(), 2, floors.data());
+            int r2 = abamd_gpu_batch_launch(2);
+            if (!r2) abamd_gpu_batch_finish_slot(2);
+        }
+    }
     PB.active = false;
     return 0;
 }
