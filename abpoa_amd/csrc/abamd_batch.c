@@ -190,7 +190,7 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
         int _pl = abpt->gap_mode == ABPOA_CONVEX_GAP ? 5 : abpt->gap_mode == ABPOA_AFFINE_GAP ? 3 : 1; \
         double _cells = (double)_gn * (2.0 * _w + 160.0); \
         if ((J)->est_cells_hint > 0) { \
-            double _t = (double)(J)->est_cells_hint * 1.25 + _q; \
+            double _t = (double)(J)->est_cells_hint * 1.5 + _q; \
             if (_t < _cells) _cells = _t; \
         } \
         _cells * _pl * _ssz; })

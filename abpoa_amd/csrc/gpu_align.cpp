@@ -409,7 +409,7 @@ static int prepare_internal(BatchJob *batch, int n_jobs, int slot, const int64_t
                                         B.beg_node_id, B.end_node_id, B.query, B.qlen);
                 if (B.est_cells_hint > 0 && !getenv("ABPOA_AMD_NO_HINT")) {
                     /* bands drift a few % per round; 25% headroom + overflow retry */
-                    int64_t tight = B.est_cells_hint + B.est_cells_hint / 4 + B.qlen;
+                    int64_t tight = B.est_cells_hint + B.est_cells_hint / 2 + B.qlen;
                     if (tight < arena_est[i]) arena_est[i] = tight;
                 }
                 if (min_est && min_est[i] > arena_est[i]) arena_est[i] = min_est[i];
