@@ -15,6 +15,7 @@
 
 void abamd_timing_report(const char *tag);
 static double g_fold_s, g_cons_s, g_collect_s;
+static long g_big_items;
 
 /* from gpu_align.cpp */
 typedef struct {
@@ -28,10 +29,12 @@ typedef struct {
     int64_t *cells_out;
 } abamd_batch_job_t;
 int abamd_gpu_align_batch(abamd_batch_job_t *batch, int n_jobs);
+int abamd_gpu_align_batch_slot(abamd_batch_job_t *batch, int n_jobs, int slot);
 int64_t abamd_gpu_free_mem(void);
 int abamd_gpu_batch_prepare(abamd_batch_job_t *batch, int n_jobs, int slot);
 int abamd_gpu_batch_launch(int slot);
 int abamd_gpu_batch_finish_slot(int slot);
+void abamd_gpu_set_arena_cap(uint64_t bytes);
 
 typedef struct {
     abpoa_t *ab;
@@ -134,10 +137,14 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
         if (s && *s) mem_gb = atof(s);
     }
     if (mem_gb <= 0.0) {
+        /* up to three pipeline groups keep three arenas resident
+         * concurrently, so each launch gets just under a third of the
+         * usable HBM (abamd_gpu_free_mem counts our own held buffers) */
         int64_t free_b = abamd_gpu_free_mem();
-        mem_gb = free_b > 0 ? (double)free_b * 0.70 / 1e9 : 48.0;
+        mem_gb = free_b > 0 ? (double)free_b * 0.28 / 1e9 : 48.0;
     }
     const double budget_bytes = mem_gb * 1e9;
+    abamd_gpu_set_arena_cap((uint64_t)budget_bytes);
 
     /* ---- round 0: first reads thread straight into their graphs ---- */
     {
@@ -156,26 +163,36 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
      * Group count is sized so one group-round fits the device memory
      * budget at the LAST (largest) round; with >= 2 groups, the host fold
      * and the next pack run while the other group's kernel executes. */
-    /* two groups: every fold/pack overlaps the other group's kernel. Launch
-     * wall time is dominated by the per-row dependence chain, so FEWER,
-     * larger launches always win; if a group-round's arena demand exceeds
-     * the budget it falls back to sequential sub-chunks for that item. */
-    int n_groups = n_sets >= 2 ? 2 : 1;
+    /* three groups, lookahead two: while item `it` is finished, items
+     * it+1 and it+2 (disjoint set groups) are already in flight on their
+     * own streams — the DP kernel is latency-bound at ~1 wave per job, so
+     * two overlapped launches nearly double chip fill. A group-round whose
+     * arena demand exceeds the budget falls back to sequential sub-chunks
+     * on the dedicated slot 3. */
+    int n_groups = n_sets >= 3 ? 3 : n_sets;
+    {
+        const char *gs = getenv("ABPOA_AMD_GROUPS");
+        if (gs && *gs) {
+            int g = atoi(gs);
+            if (g >= 1 && g <= 3 && g <= n_sets) n_groups = g;
+        }
+    }
     int *grp_of = (int*)abamd_malloc((size_t)n_sets * sizeof(int));
     for (i = 0; i < n_sets; ++i) grp_of[i] = i % n_groups;
 
     /* item = (round r >= 1, group g); g-major within a round */
     long n_items = (long)(max_reads - 1) * n_groups;
-    abamd_batch_job_t *slot_jobs[2];
+    abamd_batch_job_t *slot_jobs[3];
     slot_jobs[0] = (abamd_batch_job_t*)abamd_malloc((size_t)n_sets * sizeof(abamd_batch_job_t));
     slot_jobs[1] = (abamd_batch_job_t*)abamd_malloc((size_t)n_sets * sizeof(abamd_batch_job_t));
-    int slot_nj[2] = {0, 0};
+    slot_jobs[2] = (abamd_batch_job_t*)abamd_malloc((size_t)n_sets * sizeof(abamd_batch_job_t));
+    int slot_nj[3] = {0, 0, 0};
 
     #define ITEM_R(it) (1 + (int)((it) / n_groups))
     #define ITEM_G(it) ((int)((it) % n_groups))
 
-    double slot_est[2] = {0, 0};
-    int slot_big[2] = {0, 0};
+    double slot_est[3] = {0, 0, 0};
+    int slot_big[3] = {0, 0, 0};
 
     /* arena bytes one job will demand (matches gpu_align.cpp's reservation,
      * including the measured-cells tightening) */
@@ -217,9 +234,11 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
             ++_nj; \
         } \
         slot_nj[slot] = _nj; slot_est[slot] = _est; slot_big[slot] = _est > budget_bytes; \
+        if (slot_big[slot]) ++g_big_items; \
     } while (0)
 
-    /* oversized item: memory-bounded sequential chunks (not pipelined) */
+    /* oversized item: memory-bounded sequential chunks (not pipelined),
+     * on device slot 3 so an in-flight pipeline slot is never clobbered */
     #define RUN_BIG_ITEM(slot) do { \
         int _done = 0; \
         while (_done < slot_nj[slot]) { \
@@ -229,7 +248,7 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
                 if (_take > 0 && _acc + _e > budget_bytes) break; \
                 _acc += _e; ++_take; \
             } \
-            abamd_gpu_align_batch(&slot_jobs[slot][_done], _take); \
+            abamd_gpu_align_batch_slot(&slot_jobs[slot][_done], _take, 3); \
             _done += _take; \
         } \
     } while (0)
@@ -248,33 +267,47 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
     } while (0)
 
     if (n_items > 0) {
-        int slot = 0;
-        BUILD_ITEM(0, slot);
-        if (slot_big[slot]) RUN_BIG_ITEM(slot);
-        else {
-            abamd_gpu_batch_prepare(slot_jobs[slot], slot_nj[slot], slot);
-            abamd_gpu_batch_launch(slot);
+        /* lookahead = n_groups-1 items launched ahead of the one being
+         * finished: item nx = it+LA is the same GROUP as item it-1, whose
+         * fold completes just before BUILD(nx), so every launched item's
+         * graphs are complete up to its round. LA=0 (single set) degrades
+         * to strict build->launch->finish->fold per round. */
+        const int LA = n_groups - 1;
+        const int n_pipe = LA > 0 ? LA + 1 : 1;
+        for (long k = 0; k < LA && k < n_items; ++k) {
+            int s = (int)(k % n_pipe);
+            BUILD_ITEM(k, s);
+            if (slot_big[s]) RUN_BIG_ITEM(s);
+            else {
+                abamd_gpu_batch_prepare(slot_jobs[s], slot_nj[s], s);
+                abamd_gpu_batch_launch(s);
+            }
         }
         for (long it = 0; it < n_items; ++it) {
-            int nslot = slot ^ 1;
-            if (it + 1 < n_items) {
-                /* overlapped with item `it`'s kernel: fold the previous item,
-                 * then pack the next one (its group folded >= 1 item ago) */
-                if (it >= 1) FOLD_ITEM(it - 1);
-                BUILD_ITEM(it + 1, nslot);
-                if (!slot_big[nslot])
-                    abamd_gpu_batch_prepare(slot_jobs[nslot], slot_nj[nslot], nslot);
-                if (!slot_big[slot]) abamd_gpu_batch_finish_slot(slot);
-                if (slot_big[nslot]) RUN_BIG_ITEM(nslot);
-                else abamd_gpu_batch_launch(nslot);
-            } else {
-                if (it >= 1) FOLD_ITEM(it - 1);
-                if (!slot_big[slot]) abamd_gpu_batch_finish_slot(slot);
+            int s = (int)(it % n_pipe);
+            if (it >= 1) FOLD_ITEM(it - 1);
+            long nx = it + LA;
+            int ns = (int)(nx % n_pipe);
+            if (LA > 0 && nx < n_items) {
+                BUILD_ITEM(nx, ns);
+                if (!slot_big[ns]) {
+                    abamd_gpu_batch_prepare(slot_jobs[ns], slot_nj[ns], ns);
+                    abamd_gpu_batch_launch(ns);
+                }
+            } else if (LA == 0) {
+                BUILD_ITEM(it, 0);
+                if (!slot_big[0]) {
+                    abamd_gpu_batch_prepare(slot_jobs[0], slot_nj[0], 0);
+                    abamd_gpu_batch_launch(0);
+                }
             }
-            slot = nslot;
+            if (!slot_big[s]) abamd_gpu_batch_finish_slot(s);
+            if (LA > 0 && nx < n_items && slot_big[ns]) RUN_BIG_ITEM(ns);
+            else if (LA == 0 && slot_big[0]) RUN_BIG_ITEM(0);
         }
         FOLD_ITEM(n_items - 1);
     }
+    free(slot_jobs[2]);
     free(grp_of); free(slot_jobs[0]); free(slot_jobs[1]);
 
     /* consensus on host threads, then emit callbacks in order */
@@ -286,8 +319,8 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
     if (cb) for (i = 0; i < n_sets; ++i) cb(i, sets[i].ab->abc, user);
 
     if (getenv("ABPOA_AMD_TIMING")) {
-        fprintf(stderr, "[abamd timing batch] fold %.2fs cons %.2fs (threads %d)\n",
-                g_fold_s, g_cons_s, n_host_threads);
+        fprintf(stderr, "[abamd timing batch] fold %.2fs cons %.2fs big_items %ld (threads %d)\n",
+                g_fold_s, g_cons_s, g_big_items, n_host_threads);
         abamd_timing_report("batch");
     }
     for (i = 0; i < n_sets; ++i) {

@@ -32,15 +32,21 @@
 
 static std::atomic<uint64_t> g_dp_cells{0}, g_kernel_ns{0}, g_launches{0}, g_alg_bytes{0};
 static std::atomic<uint64_t> g_pack_ns{0}, g_stage_ns{0}, g_gpu_ns{0}, g_unpack_ns{0};
+static std::atomic<uint64_t> g_retry_jobs{0};
+static std::atomic<uint64_t> g_dev_held{0};   /* bytes held by our DevBufs */
+/* per-slot arena growth clamp, set by the batch driver to its per-launch
+ * memory budget so two resident arenas can never overshoot free HBM */
+static std::atomic<uint64_t> g_arena_cap{0};
+extern "C" void abamd_gpu_set_arena_cap(uint64_t bytes) { g_arena_cap = bytes; }
 static inline uint64_t now_ns() {
     struct timespec ts; clock_gettime(CLOCK_MONOTONIC, &ts);
     return (uint64_t)ts.tv_sec * 1000000000ull + ts.tv_nsec;
 }
 extern "C" void abamd_timing_report(const char *tag) {
     if (!getenv("ABPOA_AMD_TIMING")) return;
-    fprintf(stderr, "[abamd timing %s] pack %.2fs stage %.2fs gpu(upload->sync) %.2fs unpack %.2fs kernel %.2fs launches %llu\n",
+    fprintf(stderr, "[abamd timing %s] pack %.2fs stage %.2fs gpu(upload->sync) %.2fs unpack %.2fs kernel %.2fs launches %llu retry_jobs %llu\n",
             tag, g_pack_ns/1e9, g_stage_ns/1e9, g_gpu_ns/1e9, g_unpack_ns/1e9, g_kernel_ns/1e9,
-            (unsigned long long)g_launches.load());
+            (unsigned long long)g_launches.load(), (unsigned long long)g_retry_jobs.load());
 }
 
 extern "C" void abpoa_amd_get_stats(uint64_t *dp_cells, uint64_t *kernel_ns, uint64_t *n_launches) {
@@ -63,15 +69,29 @@ namespace {
 struct DevBuf {
     void *p = nullptr;
     size_t cap = 0;
-    /* grow exactly (x1.5 headroom), never shrink: repeated 100+GB
-     * free/malloc cycles cost seconds, and power-of-2 rounding on a
-     * ~200 GB arena would overshoot the device */
-    void ensure(size_t n) {
+    /* grow with headroom, never shrink: each regrow is a hipFree+hipMalloc
+     * (hipFree synchronizes the WHOLE device, stalling the other slot's
+     * in-flight kernel) and mapping tens of GB costs ~1 s, so regrows must
+     * stay rare. The arena's demand grows monotonically with the graphs
+     * (~3x first-to-last round), so x2 headroom caps regrows at ~2-3 per
+     * slot; power-of-2 rounding on a ~100 GB arena would overshoot the
+     * device, exact x2 of demand does not. */
+    void ensure(size_t n, int big = 0, size_t limit = 0) {
         if (n <= cap) return;
-        size_t want = n + n / 2;
+        size_t want = big ? n * 2 : n + n / 2;
+        if (big && limit) {
+            /* budget-capped arena: overshoot aggressively toward the cap so
+             * the slot allocates once and never regrows mid-pipeline */
+            want = n * 8;
+            if (want > limit) want = limit;
+            if (want < n) want = n;
+        } else if (limit && want > limit) {
+            want = n > limit ? n : limit;
+        }
         if (want < 4096) want = 4096;
         if (p) HIP_CHECK(hipFree(p));
         HIP_CHECK(hipMalloc(&p, want));
+        g_dev_held += want - cap;
         cap = want;
     }
 };
@@ -103,18 +123,35 @@ struct PinnedBuf {
     }
 };
 
-struct GpuCtx {
+/* Per-slot device context: slots 0/1 are the two pipeline groups, slot 2 the
+ * overflow-retry sub-batch. Each slot owns its stream and device buffers so
+ * the two groups' kernels can be IN FLIGHT CONCURRENTLY (the per-row latency
+ * chain leaves most CUs idle at ~500 waves/launch; two overlapped launches
+ * roughly double chip fill). Streams are non-blocking so the null-stream
+ * hipMemcpy in unpack never serializes against the other slot's kernel. */
+struct SlotDev {
     bool init = false;
     hipStream_t stream;
     hipEvent_t ev0, ev1, ev_h2d;
     DevBuf slab;        /* all per-job inputs + metadata */
     DevBuf arena;       /* DP plane arena */
     DevBuf jobs, results, cigars;
-    PinnedBuf stage2[3];              /* per-slot pinned H2D staging (slot 2 = retry) */
-    std::vector<HostBuf> jb_bufs2[3]; /* per-slot per-job pack buffers */
+    void ensure_init() {
+        if (init) return;
+        HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
+        HIP_CHECK(hipEventCreate(&ev0));
+        HIP_CHECK(hipEventCreate(&ev1));
+        HIP_CHECK(hipEventCreate(&ev_h2d));
+        init = true;
+    }
+};
+
+struct GpuCtx {
+    bool init = false;
+    SlotDev dev[4];
+    PinnedBuf stage2[4];              /* per-slot pinned H2D staging (slot 3 = retry/big) */
+    std::vector<HostBuf> jb_bufs2[4]; /* per-slot per-job pack buffers */
     HostBuf hb;
-    std::vector<abamd_gpu_job_t> hjobs;
-    std::vector<abamd_gpu_res_t> hres;
     std::vector<uint64_t> hcig;
     void ensure_init() {
         if (init) return;
@@ -126,10 +163,6 @@ struct GpuCtx {
                     hipGetErrorString(e));
             exit(EXIT_FAILURE);
         }
-        HIP_CHECK(hipStreamCreate(&stream));
-        HIP_CHECK(hipEventCreate(&ev0));
-        HIP_CHECK(hipEventCreate(&ev1));
-        HIP_CHECK(hipEventCreate(&ev_h2d));
         init = true;
     }
 };
@@ -356,17 +389,21 @@ struct PendingBatch {
     size_t total = 0;
     bool active = false;
 };
-thread_local PendingBatch g_slots[3];
+thread_local PendingBatch g_slots[4];
 }
 
 static int batch_launch(GpuCtx &C, PendingBatch &P); /* fwd */
 
+extern "C" int abamd_gpu_align_batch_slot(BatchJob *batch, int n_jobs, int slot) {
+    int r = abamd_gpu_batch_prepare(batch, n_jobs, slot);
+    if (r) return r;
+    r = abamd_gpu_batch_launch(slot);
+    if (r) return r;
+    return abamd_gpu_batch_finish_slot(slot);
+}
+
 extern "C" int abamd_gpu_align_batch(BatchJob *batch, int n_jobs) {
-    int r = abamd_gpu_batch_prepare(batch, n_jobs, 0);
-    if (r) return r;
-    r = abamd_gpu_batch_launch(0);
-    if (r) return r;
-    return abamd_gpu_batch_finish_slot(0);
+    return abamd_gpu_align_batch_slot(batch, n_jobs, 0);
 }
 
 static int prepare_internal(BatchJob *batch, int n_jobs, int slot, const int64_t *min_est);
@@ -500,7 +537,7 @@ extern "C" int abamd_gpu_batch_launch(int slot) {
     int rr = batch_launch(C, PB);
     /* wait for the H2D copies only: the caller may repack the pinned staging
      * buffer for the other slot while this kernel runs */
-    HIP_CHECK(hipEventSynchronize(C.ev_h2d));
+    HIP_CHECK(hipEventSynchronize(C.dev[slot].ev_h2d));
     g_gpu_ns += now_ns() - t_gpu0;
     return rr;
 }
@@ -515,6 +552,8 @@ static int batch_launch(GpuCtx &C, PendingBatch &PB) {
     const int planes = PB.planes;
     const size_t total = PB.total;
     const size_t ssz = bits == 16 ? 2 : 4;
+    SlotDev &D = C.dev[PB.slot];
+    D.ensure_init();
     (void)abpt;
     {
         /* arena layout */
@@ -523,7 +562,7 @@ static int batch_launch(GpuCtx &C, PendingBatch &PB) {
             packs[i].arena_off = arena_cells;
             arena_cells += arena_est[i];
         }
-        C.arena.ensure((size_t)arena_cells * planes * ssz);
+        D.arena.ensure((size_t)arena_cells * planes * ssz, 1, g_arena_cap.load());
 
         /* cigar buffers */
         int cig_total = 0;
@@ -531,12 +570,12 @@ static int batch_launch(GpuCtx &C, PendingBatch &PB) {
             packs[i].cigar_off = cig_total;
             cig_total += 2 * packs[i].qlen + 1024;
         }
-        C.cigars.ensure((size_t)cig_total * 8);
+        D.cigars.ensure((size_t)cig_total * 8);
 
         /* upload slab + jobs */
-        C.slab.ensure(total);
-        HIP_CHECK(hipMemcpyAsync(C.slab.p, C.stage2[PB.slot].p, total, hipMemcpyHostToDevice, C.stream));
-        uint8_t *S0 = (uint8_t*)C.slab.p;
+        D.slab.ensure(total);
+        HIP_CHECK(hipMemcpyAsync(D.slab.p, C.stage2[PB.slot].p, total, hipMemcpyHostToDevice, D.stream));
+        uint8_t *S0 = (uint8_t*)D.slab.p;
         for (int i = 0; i < n_jobs; ++i) {
             JobPack &P = packs[i];
             abamd_gpu_job_t &jb = P.jb;
@@ -554,30 +593,30 @@ static int batch_launch(GpuCtx &C, PendingBatch &PB) {
             jb.max_right = (int*)(S + P.o_mr);
             jb.row_meta = S + P.o_meta;
             jb.mat = (int*)(S0 + 0); /* matrix is at slab offset 0 */
-            jb.arena = (uint8_t*)C.arena.p + (size_t)P.arena_off * planes * ssz;
+            jb.arena = (uint8_t*)D.arena.p + (size_t)P.arena_off * planes * ssz;
             jb.arena_cap = arena_est[i];
-            jb.cigar = (uint64_t*)C.cigars.p + P.cigar_off;
+            jb.cigar = (uint64_t*)D.cigars.p + P.cigar_off;
             jb.cigar_cap = 2 * P.qlen + 1024;
             PB.hjobs[i] = jb;
         }
-        C.jobs.ensure((size_t)n_jobs * sizeof(abamd_gpu_job_t));
-        C.results.ensure((size_t)n_jobs * sizeof(abamd_gpu_res_t));
-        HIP_CHECK(hipMemcpyAsync(C.jobs.p, PB.hjobs.data(), (size_t)n_jobs * sizeof(abamd_gpu_job_t), hipMemcpyHostToDevice, C.stream));
-        HIP_CHECK(hipEventRecord(C.ev_h2d, C.stream));
+        D.jobs.ensure((size_t)n_jobs * sizeof(abamd_gpu_job_t));
+        D.results.ensure((size_t)n_jobs * sizeof(abamd_gpu_res_t));
+        HIP_CHECK(hipMemcpyAsync(D.jobs.p, PB.hjobs.data(), (size_t)n_jobs * sizeof(abamd_gpu_job_t), hipMemcpyHostToDevice, D.stream));
+        HIP_CHECK(hipEventRecord(D.ev_h2d, D.stream));
 
-        HIP_CHECK(hipEventRecord(C.ev0, C.stream));
+        HIP_CHECK(hipEventRecord(D.ev0, D.stream));
         {
-            abamd_gpu_job_t *J = (abamd_gpu_job_t*)C.jobs.p;
-            abamd_gpu_res_t *R = (abamd_gpu_res_t*)C.results.p;
+            abamd_gpu_job_t *J = (abamd_gpu_job_t*)D.jobs.p;
+            abamd_gpu_res_t *R = (abamd_gpu_res_t*)D.results.p;
             if (abpt->gap_mode == ABPOA_CONVEX_GAP)
-                bits == 16 ? abamd_launch_cg_i16(J, R, n_jobs, C.stream) : abamd_launch_cg_i32(J, R, n_jobs, C.stream);
+                bits == 16 ? abamd_launch_cg_i16(J, R, n_jobs, D.stream) : abamd_launch_cg_i32(J, R, n_jobs, D.stream);
             else if (abpt->gap_mode == ABPOA_AFFINE_GAP)
-                bits == 16 ? abamd_launch_ag_i16(J, R, n_jobs, C.stream) : abamd_launch_ag_i32(J, R, n_jobs, C.stream);
+                bits == 16 ? abamd_launch_ag_i16(J, R, n_jobs, D.stream) : abamd_launch_ag_i32(J, R, n_jobs, D.stream);
             else
-                bits == 16 ? abamd_launch_lg_i16(J, R, n_jobs, C.stream) : abamd_launch_lg_i32(J, R, n_jobs, C.stream);
+                bits == 16 ? abamd_launch_lg_i16(J, R, n_jobs, D.stream) : abamd_launch_lg_i32(J, R, n_jobs, D.stream);
         }
         HIP_CHECK(hipGetLastError());
-        HIP_CHECK(hipEventRecord(C.ev1, C.stream));
+        HIP_CHECK(hipEventRecord(D.ev1, D.stream));
         /* results D2H happens in finish(): an async copy into pageable host
          * memory would silently synchronize and stall the pipeline */
     }
@@ -593,17 +632,18 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
     std::vector<JobPack> &packs = PB.packs;
     const size_t ssz = PB.bits == 16 ? 2 : 4;
     const int planes = PB.planes;
+    SlotDev &D = C.dev[slot];
     uint64_t t_gpu0 = now_ns();
-    HIP_CHECK(hipStreamSynchronize(C.stream));
-    HIP_CHECK(hipMemcpy(PB.hres.data(), C.results.p, (size_t)n_jobs * sizeof(abamd_gpu_res_t), hipMemcpyDeviceToHost));
+    HIP_CHECK(hipStreamSynchronize(D.stream));
+    HIP_CHECK(hipMemcpy(PB.hres.data(), D.results.p, (size_t)n_jobs * sizeof(abamd_gpu_res_t), hipMemcpyDeviceToHost));
     {
         float ms = 0.f;
-        HIP_CHECK(hipEventElapsedTime(&ms, C.ev0, C.ev1));
+        HIP_CHECK(hipEventElapsedTime(&ms, D.ev0, D.ev1));
         g_kernel_ns += (uint64_t)(ms * 1e6);
         g_launches += 1;
     }
     g_gpu_ns += now_ns() - t_gpu0;
-    if (slot == 2) {
+    if (slot == 3) {
         /* retry slot itself overflowed: escalate reservations in place */
         for (int attempt = 0;; ++attempt) {
             bool overflow = false;
@@ -617,12 +657,12 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
                 fprintf(stderr, "[abpoa_amd] arena overflow persists after %d retries\n", attempt);
                 exit(EXIT_FAILURE);
             }
-            prepare_internal(batch, n_jobs, 2, floor_est.data());
+            prepare_internal(batch, n_jobs, 3, floor_est.data());
             batch_launch(C, PB);
-            HIP_CHECK(hipStreamSynchronize(C.stream));
-            HIP_CHECK(hipMemcpy(PB.hres.data(), C.results.p, (size_t)n_jobs * sizeof(abamd_gpu_res_t), hipMemcpyDeviceToHost));
+            HIP_CHECK(hipStreamSynchronize(D.stream));
+            HIP_CHECK(hipMemcpy(PB.hres.data(), D.results.p, (size_t)n_jobs * sizeof(abamd_gpu_res_t), hipMemcpyDeviceToHost));
             float ms = 0.f;
-            HIP_CHECK(hipEventElapsedTime(&ms, C.ev0, C.ev1));
+            HIP_CHECK(hipEventElapsedTime(&ms, D.ev0, D.ev1));
             g_kernel_ns += (uint64_t)(ms * 1e6);
             g_launches += 1;
         }
@@ -632,7 +672,7 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
     C.hcig.clear();
     for (int i = 0; i < n_jobs; ++i) {
         abamd_gpu_res_t &R = PB.hres[i];
-        if (slot != 2 && R.status == ABAMD_JOB_ARENA_OVERFLOW)
+        if (slot != 3 && R.status == ABAMD_JOB_ARENA_OVERFLOW)
             continue; /* re-run and unpacked by the retry sub-batch */
         if (R.status != ABAMD_JOB_OK) {
             fprintf(stderr, "[abpoa_amd] GPU job %d failed with status %d\n", i, R.status);
@@ -645,7 +685,7 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
         res->best_score = R.best_score;
         if (batch[i].abpt->ret_cigar && R.n_cigar > 0) {
             uint64_t *tmp = (uint64_t*)malloc((size_t)R.n_cigar * 8);
-            HIP_CHECK(hipMemcpy(tmp, (uint64_t*)C.cigars.p + packs[i].cigar_off, (size_t)R.n_cigar * 8, hipMemcpyDeviceToHost));
+            HIP_CHECK(hipMemcpy(tmp, (uint64_t*)D.cigars.p + packs[i].cigar_off, (size_t)R.n_cigar * 8, hipMemcpyDeviceToHost));
             if (!batch[i].abpt->rev_cigar) { /* reverse to front-to-back order */
                 for (int a = 0, b = R.n_cigar - 1; a < b; ++a, --b) {
                     uint64_t t = tmp[a]; tmp[a] = tmp[b]; tmp[b] = t;
@@ -666,7 +706,7 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
      * (slot 2) with doubled reservations — a whole-batch relaunch costs a
      * full latency-bound kernel pass. The sub-batch repacks from the graphs
      * (the pinned stage may already hold another slot's data). */
-    if (slot != 2) {
+    if (slot != 3) {
         std::vector<BatchJob> failed;
         std::vector<int64_t> floors;
         for (int i = 0; i < n_jobs; ++i) {
@@ -678,19 +718,23 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
             }
         }
         if (!failed.empty()) {
-            prepare_internal(failed.data(), (int)failed.size(), 2, floors.data());
-            int r2 = abamd_gpu_batch_launch(2);
-            if (!r2) abamd_gpu_batch_finish_slot(2);
+            g_retry_jobs += failed.size();
+            prepare_internal(failed.data(), (int)failed.size(), 3, floors.data());
+            int r2 = abamd_gpu_batch_launch(3);
+            if (!r2) abamd_gpu_batch_finish_slot(3);
         }
     }
     PB.active = false;
     return 0;
 }
 
+/* free HBM PLUS what our own never-shrink buffers already hold: the batch
+ * driver's per-launch budget must not ratchet down as the arenas it sized on
+ * a previous call stay resident (they are reused, not re-allocated) */
 extern "C" int64_t abamd_gpu_free_mem(void) {
     size_t free_b = 0, total_b = 0;
     if (hipMemGetInfo(&free_b, &total_b) != hipSuccess) return 0;
-    return (int64_t)free_b;
+    return (int64_t)(free_b + g_dev_held.load());
 }
 
 extern "C" int abamd_gpu_align_sequence_to_subgraph(abpoa_t *ab, abpoa_para_t *abpt,
