@@ -14,8 +14,9 @@
 #include "abamd_util.h"
 
 void abamd_timing_report(const char *tag);
-static double g_fold_s, g_cons_s, g_collect_s;
+static double g_fold_s, g_cons_s, g_collect_s, g_build_s;
 static long g_big_items;
+static _Atomic unsigned long long g_fold_work_ns; /* summed across workers */
 
 /* from gpu_align.cpp */
 typedef struct {
@@ -72,13 +73,16 @@ static void fold_one(fold_work_t *w, int si) {
     if (r >= st->n_seqs) return;
     int qlen = st->seq_lens[r];
     uint8_t *q = (uint8_t*)st->seqs[r];
+    double t0 = abamd_realtime();
     abpoa_add_graph_alignment(st->ab, w->abpt, q, ones_weight(st, qlen), qlen, NULL,
                               st->res, r, st->n_seqs, 1);
+    g_fold_work_ns += (unsigned long long)((abamd_realtime() - t0) * 1e9);
     if (st->res.n_cigar) { free(st->res.graph_cigar); st->res.graph_cigar = NULL; st->res.n_cigar = 0; }
 }
 
-static void *fold_worker(void *arg) {
+static void fold_worker(void *arg, int tid, int nthr) {
     fold_work_t *w = (fold_work_t*)arg;
+    (void)tid; (void)nthr;
     for (;;) {
         pthread_mutex_lock(&w->mu);
         int si = w->next++;
@@ -86,7 +90,6 @@ static void *fold_worker(void *arg) {
         if (si >= w->n_sets) break;
         if (w->sets[si].active) fold_one(w, si);
     }
-    return NULL;
 }
 
 typedef struct {
@@ -97,8 +100,9 @@ typedef struct {
     pthread_mutex_t mu;
 } cons_work_t;
 
-static void *cons_worker(void *arg) {
+static void cons_worker(void *arg, int tid, int nthr) {
     cons_work_t *w = (cons_work_t*)arg;
+    (void)tid; (void)nthr;
     for (;;) {
         pthread_mutex_lock(&w->mu);
         int si = w->next++;
@@ -107,7 +111,6 @@ static void *cons_worker(void *arg) {
         w->sets[si].ab->abs->n_seq = w->sets[si].n_seqs; /* consensus reads n_seq */
         abpoa_generate_consensus(w->sets[si].ab, w->abpt);
     }
-    return NULL;
 }
 
 int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
@@ -154,8 +157,7 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
             memset(&sets[i].res, 0, sizeof(abpoa_res_t));
         }
         fold_work_t fw = { sets, n_sets, abpt, 0, 0, PTHREAD_MUTEX_INITIALIZER };
-        for (i = 0; i < n_host_threads; ++i) pthread_create(&tids[i], NULL, fold_worker, &fw);
-        for (i = 0; i < n_host_threads; ++i) pthread_join(tids[i], NULL);
+        abamd_pool_run(fold_worker, &fw, n_host_threads);
         g_fold_s += abamd_realtime() - tf0;
     }
 
@@ -214,6 +216,7 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
 
     /* build the job list for one (round, group) item into a slot */
     #define BUILD_ITEM(it, slot) do { \
+        double _tb0 = abamd_realtime(); \
         int _r = ITEM_R(it), _g = ITEM_G(it), _nj = 0; \
         double _est = 0; \
         for (i = 0; i < n_sets; ++i) { \
@@ -235,6 +238,7 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
         } \
         slot_nj[slot] = _nj; slot_est[slot] = _est; slot_big[slot] = _est > budget_bytes; \
         if (slot_big[slot]) ++g_big_items; \
+        g_build_s += abamd_realtime() - _tb0; \
     } while (0)
 
     /* oversized item: memory-bounded sequential chunks (not pipelined),
@@ -260,9 +264,7 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
         for (i = 0; i < n_sets; ++i) \
             sets[i].active = (grp_of[i] == _g && _r < sets[i].n_seqs); \
         fold_work_t fw = { sets, n_sets, abpt, _r, 0, PTHREAD_MUTEX_INITIALIZER }; \
-        int _nt = n_host_threads; \
-        for (i = 0; i < _nt; ++i) pthread_create(&tids[i], NULL, fold_worker, &fw); \
-        for (i = 0; i < _nt; ++i) pthread_join(tids[i], NULL); \
+        abamd_pool_run(fold_worker, &fw, n_host_threads); \
         g_fold_s += abamd_realtime() - _tf0; \
     } while (0)
 
@@ -313,14 +315,13 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
     /* consensus on host threads, then emit callbacks in order */
     double tc0 = abamd_realtime();
     cons_work_t cw = { sets, n_sets, 0, abpt, cb, user, PTHREAD_MUTEX_INITIALIZER };
-    for (i = 0; i < n_host_threads; ++i) pthread_create(&tids[i], NULL, cons_worker, &cw);
-    for (i = 0; i < n_host_threads; ++i) pthread_join(tids[i], NULL);
+    abamd_pool_run(cons_worker, &cw, n_host_threads);
     g_cons_s += abamd_realtime() - tc0;
     if (cb) for (i = 0; i < n_sets; ++i) cb(i, sets[i].ab->abc, user);
 
     if (getenv("ABPOA_AMD_TIMING")) {
-        fprintf(stderr, "[abamd timing batch] fold %.2fs cons %.2fs big_items %ld (threads %d)\n",
-                g_fold_s, g_cons_s, g_big_items, n_host_threads);
+        fprintf(stderr, "[abamd timing batch] fold %.2fs (work %.2f cpu-s) build+toposort %.2fs cons %.2fs big_items %ld (threads %d)\n",
+                g_fold_s, (double)g_fold_work_ns / 1e9, g_build_s, g_cons_s, g_big_items, n_host_threads);
         abamd_timing_report("batch");
     }
     for (i = 0; i < n_sets; ++i) {

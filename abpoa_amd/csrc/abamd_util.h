@@ -72,6 +72,12 @@ double abamd_realtime(void);
 double abamd_cputime(void);
 double abamd_peakrss(void);
 
+/* persistent host thread pool (abamd_pool.c): fn(arg, tid, nthr) runs on
+ * workers tid = 0..nthr-1; returns when all are done. Calls must not nest. */
+typedef void (*abamd_pool_fn)(void *arg, int tid, int nthr);
+void abamd_pool_run(abamd_pool_fn fn, void *arg, int nthr);
+int abamd_pool_size(void);
+
 #ifdef __cplusplus
 }
 #endif

@@ -25,6 +25,7 @@
 #include <thread>
 #include "abpoa_amd.h"
 #include "gpu_core.h"
+#include "abamd_util.h"
 
 #define HIP_CHECK(x) do { hipError_t _e = (x); if (_e != hipSuccess) { \
     fprintf(stderr, "[abpoa_amd] HIP error %s at %s:%d: %s\n", hipGetErrorName(_e), __FILE__, __LINE__, hipGetErrorString(_e)); \
@@ -429,38 +430,47 @@ static int prepare_internal(BatchJob *batch, int n_jobs, int slot, const int64_t
     std::vector<int64_t> arena_est(n_jobs);
     std::vector<int> bits_v(n_jobs, 16);
 
-    /* parallel pack: each job into its own (reused) host buffer */
+    /* parallel pack on the persistent pool: each job into its own (reused)
+     * host buffer; pool threads keep their thread-local scratch warm */
     {
-        int nthr = (int)std::thread::hardware_concurrency();
-        if (nthr < 1) nthr = 1;
-        if (nthr > 64) nthr = 64;
-        if (nthr > n_jobs) nthr = n_jobs;
-        std::atomic<int> next{0};
-        auto worker = [&]() {
+        struct PackCtx {
+            std::atomic<int> next{0};
+            BatchJob *batch; int n_jobs;
+            std::vector<HostBuf> *jbufs;
+            std::vector<JobPack> *packs;
+            std::vector<int64_t> *arena_est;
+            std::vector<int> *bits_v;
+            const int64_t *min_est;
+            bool no_hint;
+        } ctx;
+        ctx.batch = batch; ctx.n_jobs = n_jobs; ctx.jbufs = &jbufs;
+        ctx.packs = &packs; ctx.arena_est = &arena_est; ctx.bits_v = &bits_v;
+        ctx.min_est = min_est; ctx.no_hint = getenv("ABPOA_AMD_NO_HINT") != nullptr;
+        auto worker = [](void *p, int, int) {
+            PackCtx &c = *(PackCtx*)p;
             for (;;) {
-                int i = next.fetch_add(1);
-                if (i >= n_jobs) break;
-                BatchJob &B = batch[i];
-                jbufs[i].reset();
-                arena_est[i] = pack_job(jbufs[i], packs[i], B.ab, B.abpt,
-                                        B.beg_node_id, B.end_node_id, B.query, B.qlen);
-                if (B.est_cells_hint > 0 && !getenv("ABPOA_AMD_NO_HINT")) {
-                    /* bands drift a few % per round; 25% headroom + overflow retry */
+                int i = c.next.fetch_add(1);
+                if (i >= c.n_jobs) break;
+                BatchJob &B = c.batch[i];
+                (*c.jbufs)[i].reset();
+                (*c.arena_est)[i] = pack_job((*c.jbufs)[i], (*c.packs)[i], B.ab, B.abpt,
+                                             B.beg_node_id, B.end_node_id, B.query, B.qlen);
+                if (B.est_cells_hint > 0 && !c.no_hint) {
+                    /* bands drift a few % per round; 50% headroom + overflow retry */
                     int64_t tight = B.est_cells_hint + B.est_cells_hint / 2 + B.qlen;
-                    if (tight < arena_est[i]) arena_est[i] = tight;
+                    if (tight < (*c.arena_est)[i]) (*c.arena_est)[i] = tight;
                 }
-                if (min_est && min_est[i] > arena_est[i]) arena_est[i] = min_est[i];
+                if (c.min_est && c.min_est[i] > (*c.arena_est)[i]) (*c.arena_est)[i] = c.min_est[i];
                 int inf_min;
                 int span = B.ab->abg->node_id_to_index[B.end_node_id] - B.ab->abg->node_id_to_index[B.beg_node_id] + 1;
-                pick_width(B.abpt, B.qlen, span, &bits_v[i], &inf_min);
-                packs[i].jb.inf_min = inf_min;
-                packs[i].cigar_off = 0;
+                pick_width(B.abpt, B.qlen, span, &(*c.bits_v)[i], &inf_min);
+                (*c.packs)[i].jb.inf_min = inf_min;
+                (*c.packs)[i].cigar_off = 0;
             }
         };
-        std::vector<std::thread> ts;
-        for (int t = 1; t < nthr; ++t) ts.emplace_back(worker);
-        worker();
-        for (auto &t : ts) t.join();
+        int nthr = abamd_pool_size();
+        if (nthr > n_jobs) nthr = n_jobs;
+        abamd_pool_run(worker, &ctx, nthr);
     }
     int bits_max = 16;
     for (int i = 0; i < n_jobs; ++i) if (bits_v[i] > bits_max) bits_max = bits_v[i];
@@ -479,21 +489,24 @@ static int prepare_internal(BatchJob *batch, int n_jobs, int slot, const int64_t
     stage.ensure(total);
     memcpy(stage.p + o_mat, abpt->mat, mat_bytes);
     {
-        int nthr = (int)std::thread::hardware_concurrency();
-        if (nthr < 1) nthr = 1;
-        if (nthr > 64) nthr = 64;
-        std::atomic<int> next{0};
-        auto worker = [&]() {
+        struct StageCtx {
+            std::atomic<int> next{0};
+            PinnedBuf *stage; int n_jobs;
+            std::vector<HostBuf> *jbufs;
+            std::vector<JobPack> *packs;
+        } ctx;
+        ctx.stage = &stage; ctx.n_jobs = n_jobs; ctx.jbufs = &jbufs; ctx.packs = &packs;
+        auto worker = [](void *p, int, int) {
+            StageCtx &c = *(StageCtx*)p;
             for (;;) {
-                int i = next.fetch_add(1);
-                if (i >= n_jobs) break;
-                memcpy(stage.p + packs[i].slab_base, jbufs[i].v.data(), jbufs[i].used);
+                int i = c.next.fetch_add(1);
+                if (i >= c.n_jobs) break;
+                memcpy(c.stage->p + (*c.packs)[i].slab_base, (*c.jbufs)[i].v.data(), (*c.jbufs)[i].used);
             }
         };
-        std::vector<std::thread> ts;
-        for (int t = 1; t < nthr; ++t) ts.emplace_back(worker);
-        worker();
-        for (auto &t : ts) t.join();
+        int nthr = 64; /* memcpy saturates well below core count */
+        if (nthr > n_jobs) nthr = n_jobs;
+        abamd_pool_run(worker, &ctx, nthr);
     }
     /* a mixed batch runs at the widest type; widths are identical across jobs
      * of one workload in practice (same scoring paras, similar qlen) */
