@@ -8,10 +8,24 @@
  * tid = 0..nthr-1; abamd_pool_run returns when all have finished. Calls
  * must not nest (all call sites run on the driver's main thread). */
 #include <pthread.h>
+#include <stdlib.h>
 #include <unistd.h>
 #include "abamd_util.h"
 
 typedef void (*abamd_pool_fn)(void *arg, int tid, int nthr);
+
+/* spawn-per-call fallback (ABPOA_AMD_NO_POOL=1), kept for A/B timing */
+typedef struct { abamd_pool_fn fn; void *arg; int tid, nthr; } spawn_t;
+static void *spawn_tramp(void *p) {
+    spawn_t *s = (spawn_t*)p;
+    s->fn(s->arg, s->tid, s->nthr);
+    return NULL;
+}
+static int pool_disabled(void) {
+    static int v = -1;
+    if (v < 0) { const char *e = getenv("ABPOA_AMD_NO_POOL"); v = (e && *e && *e != '0') ? 1 : 0; }
+    return v;
+}
 
 static pthread_mutex_t mu = PTHREAD_MUTEX_INITIALIZER;
 static pthread_cond_t cv_work = PTHREAD_COND_INITIALIZER;
@@ -53,6 +67,18 @@ void abamd_pool_run(abamd_pool_fn fn, void *arg, int nthr) {
     int N = abamd_pool_size();
     if (nthr > N) nthr = N;
     if (nthr <= 1) { fn(arg, 0, 1); return; }
+    if (pool_disabled()) {
+        spawn_t *ss = (spawn_t*)abamd_malloc((size_t)nthr * sizeof(spawn_t));
+        pthread_t *ts = (pthread_t*)abamd_malloc((size_t)nthr * sizeof(pthread_t));
+        for (int t = 1; t < nthr; ++t) {
+            ss[t].fn = fn; ss[t].arg = arg; ss[t].tid = t; ss[t].nthr = nthr;
+            pthread_create(&ts[t], NULL, spawn_tramp, &ss[t]);
+        }
+        fn(arg, 0, nthr);
+        for (int t = 1; t < nthr; ++t) pthread_join(ts[t], NULL);
+        free(ss); free(ts);
+        return;
+    }
     pthread_mutex_lock(&mu);
     if (!started) {
         pool_n = N;

@@ -469,6 +469,7 @@ static int prepare_internal(BatchJob *batch, int n_jobs, int slot, const int64_t
             }
         };
         int nthr = abamd_pool_size();
+        if (nthr > 64) nthr = 64; /* measured: pack scales poorly past ~64 */
         if (nthr > n_jobs) nthr = n_jobs;
         abamd_pool_run(worker, &ctx, nthr);
     }
