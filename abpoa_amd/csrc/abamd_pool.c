@@ -14,7 +14,11 @@
 
 typedef void (*abamd_pool_fn)(void *arg, int tid, int nthr);
 
-/* spawn-per-call fallback (ABPOA_AMD_NO_POOL=1), kept for A/B timing */
+/* Spawn-per-call is the DEFAULT: a same-box A/B (profiles/r01_pool_ab.txt)
+ * showed the persistent broadcast pool tripling per-fold CPU time on the
+ * 254-thread GPU-box host (fold work 5255 vs 1717 cpu-s, pack 16.1 vs
+ * 12.8 s wall) — staggered thread starts behave better for these short
+ * memory-bound passes. The pool stays opt-in via ABPOA_AMD_POOL=1. */
 typedef struct { abamd_pool_fn fn; void *arg; int tid, nthr; } spawn_t;
 static void *spawn_tramp(void *p) {
     spawn_t *s = (spawn_t*)p;
@@ -23,7 +27,7 @@ static void *spawn_tramp(void *p) {
 }
 static int pool_disabled(void) {
     static int v = -1;
-    if (v < 0) { const char *e = getenv("ABPOA_AMD_NO_POOL"); v = (e && *e && *e != '0') ? 1 : 0; }
+    if (v < 0) { const char *e = getenv("ABPOA_AMD_POOL"); v = (e && *e && *e != '0') ? 0 : 1; }
     return v;
 }
 
