@@ -1,13 +1,17 @@
 /* Consensus extraction and result emission.
  *   heaviest-bundle consensus:  abpoa_output.c:476-548 (tie-breaks preserved)
  *   majority-vote consensus:    abpoa_output.c:394-452, 550-587
+ *   multi-consensus clustering: abpoa_output.c:650-1182 (candidate het
+ *     positions from the MSA matrix -> per-pair distance matrix ->
+ *     k-medoids -> per-cluster read-id bitsets -> cluster-filtered
+ *     consensus walks; exact tie-breaks and iteration orders preserved)
  *   RC-MSA:                     abpoa_output.c:106-193
  *   GFA:                        abpoa_output.c:196-295
  *   FASTA/FASTQ emission:       abpoa_output.c:297-303, 589-628
- * Multi-consensus clustering (k-medoids, -d >= 2) is not in this build yet;
- * requesting it aborts with a clear message.
  */
+#include <limits.h>
 #include <math.h>
+#include <string.h>
 #include "abpoa_amd.h"
 #include "abamd_util.h"
 
@@ -42,74 +46,542 @@ static abpoa_cons_t *allocate_cons(abpoa_cons_t *c, int n_node, int n_seq, int n
     return c;
 }
 
-/* single-cluster edge weight is just the stored weight (abpoa_output.c:330-336) */
-static int edge_weight(const abpoa_node_t *v, int edge_i) { return v->out_edge_weight[edge_i]; }
+/* ---------------- multi-consensus read clustering (-d >= 2) -------------
+ * Restates abpoa_output.c:650-1182. The distance space is the candidate
+ * heterozygous MSA columns; reads are clustered around k medoids and each
+ * cluster gets its own consensus walk with read-filtered edge weights. */
 
-/* Reverse-BFS heaviest-bundle walk. Per node choose the out-edge with max
- * weight; ties prefer the later-scanned edge when its downstream score is >=
- * (abpoa_output.c:521-535); the src node instead prefers strictly higher
- * weight with score as secondary (:507-517). */
+static int group_msa_rank(abpoa_graph_t *g, int id);
+static void msa_place_node_full(const abpoa_node_t *v, int rank, uint8_t **msa_base);
+
+typedef struct {
+    int pos, depth, var_type, count, n_uniq_alles;
+    int *n_clu_reads;          /* [m+1] reads carrying each allele */
+    int **clu_read_ids;        /* [m+1][...] read ids per allele */
+    int *read_id_to_allele_idx;/* [n_seq] -> allele index or -1 */
+    uint8_t *alle_bases;       /* [n_uniq_alles] allele codes, first-seen order */
+} het_pos_t;
+
+/* full per-read MSA matrix, gap code = m (abpoa_collect_msa, :125-148) */
+static int collect_msa_matrix(abpoa_graph_t *g, abpoa_para_t *abpt, uint8_t **msa, int n_seq) {
+    if (g->node_n <= 2) return 0;
+    abamd_set_msa_rank(g, ABPOA_SRC_NODE_ID, ABPOA_SINK_NODE_ID);
+    int msa_l = g->node_id_to_msa_rank[ABPOA_SINK_NODE_ID] - 1;
+    int i, j;
+    for (i = 0; i < n_seq; ++i) {
+        msa[i] = (uint8_t*)abamd_malloc((size_t)msa_l);
+        for (j = 0; j < msa_l; ++j) msa[i][j] = (uint8_t)abpt->m;
+    }
+    for (i = 2; i < g->node_n; ++i)
+        msa_place_node_full(&g->node[i], group_msa_rank(g, i), msa);
+    return msa_l;
+}
+
+/* a candidate position whose allele->read partition already appeared is
+ * only counted, scanning newest-first (allele_clu_exist, :650-674) */
+static int het_clu_exists(het_pos_t *hp, int n_het, int n_uniq, const int *alleles,
+                          const int *n_clu_reads, int *const *clu_read_ids) {
+    int i, j, k;
+    for (i = n_het - 1; i >= 0; --i) {
+        if (hp[i].n_uniq_alles != n_uniq) continue;
+        int exist = 1;
+        for (j = 0; j < n_uniq; ++j) {
+            int ax = hp[i].alle_bases[j], ay = alleles[j];
+            if (hp[i].n_clu_reads[ax] != n_clu_reads[ay]) { exist = 0; break; }
+            for (k = 0; k < n_clu_reads[ay]; ++k)
+                if (hp[i].clu_read_ids[ax][k] != clu_read_ids[ay][k]) { exist = 0; break; }
+            if (!exist) break;
+        }
+        if (exist) return i;
+    }
+    return -1;
+}
+
+/* candidate het columns: >= 2 alleles each covering [min_het, n_seq-min_het]
+ * reads; alleles ordered by first appearance; priorities bubble-sorted by
+ * (count desc, depth desc, SNP-before-indel) (abpoa_collect_cand_het_pos,
+ * :676-829) */
+static int collect_cand_het_pos(uint8_t **msa, int msa_l, int n_seq, int m, int min_het,
+                                het_pos_t *hp, int *prio) {
+    int n_het = 0, i, j, k;
+    min_het = AB_MAX2(2, min_het / 2);
+    int min_hom = n_seq - min_het;
+    int *depth = (int*)abamd_malloc((size_t)(m + 1) * sizeof(int));
+    int *alleles = (int*)abamd_malloc((size_t)(m + 1) * sizeof(int));
+    int *allele_to_idx = (int*)abamd_malloc((size_t)(m + 1) * sizeof(int));
+    int *first_seen = (int*)abamd_malloc((size_t)(m + 1) * sizeof(int));
+    int *n_clu_reads = (int*)abamd_malloc((size_t)(m + 1) * sizeof(int));
+    int **clu_read_ids = (int**)abamd_malloc((size_t)(m + 1) * sizeof(int*));
+    for (j = 0; j < m + 1; ++j) clu_read_ids[j] = (int*)abamd_malloc((size_t)n_seq * sizeof(int));
+    for (i = 0; i < msa_l; ++i) {
+        int n_uniq = 0, var_type = 0, total_depth = 0;
+        memset(depth, 0, (size_t)(m + 1) * sizeof(int));
+        memset(n_clu_reads, 0, (size_t)(m + 1) * sizeof(int));
+        for (j = 0; j < n_seq; ++j) {
+            if (++depth[msa[j][i]] == 1) first_seen[msa[j][i]] = j;
+        }
+        for (j = 0; j < m + 1; ++j) {
+            if (depth[j] >= min_het && depth[j] <= min_hom) {
+                alleles[n_uniq++] = j;
+                total_depth += depth[j];
+                if (j == m) var_type = 1; /* gap allele => indel column */
+            }
+        }
+        if (n_uniq < 2) continue;
+        for (j = 0; j < n_uniq - 1; ++j)
+            for (k = j + 1; k < n_uniq; ++k)
+                if (first_seen[alleles[j]] > first_seen[alleles[k]]) {
+                    int t = alleles[j]; alleles[j] = alleles[k]; alleles[k] = t;
+                }
+        for (j = 0; j < n_uniq; ++j) allele_to_idx[alleles[j]] = j;
+        for (j = 0; j < n_seq; ++j)
+            for (k = 0; k < n_uniq; ++k)
+                if (msa[j][i] == alleles[k]) {
+                    clu_read_ids[alleles[k]][n_clu_reads[alleles[k]]++] = j;
+                    break;
+                }
+        int het_i = het_clu_exists(hp, n_het, n_uniq, alleles, n_clu_reads, clu_read_ids);
+        if (het_i >= 0) {
+            hp[het_i].count++;
+            if (var_type == 0) hp[het_i].var_type = 0; /* SNP wins if any column is a SNP */
+            continue;
+        }
+        hp[n_het].pos = i;
+        hp[n_het].depth = total_depth;
+        hp[n_het].var_type = var_type;
+        hp[n_het].count = 1;
+        hp[n_het].n_uniq_alles = n_uniq;
+        hp[n_het].n_clu_reads = (int*)abamd_malloc((size_t)(m + 1) * sizeof(int));
+        hp[n_het].clu_read_ids = (int**)abamd_malloc((size_t)(m + 1) * sizeof(int*));
+        hp[n_het].read_id_to_allele_idx = (int*)abamd_malloc((size_t)n_seq * sizeof(int));
+        for (j = 0; j < n_seq; ++j) hp[n_het].read_id_to_allele_idx[j] = -1;
+        for (j = 0; j < m + 1; ++j) {
+            hp[n_het].n_clu_reads[j] = n_clu_reads[j];
+            hp[n_het].clu_read_ids[j] = (int*)abamd_malloc((size_t)n_seq * sizeof(int));
+            memcpy(hp[n_het].clu_read_ids[j], clu_read_ids[j], (size_t)n_clu_reads[j] * sizeof(int));
+            for (k = 0; k < n_clu_reads[j]; ++k)
+                hp[n_het].read_id_to_allele_idx[clu_read_ids[j][k]] = allele_to_idx[j];
+        }
+        hp[n_het].alle_bases = (uint8_t*)abamd_malloc((size_t)n_uniq);
+        for (j = 0; j < n_uniq; ++j) hp[n_het].alle_bases[j] = (uint8_t)alleles[j];
+        prio[n_het] = n_het;
+        n_het++;
+    }
+    for (i = 0; i < n_het; ++i) prio[i] = i;
+    int swapped;
+    do {
+        swapped = 0;
+        for (j = 0; j < n_het - 1; ++j) {
+            int a = prio[j], b = prio[j + 1];
+            if (hp[a].count < hp[b].count ||
+                (hp[a].count == hp[b].count && hp[a].depth < hp[b].depth) ||
+                (hp[a].count == hp[b].count && hp[a].depth == hp[b].depth && hp[a].var_type > hp[b].var_type)) {
+                prio[j] = b; prio[j + 1] = a; swapped = 1;
+            }
+        }
+    } while (swapped);
+    for (j = 0; j < m + 1; ++j) free(clu_read_ids[j]);
+    free(clu_read_ids); free(n_clu_reads); free(alleles); free(allele_to_idx);
+    free(first_seen); free(depth);
+    return n_het;
+}
+
+/* pairwise read distance over het columns only: SNP columns weigh 2x, gap
+ * columns 1x, scaled by the column's collapse count; positions where either
+ * read carries a non-candidate allele are skipped (:824-864) */
+static int **collect_dis_matrix(uint8_t **msa, int n_seq, const het_pos_t *hp, int n_het) {
+    int i, j, k, a;
+    int **dm = (int**)abamd_malloc((size_t)n_seq * sizeof(int*));
+    for (i = 0; i < n_seq; ++i) dm[i] = (int*)abamd_calloc(n_seq, sizeof(int));
+    for (i = 0; i < n_seq; ++i)
+        for (j = i + 1; j < n_seq; ++j) {
+            int dis = 0;
+            for (k = 0; k < n_het; ++k) {
+                int pos = hp[k].pos;
+                int w = hp[k].var_type == 0 ? 2 : 1;
+                uint8_t b1 = msa[i][pos], b2 = msa[j][pos];
+                int v1 = 0, v2 = 0;
+                for (a = 0; a < hp[k].n_uniq_alles; ++a) {
+                    if (b1 == hp[k].alle_bases[a]) v1 = 1;
+                    if (b2 == hp[k].alle_bases[a]) v2 = 1;
+                }
+                if (!v1 || !v2) continue;
+                if (b1 != b2) dis += w * hp[k].count;
+            }
+            dm[i][j] = dm[j][i] = dis;
+        }
+    return dm;
+}
+
+/* first two medoids: the max-distance read pair across the top het column's
+ * allele clusters (:866-891) */
+static int collect_2medoids(const het_pos_t *hp, int het_i, int **dm, int *med) {
+    int max_dis = 0, max_i = -1, max_j = -1, i, j, r1, r2;
+    for (i = 0; i < hp[het_i].n_uniq_alles - 1; ++i) {
+        int ai = hp[het_i].alle_bases[i];
+        for (j = i + 1; j < hp[het_i].n_uniq_alles; ++j) {
+            int aj = hp[het_i].alle_bases[j];
+            for (r1 = 0; r1 < hp[het_i].n_clu_reads[ai]; ++r1) {
+                int ri = hp[het_i].clu_read_ids[ai][r1];
+                for (r2 = 0; r2 < hp[het_i].n_clu_reads[aj]; ++r2) {
+                    int rj = hp[het_i].clu_read_ids[aj][r2];
+                    if (dm[ri][rj] > max_dis) { max_dis = dm[ri][rj]; max_i = ri; max_j = rj; }
+                }
+            }
+        }
+    }
+    if (max_dis > 0) { med[0] = max_i; med[1] = max_j; return 2; }
+    return 0;
+}
+
+/* partition index of a read over het columns 0..het_i (:893-902) */
+static int partition_index(const het_pos_t *hp, int het_i, int read_i) {
+    int k, idx = 0;
+    for (k = 0; k <= het_i; ++k) {
+        idx *= hp[k].n_uniq_alles + 1;
+        idx += hp[k].read_id_to_allele_idx[read_i] + 1;
+    }
+    return idx;
+}
+
+/* add one medoid: prefer a read from a partition no medoid occupies (most
+ * populated partition, then max of min-distance-to-medoids); else the
+ * max-min-distance read among the top column's allele clusters (:904-971) */
+static int collect_1medoid(const het_pos_t *hp, int het_i, int **dm, int n_seq,
+                           int *med, int n_medoids) {
+    int n_partitions = 1, k, i, j;
+    for (k = 0; k <= het_i; ++k) n_partitions *= hp[k].n_uniq_alles + 1;
+    int *pcount = (int*)abamd_calloc(n_partitions, sizeof(int));
+    for (i = 0; i < n_seq; ++i) pcount[partition_index(hp, het_i, i)] += 1;
+    int max_dis = 0, max_read_i = -1, max_ptotal = -1;
+    for (i = 0; i < n_seq; ++i) {
+        int found = 0;
+        int pi = partition_index(hp, het_i, i);
+        int ptotal = pcount[pi];
+        for (j = 0; j < n_medoids; ++j)
+            if (pi == partition_index(hp, het_i, med[j])) { found = 1; break; }
+        if (!found) {
+            int min_dis = INT_MAX;
+            for (j = 0; j < n_medoids; ++j)
+                if (dm[i][med[j]] < min_dis) min_dis = dm[i][med[j]];
+            if (ptotal > max_ptotal || (ptotal == max_ptotal && min_dis > max_dis)) {
+                max_dis = min_dis; max_read_i = i; max_ptotal = ptotal;
+            }
+        }
+    }
+    if (max_read_i == -1) {
+        for (i = 0; i < hp[het_i].n_uniq_alles; ++i) {
+            int allele = hp[het_i].alle_bases[i];
+            int r;
+            for (r = 0; r < hp[het_i].n_clu_reads[allele]; ++r) {
+                int ri = hp[het_i].clu_read_ids[allele][r];
+                int min_dis = INT_MAX, skip = 0;
+                for (j = 0; j < n_medoids; ++j) {
+                    if (med[j] == ri) { skip = 1; continue; }
+                    if (dm[ri][med[j]] < min_dis) min_dis = dm[ri][med[j]];
+                }
+                if (min_dis > max_dis && !skip) { max_dis = min_dis; max_read_i = ri; }
+            }
+        }
+    }
+    free(pcount);
+    if (max_read_i != -1) { med[n_medoids] = max_read_i; return 1; }
+    return 0;
+}
+
+static int collect_multi_medoids(const het_pos_t *hp, int het_i, int **dm, int n_seq,
+                                 int max_n_cons, int *med, int n_medoids) {
+    int n_to_collect = AB_MIN2(hp[het_i].n_uniq_alles, max_n_cons);
+    while (n_medoids < n_to_collect) {
+        int got = n_medoids == 0 ? collect_2medoids(hp, het_i, dm, med)
+                                 : collect_1medoid(hp, het_i, dm, n_seq, med, n_medoids);
+        if (got == 0) break;
+        n_medoids += got;
+    }
+    return n_medoids;
+}
+
+static int init_kmedoids(const het_pos_t *hp, const int *prio, int n_het, int **dm,
+                         int n_seq, int max_n_cons, int *med) {
+    int n_medoids = 0, het_i = 0;
+    while (n_medoids < max_n_cons) {
+        if (n_medoids == 0)
+            n_medoids += collect_multi_medoids(hp, prio[het_i], dm, n_seq, max_n_cons, med, n_medoids);
+        else
+            n_medoids += collect_1medoid(hp, prio[het_i], dm, n_seq, med, n_medoids);
+        het_i++;
+        if (het_i >= n_het) break;
+    }
+    return n_medoids;
+}
+
+/* re-pick each cluster's medoid = member with min summed distance, then sort
+ * medoids ascending (:1003-1029) */
+static void kmedoids_repick(int **dm, int max_n_cons, int **clu_reads,
+                            const int *n_clu_seqs, int *med) {
+    int i, j, k;
+    for (i = 0; i < max_n_cons; ++i) {
+        int best = INT_MAX, best_read = -1;
+        for (j = 0; j < n_clu_seqs[i]; ++j) {
+            int sum = 0, ri = clu_reads[i][j];
+            for (k = 0; k < n_clu_seqs[i]; ++k) {
+                if (j == k) continue;
+                sum += dm[ri][clu_reads[i][k]];
+            }
+            if (sum < best) { best = sum; best_read = ri; }
+        }
+        if (best_read != -1) med[i] = best_read;
+    }
+    for (i = 0; i < max_n_cons - 1; ++i)
+        for (j = i + 1; j < max_n_cons; ++j)
+            if (med[i] > med[j]) { int t = med[i]; med[i] = med[j]; med[j] = t; }
+}
+
+/* one assignment + medoid-update round; ties between medoids send the read
+ * to the currently smaller of clusters 0/1, exactly as the reference does
+ * even for k > 2 (:1031-1087) */
+static int kmedoids_update(int **dm, int n_seq, int max_n_cons, int **med,
+                           int **clu_reads, int *n_clu_seqs) {
+    int i, j;
+    int *new_med = (int*)abamd_malloc((size_t)max_n_cons * sizeof(int));
+    for (i = 0; i < max_n_cons; ++i) new_med[i] = -1;
+    memset(n_clu_seqs, 0, (size_t)max_n_cons * sizeof(int));
+    for (i = 0; i < n_seq; ++i) {
+        int min_dis = INT_MAX, min_clu = -1, tied = 0;
+        for (j = 0; j < max_n_cons; ++j) {
+            if (dm[i][(*med)[j]] < min_dis) { min_dis = dm[i][(*med)[j]]; min_clu = j; tied = 0; }
+            else if (dm[i][(*med)[j]] == min_dis) tied = 1;
+        }
+        if (min_clu == -1) continue;
+        if (tied == 1) min_clu = n_clu_seqs[0] < n_clu_seqs[1] ? 0 : 1;
+        clu_reads[min_clu][n_clu_seqs[min_clu]++] = i;
+    }
+    kmedoids_repick(dm, max_n_cons, clu_reads, n_clu_seqs, new_med);
+    int changed = 0;
+    for (i = 0; i < max_n_cons; ++i) {
+        if (new_med[i] == -1) { changed = 0; break; } /* empty cluster: stop */
+        if (new_med[i] != (*med)[i]) changed = 1;
+    }
+    free(*med);
+    *med = new_med;
+    return changed;
+}
+
+/* iterate k-medoids, shrinking k while a cluster stays below min_het reads
+ * or <80% of reads are clustered (:1089-1134) */
+static int clu_reads_kmedoids(const het_pos_t *hp, const int *prio, int n_het, int **dm,
+                              int n_seq, int min_het, int max_n_cons, uint64_t ***clu_read_ids) {
+    int i, j;
+    int *med = (int*)abamd_malloc((size_t)max_n_cons * sizeof(int));
+    int **clu_reads = (int**)abamd_malloc((size_t)max_n_cons * sizeof(int*));
+    int *n_clu_seqs = (int*)abamd_malloc((size_t)max_n_cons * sizeof(int));
+    for (i = 0; i < max_n_cons; ++i) clu_reads[i] = (int*)abamd_malloc((size_t)n_seq * sizeof(int));
+    int to_collect = max_n_cons, n_clusters = 1;
+    while (1) {
+        if (init_kmedoids(hp, prio, n_het, dm, n_seq, to_collect, med) <= 0) break;
+        int iter = 0;
+        while (1) {
+            int changed = kmedoids_update(dm, n_seq, to_collect, &med, clu_reads, n_clu_seqs);
+            if (changed == 0 || ++iter >= 10) break;
+        }
+        int n_clu = 0, n_clustered = 0;
+        for (i = 0; i < to_collect; ++i) {
+            if (n_clu_seqs[i] >= min_het) n_clu++;
+            n_clustered += n_clu_seqs[i];
+        }
+        if (n_clu != to_collect || n_clustered < (int)ceil(n_seq * 0.8)) {
+            if (--to_collect < 2) break;
+        } else { n_clusters = n_clu; break; }
+    }
+    if (n_clusters != 1) {
+        int rid_n = (n_seq - 1) / 64 + 1;
+        *clu_read_ids = (uint64_t**)abamd_malloc((size_t)n_clusters * sizeof(uint64_t*));
+        for (i = 0; i < n_clusters; ++i) {
+            (*clu_read_ids)[i] = (uint64_t*)abamd_calloc(rid_n, sizeof(uint64_t));
+            for (j = 0; j < n_clu_seqs[i]; ++j) {
+                int ri = clu_reads[i][j];
+                (*clu_read_ids)[i][ri / 64] |= 1ull << (ri & 0x3f);
+            }
+        }
+    }
+    for (i = 0; i < max_n_cons; ++i) free(clu_reads[i]);
+    free(clu_reads); free(n_clu_seqs); free(med);
+    return n_clusters;
+}
+
+/* MSA -> het positions -> distances -> k-medoids clusters
+ * (abpoa_multip_read_clu_kmedoids, :1136-1182) */
+static int multip_read_clu(abpoa_graph_t *g, abpoa_para_t *abpt, int n_seq,
+                           uint64_t ***clu_read_ids) {
+    int i, j, n_clu;
+    uint8_t **msa = (uint8_t**)abamd_malloc((size_t)n_seq * sizeof(uint8_t*));
+    int msa_l = collect_msa_matrix(g, abpt, msa, n_seq);
+    int min_w = AB_MAX2(2, (int)ceil(n_seq * abpt->min_freq));
+    het_pos_t *hp = (het_pos_t*)abamd_malloc((size_t)msa_l * sizeof(het_pos_t));
+    int *prio = (int*)abamd_malloc((size_t)msa_l * sizeof(int));
+    int n_het = collect_cand_het_pos(msa, msa_l, n_seq, abpt->m, min_w, hp, prio);
+    if (n_het < 1) n_clu = 1;
+    else {
+        int **dm = collect_dis_matrix(msa, n_seq, hp, n_het);
+        n_clu = clu_reads_kmedoids(hp, prio, n_het, dm, n_seq, min_w, abpt->max_n_cons, clu_read_ids);
+        for (i = 0; i < n_seq; ++i) free(dm[i]);
+        free(dm);
+    }
+    for (i = 0; i < n_het; ++i) {
+        free(hp[i].alle_bases);
+        for (j = 0; j < abpt->m + 1; ++j) free(hp[i].clu_read_ids[j]);
+        free(hp[i].clu_read_ids); free(hp[i].n_clu_reads); free(hp[i].read_id_to_allele_idx);
+    }
+    free(hp); free(prio);
+    for (i = 0; i < n_seq; ++i) free(msa[i]);
+    free(msa);
+    return n_clu;
+}
+
+/* ---------------- cluster-filtered edge weights (:305-374) ---------------- */
+
+static int edge_inclu_read_count(const abpoa_node_t *v, int edge_i, int cons_i,
+                                 uint64_t *const *clu_read_ids) {
+    int n = 0, i;
+    for (i = 0; i < v->read_ids_n; ++i)
+        n += ab_amd_popcnt64(v->read_ids[edge_i][i] & clu_read_ids[cons_i][i]);
+    return n;
+}
+
+static int inclu_edge_weight(const abpoa_node_t *v, int edge_i, int cons_i,
+                             uint64_t *const *clu_read_ids, int use_qv) {
+    if (use_qv == 0) return edge_inclu_read_count(v, edge_i, cons_i, clu_read_ids);
+    int w = 0, i;
+    for (i = 0; i < v->m_read; ++i) {
+        if (v->read_weight[i] > 0) {
+            uint64_t c = v->read_ids[edge_i][i / 64] & clu_read_ids[cons_i][i / 64];
+            if (c & (1ull << (i & 0x3f))) w += v->read_weight[i];
+        }
+    }
+    return w;
+}
+
+/* single-cluster edge weight is just the stored weight (abpoa_output.c:330-336) */
+static int edge_weight_clu(const abpoa_node_t *v, int edge_i, int cons_i,
+                           uint64_t *const *clu_read_ids, int use_qv, int n_clu) {
+    if (n_clu == 1) return v->out_edge_weight[edge_i];
+    return inclu_edge_weight(v, edge_i, cons_i, clu_read_ids, use_qv);
+}
+
+static int node_out_cov(const abpoa_node_t *nodes, int id, uint64_t *const *clu_read_ids,
+                        int cons_i, int n_cons) {
+    if (n_cons == 1) return nodes[id].n_read;
+    int i, cov = 0;
+    for (i = 0; i < nodes[id].out_edge_n; ++i)
+        cov += edge_inclu_read_count(&nodes[id], i, cons_i, clu_read_ids);
+    return cov;
+}
+
+/* NOTE: replicates the reference exactly, including its in-cov loop bound of
+ * nodes[0].in_edge_n (the SRC node, in-degree 0), so the in-side term is
+ * always 0 and multi-cluster coverage is effectively the out-side coverage
+ * (abpoa_node_in_cov/abpoa_node_cov, :356-374). */
+static int node_cov_clu(const abpoa_node_t *nodes, int id, uint64_t *const *clu_read_ids,
+                        int cons_i, int n_cons) {
+    if (n_cons == 1) return nodes[id].n_read;
+    int i, j, in_cov = 0;
+    for (i = 0; i < nodes[0].in_edge_n; ++i) {
+        int in_id = nodes[id].in_id[i];
+        for (j = 0; j < nodes[in_id].out_edge_n; ++j)
+            if (nodes[in_id].out_id[j] == id) {
+                in_cov += edge_inclu_read_count(&nodes[in_id], j, cons_i, clu_read_ids);
+                break;
+            }
+    }
+    int out_cov = node_out_cov(nodes, id, clu_read_ids, cons_i, n_cons);
+    return AB_MAX2(in_cov, out_cov);
+}
+
+/* Reverse-BFS heaviest-bundle walk, one pass per cluster. Per node choose
+ * the out-edge with max (cluster-filtered) weight; ties prefer the
+ * later-scanned edge when its downstream score is >= (abpoa_output.c:521-535);
+ * the src node instead prefers strictly higher weight with score as secondary
+ * (:507-517). */
 static void heaviest_bundling(abpoa_graph_t *g, abpoa_para_t *abpt, int src_id, int sink_id,
-                              const int *out_degree, abpoa_cons_t *abc) {
-    (void)abpt;
-    int i, cur;
+                              const int *out_degree, int n_clu, int read_ids_n,
+                              uint64_t **clu_read_ids, abpoa_cons_t *abc) {
+    int i, cur, cons_i;
     int n = g->node_n;
     int *deg = (int*)abamd_malloc((size_t)n * sizeof(int));
     int *score = (int*)abamd_malloc((size_t)n * sizeof(int));
-    int *max_out_id = (int*)abamd_malloc((size_t)n * sizeof(int));
-    abc->clu_n_seq[0] = abc->n_seq;
-    for (i = 0; i < abc->n_seq; ++i) abc->clu_read_ids[0][i] = i;
-
-    for (i = 0; i < n; ++i) deg[i] = out_degree[i];
-    int cap = n, sp_head = 0, sp_tail = 0;
-    int *q = (int*)abamd_malloc((size_t)cap * sizeof(int));
-    q[sp_tail++] = sink_id;
-    while (sp_head < sp_tail) {
-        cur = q[sp_head++];
-        if (cur == sink_id) {
-            max_out_id[cur] = -1;
-            score[cur] = 0;
-        } else if (cur == src_id) {
-            int max_id = -1, path_score = -1, path_max_w = -1;
-            for (i = 0; i < g->node[cur].out_edge_n; ++i) {
-                int out_id = g->node[cur].out_id[i];
-                int w = edge_weight(&g->node[cur], i);
-                if (w > path_max_w || (w == path_max_w && score[out_id] > path_score)) {
-                    max_id = out_id; path_score = score[out_id]; path_max_w = w;
-                }
-            }
-            max_out_id[cur] = max_id;
-            break;
-        } else {
-            int max_id = -1, max_w = INT32_MIN;
-            for (i = 0; i < g->node[cur].out_edge_n; ++i) {
-                int out_id = g->node[cur].out_id[i];
-                int w = edge_weight(&g->node[cur], i);
-                if (max_w < w) { max_w = w; max_id = out_id; }
-                else if (max_w == w && score[max_id] <= score[out_id]) max_id = out_id;
-            }
-            score[cur] = max_w + score[max_id];
-            max_out_id[cur] = max_id;
+    int **max_out_id = (int**)abamd_malloc((size_t)n_clu * sizeof(int*));
+    for (i = 0; i < n_clu; ++i) max_out_id[i] = (int*)abamd_malloc((size_t)n * sizeof(int));
+    if (n_clu == 1) {
+        abc->clu_n_seq[0] = abc->n_seq;
+        for (i = 0; i < abc->n_seq; ++i) abc->clu_read_ids[0][i] = i;
+    } else {
+        for (cons_i = 0; cons_i < n_clu; ++cons_i) {
+            int cnt = 0, r;
+            for (i = 0; i < read_ids_n; ++i) cnt += ab_amd_popcnt64(clu_read_ids[cons_i][i]);
+            abc->clu_n_seq[cons_i] = cnt;
+            for (r = 0, cnt = 0; r < abc->n_seq; ++r)
+                if (clu_read_ids[cons_i][r / 64] & (1ull << (r & 0x3f)))
+                    abc->clu_read_ids[cons_i][cnt++] = r;
         }
-        for (i = 0; i < g->node[cur].in_edge_n; ++i) {
-            int in_id = g->node[cur].in_id[i];
-            if (--deg[in_id] == 0) q[sp_tail++] = in_id;
+    }
+
+    int cap = n;
+    int *q = (int*)abamd_malloc((size_t)cap * sizeof(int));
+    for (cons_i = 0; cons_i < n_clu; ++cons_i) {
+        int sp_head = 0, sp_tail = 0;
+        for (i = 0; i < n; ++i) deg[i] = out_degree[i];
+        q[sp_tail++] = sink_id;
+        while (sp_head < sp_tail) {
+            cur = q[sp_head++];
+            if (cur == sink_id) {
+                max_out_id[cons_i][cur] = -1;
+                score[cur] = 0;
+            } else if (cur == src_id) {
+                int max_id = -1, path_score = -1, path_max_w = -1;
+                for (i = 0; i < g->node[cur].out_edge_n; ++i) {
+                    int out_id = g->node[cur].out_id[i];
+                    int w = edge_weight_clu(&g->node[cur], i, cons_i, clu_read_ids, abpt->use_qv, n_clu);
+                    if (w > path_max_w || (w == path_max_w && score[out_id] > path_score)) {
+                        max_id = out_id; path_score = score[out_id]; path_max_w = w;
+                    }
+                }
+                max_out_id[cons_i][cur] = max_id;
+                break;
+            } else {
+                int max_id = -1, max_w = INT32_MIN;
+                for (i = 0; i < g->node[cur].out_edge_n; ++i) {
+                    int out_id = g->node[cur].out_id[i];
+                    int w = edge_weight_clu(&g->node[cur], i, cons_i, clu_read_ids, abpt->use_qv, n_clu);
+                    if (max_w < w) { max_w = w; max_id = out_id; }
+                    else if (max_w == w && score[max_id] <= score[out_id]) max_id = out_id;
+                }
+                score[cur] = max_w + score[max_id];
+                max_out_id[cons_i][cur] = max_id;
+            }
+            for (i = 0; i < g->node[cur].in_edge_n; ++i) {
+                int in_id = g->node[cur].in_id[i];
+                if (--deg[in_id] == 0) q[sp_tail++] = in_id;
+            }
         }
     }
     /* walk src -> sink through the chosen edges (abpoa_set_hb_cons, :376-392) */
-    int j = 0;
-    cur = max_out_id[src_id];
-    while (cur != sink_id) {
-        abc->cons_node_ids[0][j] = cur;
-        abc->cons_base[0][j] = g->node[cur].base;
-        abc->cons_cov[0][j] = g->node[cur].n_read;
-        abc->cons_phred_score[0][j] = cons_phred_score(abc->cons_cov[0][j], abc->clu_n_seq[0]);
-        ++j;
-        cur = max_out_id[cur];
+    abc->n_cons = n_clu;
+    for (cons_i = 0; cons_i < n_clu; ++cons_i) {
+        int j = 0;
+        cur = max_out_id[cons_i][src_id];
+        while (cur != sink_id) {
+            abc->cons_node_ids[cons_i][j] = cur;
+            abc->cons_base[cons_i][j] = g->node[cur].base;
+            abc->cons_cov[cons_i][j] = node_cov_clu(g->node, cur, clu_read_ids, cons_i, n_clu);
+            abc->cons_phred_score[cons_i][j] = cons_phred_score(abc->cons_cov[cons_i][j], abc->clu_n_seq[cons_i]);
+            ++j;
+            cur = max_out_id[cons_i][cur];
+        }
+        abc->cons_len[cons_i] = j;
     }
-    abc->cons_len[0] = j;
-    abc->n_cons = 1;
-    free(deg); free(score); free(max_out_id); free(q);
+    free(deg); free(score); free(q);
+    for (i = 0; i < n_clu; ++i) free(max_out_id[i]);
+    free(max_out_id);
 }
 
 /* node msa rank = max over its aligned group (abpoa_output.c:136-144) */
@@ -122,48 +594,68 @@ static int group_msa_rank(abpoa_graph_t *g, int id) {
     return rank;
 }
 
-/* majority-vote consensus over MSA columns (abpoa_output.c:394-452, 550-587) */
-static void most_frequent(abpoa_graph_t *g, abpoa_para_t *abpt, int src_id, int sink_id, abpoa_cons_t *abc) {
+/* majority-vote consensus over MSA columns, one pass per cluster
+ * (abpoa_output.c:394-452, 550-587) */
+static void most_frequent(abpoa_graph_t *g, abpoa_para_t *abpt, int src_id, int sink_id,
+                          int n_clu, int read_ids_n, uint64_t **clu_read_ids, abpoa_cons_t *abc) {
     int use_span = abpt->sub_aln;
     abamd_set_msa_rank(g, src_id, sink_id);
-    int m = abpt->m, i, j;
+    int m = abpt->m, i, j, cons_i;
     int msa_l = g->node_id_to_msa_rank[sink_id] - 1;
-    int *rc_weight = (int*)abamd_calloc((size_t)msa_l * m, sizeof(int));
+    int *rc_weight = (int*)abamd_calloc((size_t)n_clu * msa_l * m, sizeof(int));
     int *msa_node_id = (int*)abamd_calloc((size_t)msa_l * m, sizeof(int));
-    for (i = 0; i < msa_l; ++i) rc_weight[i * m + m - 1] = abc->n_seq;
-    abc->n_cons = 1;
-    abc->clu_n_seq[0] = abc->n_seq;
-    for (i = 0; i < abc->n_seq; ++i) abc->clu_read_ids[0][i] = i;
+    abc->n_cons = n_clu;
+    if (n_clu == 1) {
+        abc->clu_n_seq[0] = abc->n_seq;
+        for (i = 0; i < abc->n_seq; ++i) abc->clu_read_ids[0][i] = i;
+    } else {
+        for (cons_i = 0; cons_i < n_clu; ++cons_i) {
+            int cnt = 0, r;
+            for (i = 0; i < read_ids_n; ++i) cnt += ab_amd_popcnt64(clu_read_ids[cons_i][i]);
+            abc->clu_n_seq[cons_i] = cnt;
+            for (r = 0, cnt = 0; r < abc->n_seq; ++r)
+                if (clu_read_ids[cons_i][r / 64] & (1ull << (r & 0x3f)))
+                    abc->clu_read_ids[cons_i][cnt++] = r;
+        }
+    }
+    for (cons_i = 0; cons_i < n_clu; ++cons_i)
+        for (i = 0; i < msa_l; ++i)
+            rc_weight[(cons_i * (size_t)msa_l + i) * m + m - 1] = abc->clu_n_seq[cons_i];
 
-    /* per-column per-base weights; the gap count starts at n_seq and is
-     * decremented by every base weight placed in the column (:427-452) */
+    /* per-column per-base weights; the gap count starts at clu_n_seq and is
+     * decremented by every base weight placed in the column
+     * (abpoa_set_row_column_weight, :427-452) */
     for (i = 2; i < g->node_n; ++i) {
         int rank = group_msa_rank(g, i);
-        int node_w = g->node[i].n_read; /* n_clu==1: out coverage = n_read */
         msa_node_id[(rank - 1) * m + g->node[i].base] = i;
-        rc_weight[(rank - 1) * m + g->node[i].base] = node_w;
-        rc_weight[(rank - 1) * m + m - 1] -= node_w;
-    }
-    int cons_l = 0;
-    for (i = 0; i < msa_l; ++i) {
-        int max_c = 0, total_c = 0, max_base = m, gap_c, c;
-        for (j = 0; j < m - 1; ++j) {
-            c = rc_weight[i * m + j];
-            if (c > max_c) { max_c = c; max_base = j; }
-            total_c += c;
-        }
-        if (use_span) gap_c = g->node[msa_node_id[i * m + max_base]].n_span_read - total_c;
-        else gap_c = abc->clu_n_seq[0] - total_c;
-        if (max_c >= gap_c) {
-            int cur_id = msa_node_id[i * m + max_base];
-            abc->cons_node_ids[0][cons_l] = cur_id;
-            abc->cons_base[0][cons_l] = (uint8_t)max_base;
-            abc->cons_cov[0][cons_l] = max_c;
-            abc->cons_phred_score[0][cons_l] = cons_phred_score(max_c, abc->clu_n_seq[0]);
-            cons_l++;
+        for (cons_i = 0; cons_i < n_clu; ++cons_i) {
+            int node_w = node_out_cov(g->node, i, clu_read_ids, cons_i, n_clu);
+            rc_weight[(cons_i * (size_t)msa_l + rank - 1) * m + g->node[i].base] = node_w;
+            rc_weight[(cons_i * (size_t)msa_l + rank - 1) * m + m - 1] -= node_w;
         }
     }
-    abc->cons_len[0] = cons_l;
+    for (cons_i = 0; cons_i < n_clu; ++cons_i) {
+        int cons_l = 0;
+        for (i = 0; i < msa_l; ++i) {
+            int max_c = 0, total_c = 0, max_base = m, gap_c, c;
+            for (j = 0; j < m - 1; ++j) {
+                c = rc_weight[(cons_i * (size_t)msa_l + i) * m + j];
+                if (c > max_c) { max_c = c; max_base = j; }
+                total_c += c;
+            }
+            if (use_span) gap_c = g->node[msa_node_id[i * m + max_base]].n_span_read - total_c;
+            else gap_c = abc->clu_n_seq[cons_i] - total_c;
+            if (max_c >= gap_c) {
+                int cur_id = msa_node_id[i * m + max_base];
+                abc->cons_node_ids[cons_i][cons_l] = cur_id;
+                abc->cons_base[cons_i][cons_l] = (uint8_t)max_base;
+                abc->cons_cov[cons_i][cons_l] = max_c;
+                abc->cons_phred_score[cons_i][cons_l] = cons_phred_score(max_c, abc->clu_n_seq[cons_i]);
+                cons_l++;
+            }
+        }
+        abc->cons_len[cons_i] = cons_l;
+    }
     free(rc_weight); free(msa_node_id);
 }
 
@@ -171,16 +663,26 @@ void abpoa_generate_consensus(abpoa_t *ab, abpoa_para_t *abpt) {
     if (ab->abg->is_called_cons == 1) return;
     abpoa_graph_t *g = ab->abg;
     if (g->node_n <= 2) return;
-    if (abpt->max_n_cons > 1)
-        abamd_fatal("abpoa_generate_consensus", "multi-consensus clustering (-d >= 2) is not implemented in abpoa_amd yet");
     int i, *out_degree = (int*)abamd_malloc((size_t)g->node_n * sizeof(int));
     for (i = 0; i < g->node_n; ++i) out_degree[i] = g->node[i].out_edge_n;
+    int n_seq = ab->abs->n_seq;
+    int read_ids_n = (n_seq - 1) / 64 + 1;
+    int n_clu = 1;
+    uint64_t **clu_read_ids = NULL;
+    if (abpt->max_n_cons > 1)
+        n_clu = multip_read_clu(g, abpt, n_seq, &clu_read_ids);
     abpoa_cons_t *abc = ab->abc;
-    allocate_cons(abc, g->node_n, ab->abs->n_seq, 1);
+    allocate_cons(abc, g->node_n, n_seq, n_clu);
     if (abpt->cons_algrm == ABPOA_HB)
-        heaviest_bundling(g, abpt, ABPOA_SRC_NODE_ID, ABPOA_SINK_NODE_ID, out_degree, abc);
+        heaviest_bundling(g, abpt, ABPOA_SRC_NODE_ID, ABPOA_SINK_NODE_ID, out_degree,
+                          n_clu, read_ids_n, clu_read_ids, abc);
     else
-        most_frequent(g, abpt, ABPOA_SRC_NODE_ID, ABPOA_SINK_NODE_ID, abc);
+        most_frequent(g, abpt, ABPOA_SRC_NODE_ID, ABPOA_SINK_NODE_ID,
+                      n_clu, read_ids_n, clu_read_ids, abc);
+    if (n_clu > 1) {
+        for (i = 0; i < n_clu; ++i) free(clu_read_ids[i]);
+        free(clu_read_ids);
+    }
     g->is_called_cons = 1;
     free(out_degree);
 }
