@@ -123,3 +123,35 @@ def test_gpu_align_modes_vs_oracle(gpu_bin, cputest_bin, oracle_env, tmp_path, o
         gpu = run_stdout([gpu_bin, str(fa)] + opts + extra)
         cpu = run_stdout([cputest_bin, str(fa)] + opts + extra, env=oracle_env)
         assert gpu == cpu, "GPU/oracle divergence opts=%r extra=%r" % (opts, extra)
+
+
+def test_gpu_multicons_vs_oracle(gpu_bin, cputest_bin, oracle_env, tmp_path):
+    """Multi-consensus (-d >= 2): GPU DP + k-medoids clustering end-to-end
+    must match the oracle-injected host pipeline byte-for-byte."""
+    import random
+    rng = random.Random(11)
+    L = 500
+    base = "".join(rng.choice("ACGT") for _ in range(L))
+    hapB = list(base)
+    for p in rng.sample(range(10, L - 10), 6):
+        hapB[p] = rng.choice([c for c in "ACGT" if c != hapB[p]])
+    haps = [base, "".join(hapB)]
+
+    def noisy(h):
+        out = []
+        for ch in h:
+            r = rng.random()
+            if r < 0.02: out.append(rng.choice("ACGT"))
+            elif r < 0.03: pass
+            elif r < 0.04: out.extend((ch, rng.choice("ACGT")))
+            else: out.append(ch)
+        return "".join(out)
+
+    fa = tmp_path / "dip.fa"
+    with open(fa, "w") as f:
+        for i in range(24):
+            f.write(">r%d\n%s\n" % (i, noisy(haps[i % 2])))
+    for opts in (["-d2"], ["-d2", "-r1"], ["-d3"], ["-d2", "-a1"]):
+        gpu = run_stdout([gpu_bin, str(fa)] + opts)
+        cpu = run_stdout([cputest_bin, str(fa)] + opts, env=oracle_env)
+        assert gpu == cpu, "GPU/oracle multicons divergence opts=%r" % (opts,)
