@@ -237,6 +237,14 @@ static int find_aligned_with_base(abpoa_graph_t *g, int node_id, uint8_t base) {
     return -1;
 }
 
+/* non-static entries for the graph-restore path (abamd_restore.c) */
+int abamd_get_aligned_id(abpoa_graph_t *g, int node_id, uint8_t base) {
+    return find_aligned_with_base(g, node_id, base);
+}
+void abamd_add_aligned_pair(abpoa_graph_t *g, int node_id, int new_id) {
+    add_aligned_pair(g, node_id, new_id);
+}
+
 /* ---------------- topological machinery ---------------- */
 
 /* simple int FIFO over a thread-local reusable buffer: the topological

@@ -1,5 +1,6 @@
 /* MSA drivers: per-set orchestration of align -> fold -> consensus.
- * Mirrors abpoa_align.c:313-540 (abpoa_poa / abpoa_msa / abpoa_msa1).
+ * Mirrors abpoa_align.c:313-540 (abpoa_poa / abpoa_msa / abpoa_msa1),
+ * including incremental graph restore (-i, abamd_restore.c).
  * Minimizer seeding / guide-tree mode (-S/-p) is not in this build; requesting
  * it aborts with a clear message (the reference default is seeding OFF,
  * abpoa_align.c:146). */
@@ -58,8 +59,6 @@ static int abamd_poa(abpoa_t *ab, abpoa_para_t *abpt, uint8_t **seqs, int **weig
 static void check_unsupported_modes(abpoa_para_t *abpt) {
     if ((abpt->disable_seeding == 0 || abpt->progressive_poa) && abpt->align_mode == ABPOA_GLOBAL_MODE)
         abamd_fatal("abpoa_amd", "minimizer seeding / progressive POA (-S/-p) is not implemented in this build");
-    if (abpt->incr_fn)
-        abamd_fatal("abpoa_amd", "incremental MSA (-i) is not implemented in this build");
 }
 
 /* library entry: align a set supplied as arrays (abpoa_msa, abpoa_align.c:402-472) */
@@ -68,7 +67,12 @@ int abpoa_msa(abpoa_t *ab, abpoa_para_t *abpt, int n_seq, char **seq_names, int 
     if (n_seq <= 0) return 0;
     check_unsupported_modes(abpt);
     abpoa_seq_t *abs = ab->abs;
-    if (abs->n_seq <= 0) abpoa_reset(ab, abpt, 1024);
+    if (abs->n_seq <= 0) {
+        abpoa_reset(ab, abpt, 1024);
+        if (abpt->incr_fn) abpoa_restore_graph(ab, abpt); /* abpoa_align.c:406-412 */
+    } else if (abpt->incr_fn) {
+        fprintf(stderr, "[abpoa_msa] Graph already exists, but incr_fn is also provided. Not restoring graph from file.\n");
+    }
     int i, j, exist_n_seq = abs->n_seq;
     abs->n_seq += n_seq; abamd_realloc_seq(abs);
     if (seq_names)
@@ -110,6 +114,7 @@ int abpoa_msa1(abpoa_t *ab, abpoa_para_t *abpt, char *read_fn, FILE *out_fp) {
     if (!abpt->out_msa && !abpt->out_cons && !abpt->out_gfa) return 0;
     check_unsupported_modes(abpt);
     abpoa_reset(ab, abpt, 1024);
+    if (abpt->incr_fn) abpoa_restore_graph(ab, abpt); /* abpoa_align.c:477 */
     abpoa_seq_t *abs = ab->abs;
     int exist_n_seq = abs->n_seq;
 
@@ -141,8 +146,3 @@ int abpoa_msa1(abpoa_t *ab, abpoa_para_t *abpt, char *read_fn, FILE *out_fp) {
     return 0;
 }
 
-abpoa_t *abpoa_restore_graph(abpoa_t *ab, abpoa_para_t *abpt) {
-    (void)abpt;
-    abamd_fatal("abpoa_restore_graph", "incremental graph restore is not implemented in abpoa_amd yet");
-    return ab;
-}
