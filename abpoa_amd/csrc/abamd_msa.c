@@ -1,9 +1,7 @@
 /* MSA drivers: per-set orchestration of align -> fold -> consensus.
  * Mirrors abpoa_align.c:313-540 (abpoa_poa / abpoa_msa / abpoa_msa1),
- * including incremental graph restore (-i, abamd_restore.c).
- * Minimizer seeding / guide-tree mode (-S/-p) is not in this build; requesting
- * it aborts with a clear message (the reference default is seeding OFF,
- * abpoa_align.c:146). */
+ * including incremental graph restore (-i, abamd_restore.c) and the
+ * minimizer-seeded / progressive guide-tree path (-S / -p, abamd_seed.c). */
 #include "abpoa_amd.h"
 #include "abamd_util.h"
 
@@ -56,16 +54,26 @@ static int abamd_poa(abpoa_t *ab, abpoa_para_t *abpt, uint8_t **seqs, int **weig
     return 0;
 }
 
-static void check_unsupported_modes(abpoa_para_t *abpt) {
-    if ((abpt->disable_seeding == 0 || abpt->progressive_poa) && abpt->align_mode == ABPOA_GLOBAL_MODE)
-        abamd_fatal("abpoa_amd", "minimizer seeding / progressive POA (-S/-p) is not implemented in this build");
+/* seeded/progressive path: guide tree + per-pair anchors, then
+ * anchor-windowed POA (abpoa_align.c:443-465) */
+static void abamd_seeded_poa(abpoa_t *ab, abpoa_para_t *abpt, uint8_t **seqs, int **weights,
+                             int *seq_lens, int exist_n_seq, int n_seq, int max_len) {
+    int *tpos_to_node_id = (int*)abamd_calloc((size_t)(max_len > 0 ? max_len : 1), sizeof(int));
+    int *qpos_to_node_id = (int*)abamd_calloc((size_t)(max_len > 0 ? max_len : 1), sizeof(int));
+    int *read_id_map = (int*)abamd_malloc((size_t)n_seq * sizeof(int));
+    abamd_u64v_t par_anchors = {0, 0, 0};
+    int *par_c = (int*)abamd_calloc(n_seq, sizeof(int));
+    abamd_build_guide_tree_partition(seqs, seq_lens, n_seq, abpt, read_id_map, &par_anchors, par_c);
+    abamd_anchor_poa(ab, abpt, seqs, weights, seq_lens, par_anchors, par_c,
+                     tpos_to_node_id, qpos_to_node_id, read_id_map, exist_n_seq, n_seq);
+    free(read_id_map); free(tpos_to_node_id); free(qpos_to_node_id); free(par_c);
+    if (par_anchors.m > 0) free(par_anchors.a);
 }
 
 /* library entry: align a set supplied as arrays (abpoa_msa, abpoa_align.c:402-472) */
 int abpoa_msa(abpoa_t *ab, abpoa_para_t *abpt, int n_seq, char **seq_names, int *seq_lens,
               uint8_t **seqs, int **qual_weights, FILE *out_fp) {
     if (n_seq <= 0) return 0;
-    check_unsupported_modes(abpt);
     abpoa_seq_t *abs = ab->abs;
     if (abs->n_seq <= 0) {
         abpoa_reset(ab, abpt, 1024);
@@ -78,6 +86,8 @@ int abpoa_msa(abpoa_t *ab, abpoa_para_t *abpt, int n_seq, char **seq_names, int 
     if (seq_names)
         for (i = 0; i < n_seq; ++i)
             abamd_cpy_str(&abs->name[exist_n_seq + i], seq_names[i], (int)strlen(seq_names[i]));
+    int max_len = 0;
+    for (i = 0; i < n_seq; ++i) if (seq_lens[i] > max_len) max_len = seq_lens[i];
     int **weights = (int**)abamd_malloc((size_t)n_seq * sizeof(int*));
     for (i = 0; i < n_seq; ++i) {
         weights[i] = (int*)abamd_malloc((size_t)seq_lens[i] * sizeof(int));
@@ -87,7 +97,10 @@ int abpoa_msa(abpoa_t *ab, abpoa_para_t *abpt, int n_seq, char **seq_names, int 
             for (j = 0; j < seq_lens[i]; ++j) weights[i][j] = 1;
         }
     }
-    abamd_poa(ab, abpt, seqs, weights, seq_lens, exist_n_seq, n_seq);
+    if ((abpt->disable_seeding && abpt->progressive_poa == 0) || abpt->align_mode != ABPOA_GLOBAL_MODE)
+        abamd_poa(ab, abpt, seqs, weights, seq_lens, exist_n_seq, n_seq);
+    else
+        abamd_seeded_poa(ab, abpt, seqs, weights, seq_lens, exist_n_seq, n_seq, max_len);
     abpoa_output(ab, abpt, out_fp);
     for (i = 0; i < n_seq; ++i) free(weights[i]);
     free(weights);
@@ -112,7 +125,6 @@ static void sort_seq_by_length(abpoa_seq_t *abs, int exist_n_seq, int n_seq) {
 /* CLI entry: one input file -> one MSA/consensus (abpoa_msa1, abpoa_align.c:474-540) */
 int abpoa_msa1(abpoa_t *ab, abpoa_para_t *abpt, char *read_fn, FILE *out_fp) {
     if (!abpt->out_msa && !abpt->out_cons && !abpt->out_gfa) return 0;
-    check_unsupported_modes(abpt);
     abpoa_reset(ab, abpt, 1024);
     if (abpt->incr_fn) abpoa_restore_graph(ab, abpt); /* abpoa_align.c:477 */
     abpoa_seq_t *abs = ab->abs;
@@ -123,6 +135,10 @@ int abpoa_msa1(abpoa_t *ab, abpoa_para_t *abpt, char *read_fn, FILE *out_fp) {
     abamd_fx_close(fx);
 
     if (abpt->sort_input_seq) sort_seq_by_length(abs, exist_n_seq, n_seq);
+
+    int max_len = 0;
+    for (i = 0; i < abs->n_seq; ++i)
+        if (abs->seq[i].l > max_len) max_len = abs->seq[i].l;
 
     uint8_t **seqs = (uint8_t**)abamd_malloc((size_t)n_seq * sizeof(uint8_t*));
     int *seq_lens = (int*)abamd_malloc((size_t)n_seq * sizeof(int));
@@ -139,7 +155,10 @@ int abpoa_msa1(abpoa_t *ab, abpoa_para_t *abpt, char *read_fn, FILE *out_fp) {
             for (j = 0; j < seq_lens[i]; ++j) weights[i][j] = 1;
         }
     }
-    abamd_poa(ab, abpt, seqs, weights, seq_lens, exist_n_seq, n_seq);
+    if ((abpt->disable_seeding && abpt->progressive_poa == 0) || abpt->align_mode != ABPOA_GLOBAL_MODE)
+        abamd_poa(ab, abpt, seqs, weights, seq_lens, exist_n_seq, n_seq);
+    else
+        abamd_seeded_poa(ab, abpt, seqs, weights, seq_lens, exist_n_seq, n_seq, max_len);
     abpoa_output(ab, abpt, out_fp);
     for (i = 0; i < n_seq; ++i) { free(seqs[i]); free(weights[i]); }
     free(seqs); free(weights); free(seq_lens);

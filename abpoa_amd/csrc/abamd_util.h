@@ -72,6 +72,19 @@ double abamd_realtime(void);
 double abamd_cputime(void);
 double abamd_peakrss(void);
 
+/* minimizer seeding + anchor-windowed POA (abamd_seed.c); the functions are
+ * declared only when abpoa_amd.h's types are visible */
+typedef struct { size_t n, m; uint64_t *a; } abamd_u64v_t;
+#ifdef ABPOA_AMD_H
+int abamd_build_guide_tree_partition(uint8_t **seqs, int *seq_lens, int n_seq,
+                                     abpoa_para_t *abpt, int *read_id_map,
+                                     abamd_u64v_t *par_anchors, int *par_c);
+int abamd_anchor_poa(abpoa_t *ab, abpoa_para_t *abpt, uint8_t **seqs, int **weights,
+                     int *seq_lens, abamd_u64v_t par_anchors, int *par_c,
+                     int *tpos_to_node_id, int *qpos_to_node_id, int *read_id_map,
+                     int exist_n_seq, int n_seq);
+#endif
+
 /* persistent host thread pool (abamd_pool.c): fn(arg, tid, nthr) runs on
  * workers tid = 0..nthr-1; returns when all are done. Calls must not nest. */
 typedef void (*abamd_pool_fn)(void *arg, int tid, int nthr);

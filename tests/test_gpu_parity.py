@@ -155,3 +155,16 @@ def test_gpu_multicons_vs_oracle(gpu_bin, cputest_bin, oracle_env, tmp_path):
         gpu = run_stdout([gpu_bin, str(fa)] + opts)
         cpu = run_stdout([cputest_bin, str(fa)] + opts, env=oracle_env)
         assert gpu == cpu, "GPU/oracle multicons divergence opts=%r" % (opts,)
+
+
+def test_gpu_seeding_vs_oracle(gpu_bin, cputest_bin, oracle_env, tmp_path):
+    """Seeded (-S) and progressive (-p) POA through the GPU subgraph kernels
+    end-to-end vs the oracle-injected host pipeline."""
+    fa = tmp_path / "s.fa"
+    subprocess.run(["python3", os.path.join(ROOT, "tests", "make_synth.py"), str(fa),
+                    "--seed", "4", "--len", "5000", "--depth", "8"],
+                   check=True, stderr=subprocess.DEVNULL)
+    for opts in (["-S"], ["-S", "-p"], ["-S", "-n", "100", "-r1"]):
+        gpu = run_stdout([gpu_bin, str(fa)] + opts)
+        cpu = run_stdout([cputest_bin, str(fa)] + opts, env=oracle_env)
+        assert gpu == cpu, "GPU/oracle seeding divergence opts=%r" % (opts,)
