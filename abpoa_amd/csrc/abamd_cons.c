@@ -880,11 +880,87 @@ void abpoa_output(abpoa_t *ab, abpoa_para_t *abpt, FILE *fp) {
         if (abpt->out_msa) abpoa_output_rc_msa(ab, abpt, fp);
         else if (abpt->out_cons) abpoa_output_fx_consensus(ab, abpt, fp);
     }
-    if (abpt->out_pog)
-        fprintf(stderr, "[abpoa_amd] graph plotting (--out-pog) is not implemented in this build; skipping.\n");
+    if (abpt->out_pog) abpoa_dump_pog(ab, abpt);
 }
 
+/* graphviz DOT dump + `dot` render (abpoa_plot.c:33-123): the .dot file is
+ * byte-identical to the reference's; rendering shells out to graphviz and
+ * fails with the same message when it is unavailable */
 void abpoa_dump_pog(abpoa_t *ab, abpoa_para_t *abpt) {
-    (void)ab; (void)abpt;
-    fprintf(stderr, "[abpoa_amd] abpoa_dump_pog is not implemented in this build.\n");
+    const char *PROG = "abpoa"; int font_size = 24;
+    abpoa_graph_t *g = ab->abg;
+    if (g->is_topological_sorted == 0) abpoa_topological_sort(g, abpt);
+    static const char node_color[5][10] = {"pink1", "red1", "gold2", "seagreen4", "gray"};
+    float node_width = 1;
+    const char *rankdir = "LR", *node_style = "filled", *node_fixedsize = "true", *node_shape = "circle";
+    int show_aligned_mismatch = 1;
+
+    int i, j, id, index, out_id; char base;
+    char **node_label = (char**)abamd_malloc((size_t)g->node_n * sizeof(char*));
+    for (i = 0; i < g->node_n; ++i) node_label[i] = (char*)abamd_malloc(128);
+
+    char *dot_fn = (char*)abamd_malloc(strlen(abpt->out_pog) + 10);
+    strcpy(dot_fn, abpt->out_pog);
+    strcat(dot_fn, ".dot");
+    FILE *fp = fopen(dot_fn, "w");
+    if (!fp) abamd_fatal("abpoa_dump_pog", "cannot open %s", dot_fn);
+    fprintf(fp, "// %s graph dot file.\n// %d nodes.\n", PROG, g->node_n);
+    fprintf(fp, "digraph ABPOA_graph {\n\tgraph [rankdir=\"%s\"];\n\tnode [width=%f, style=%s, fixedsize=%s, shape=%s];\n",
+            rankdir, node_width, node_style, node_fixedsize, node_shape);
+    for (i = 0; i < g->node_n; ++i) {
+        id = g->index_to_node_id[i];
+        index = i;
+        if (id == ABPOA_SRC_NODE_ID) {
+            base = 'S';
+            sprintf(node_label[id], "\"%c\n%d\"", base, index);
+            fprintf(fp, "%s [color=%s, fontsize=%d]\n", node_label[id], node_color[4], font_size);
+        } else if (id == ABPOA_SINK_NODE_ID) {
+            base = 'E';
+            sprintf(node_label[id], "\"%c\n%d\"", base, index);
+            fprintf(fp, "%s [color=%s, fontsize=%d]\n", node_label[id], node_color[4], font_size);
+        } else {
+            base = ab_amd_nt256_table[g->node[id].base];
+            sprintf(node_label[id], "\"%c\n%d\"", base, index);
+            fprintf(fp, "%s [color=%s, fontsize=%d]\n", node_label[id], node_color[g->node[id].base], font_size);
+        }
+    }
+    int x_index = -1;
+    for (i = 0; i < g->node_n; ++i) {
+        id = g->index_to_node_id[i];
+        for (j = 0; j < g->node[id].out_edge_n; ++j) {
+            out_id = g->node[id].out_id[j];
+            fprintf(fp, "\t%s -> %s [label=\"%d\", fontsize=20, fontcolor=red, penwidth=%d]\n",
+                    node_label[id], node_label[out_id], g->node[id].out_edge_weight[j], g->node[id].out_edge_weight[j] + 1);
+        }
+        if (g->node[id].aligned_node_n > 0) {
+            fprintf(fp, "\t{rank=same; %s ", node_label[id]);
+            for (j = 0; j < g->node[id].aligned_node_n; ++j)
+                fprintf(fp, "%s ", node_label[g->node[id].aligned_node_id[j]]);
+            fprintf(fp, "};\n");
+            if (show_aligned_mismatch) {
+                if (i > x_index) {
+                    x_index = i;
+                    fprintf(fp, "\t{ edge [style=dashed, arrowhead=none]; %s ", node_label[id]);
+                    for (j = 0; j < g->node[id].aligned_node_n; ++j) {
+                        fprintf(fp, "-> %s ", node_label[g->node[id].aligned_node_id[j]]);
+                        index = g->node_id_to_index[g->node[id].aligned_node_id[j]];
+                        x_index = index > x_index ? index : x_index;
+                    }
+                    fprintf(fp, "}\n");
+                }
+            }
+        }
+    }
+    fprintf(fp, "}\n");
+    for (i = 0; i < g->node_n; ++i) free(node_label[i]);
+    free(node_label);
+    fclose(fp);
+
+    char cmd[1024];
+    char *type = strrchr(abpt->out_pog, '.');
+    if (type == NULL || (strcmp(type + 1, "pdf") != 0 && strcmp(type + 1, "png") != 0))
+        abamd_fatal("abpoa_dump_pog", "POG can only be dump to .pdf/.png file");
+    snprintf(cmd, sizeof(cmd), "dot %s -T%s > %s", dot_fn, type + 1, abpt->out_pog);
+    free(dot_fn);
+    if (system(cmd) != 0) abamd_fatal("abpoa_dump_pog", "Fail to plot %s DAG.", PROG);
 }

@@ -11,7 +11,7 @@ import os
 import subprocess
 import pytest
 
-from conftest import GOLDEN, ROOT, run_stdout
+from conftest import GOLDEN, ROOT, REF_BIN, run_stdout
 
 CASES = [
     ("seq.fa", [], "expected_seq_cons.txt"),
@@ -102,3 +102,19 @@ def test_aa_mode_vs_reference(cputest_bin, oracle_env, ref_bin, tmp_path):
         ref = run_stdout([ref_bin, str(fa)] + opts)
         got = run_stdout([cputest_bin, str(fa)] + opts, env=oracle_env)
         assert got == ref
+
+
+def test_pog_dot_output(cputest_bin, oracle_env, tmp_path):
+    """--out-pog DOT file is byte-identical to the reference's (the graphviz
+    render step itself needs `dot`, absent here; both sides fail alike)."""
+    import subprocess
+    ref_png = tmp_path / "ref.png"
+    amd_png = tmp_path / "amd.png"
+    fa = os.path.join(GOLDEN, "seq.fa")
+    subprocess.run([REF_BIN, fa, "-g", str(ref_png)], stdout=subprocess.DEVNULL,
+                   stderr=subprocess.DEVNULL)
+    subprocess.run([cputest_bin, fa, "-g", str(amd_png)], env=oracle_env,
+                   stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    ref_dot = open(str(ref_png) + ".dot", "rb").read()
+    amd_dot = open(str(amd_png) + ".dot", "rb").read()
+    assert ref_dot == amd_dot
