@@ -61,3 +61,19 @@ def test_stats_api(handle):
     handle.abpoa_amd_reset_stats()
     handle.abpoa_amd_get_stats(ctypes.byref(a), ctypes.byref(b), ctypes.byref(c))
     assert (a.value, b.value, c.value) == (0, 0, 0)
+
+
+CSRC = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "abpoa_amd", "csrc")
+
+
+def test_align_shim_resolves():
+    """The hybrid seam library must carry every symbol it needs (a missing
+    object in its link list only surfaced on the GPU box otherwise)."""
+    import subprocess
+    so = os.path.join(CSRC, "libabpoa_amd_align.so")
+    if not os.path.exists(so):
+        subprocess.run(["make", "align-shim"], cwd=CSRC, check=True,
+                       stdout=subprocess.DEVNULL)
+    out = subprocess.run(["ldd", "-r", so], stdout=subprocess.PIPE,
+                         stderr=subprocess.STDOUT).stdout.decode()
+    assert "undefined symbol" not in out, out
