@@ -20,9 +20,10 @@ def lib():
     """Load (once) and return the native library handle."""
     global _lib
     if _lib is None:
-        if not os.path.exists(LIB_PATH):
+        path = os.environ.get("ABPOA_AMD_LIB", LIB_PATH)
+        if not os.path.exists(path):
             raise RuntimeError("libabpoa_amd.so not built; run __graft_entry__.build()")
-        _lib = ctypes.CDLL(LIB_PATH)
+        _lib = ctypes.CDLL(path)
         _lib.abpoa_amd_get_stats.argtypes = [ctypes.POINTER(ctypes.c_uint64)] * 3
         _lib.abpoa_amd_get_stats.restype = None
         _lib.abpoa_amd_reset_stats.restype = None
