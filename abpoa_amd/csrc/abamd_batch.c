@@ -117,6 +117,13 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
                         const int *const *seq_lens, const uint8_t *const *const *seqs,
                         abpoa_amd_cons_cb cb, void *user, int n_host_threads) {
     if (n_sets <= 0) return 0;
+    {
+        /* fold thread-count override: at ~250 threads the concurrent folds'
+         * ~10 MB graph working sets thrash the LLC (measured ~20x per-call
+         * inflation); knob for tuning without a rebuild */
+        const char *ft = getenv("ABPOA_AMD_FOLD_THREADS");
+        if (ft && *ft) n_host_threads = atoi(ft);
+    }
     if (n_host_threads < 1) n_host_threads = 1;
     int i, r, max_reads = 0;
     set_state_t *sets = (set_state_t*)abamd_calloc(n_sets, sizeof(set_state_t));
