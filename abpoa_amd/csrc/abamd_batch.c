@@ -14,7 +14,7 @@
 #include "abamd_util.h"
 
 void abamd_timing_report(const char *tag);
-static double g_fold_s, g_cons_s, g_collect_s, g_build_s;
+static double g_fold_s, g_cons_s, g_build_s;
 static long g_big_items;
 static _Atomic unsigned long long g_fold_work_ns; /* summed across workers */
 
@@ -125,7 +125,7 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
         if (ft && *ft) n_host_threads = atoi(ft);
     }
     if (n_host_threads < 1) n_host_threads = 1;
-    int i, r, max_reads = 0;
+    int i, max_reads = 0;
     set_state_t *sets = (set_state_t*)abamd_calloc(n_sets, sizeof(set_state_t));
     for (i = 0; i < n_sets; ++i) {
         sets[i].ab = abpoa_init();
@@ -200,7 +200,6 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
     #define ITEM_R(it) (1 + (int)((it) / n_groups))
     #define ITEM_G(it) ((int)((it) % n_groups))
 
-    double slot_est[3] = {0, 0, 0};
     int slot_big[3] = {0, 0, 0};
 
     /* arena bytes one job will demand (matches gpu_align.cpp's reservation,
@@ -243,7 +242,7 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
             _est += JOB_EST(J); \
             ++_nj; \
         } \
-        slot_nj[slot] = _nj; slot_est[slot] = _est; slot_big[slot] = _est > budget_bytes; \
+        slot_nj[slot] = _nj; slot_big[slot] = _est > budget_bytes; \
         if (slot_big[slot]) ++g_big_items; \
         g_build_s += abamd_realtime() - _tb0; \
     } while (0)
