@@ -118,3 +118,47 @@ def test_pog_dot_output(cputest_bin, oracle_env, tmp_path):
     ref_dot = open(str(ref_png) + ".dot", "rb").read()
     amd_dot = open(str(amd_png) + ".dot", "rb").read()
     assert ref_dot == amd_dot
+
+
+def test_list_input_mode(cputest_bin, oracle_env, tmp_path):
+    """-l: input file is a list of FASTA paths, one MSA per file with
+    batch_index-numbered consensus headers (abpoa.c:152-161)."""
+    lst = tmp_path / "list.txt"
+    fas = []
+    for seed in (61, 62):
+        fa = tmp_path / ("s%d.fa" % seed)
+        subprocess.run(["python3", os.path.join(ROOT, "tests", "make_synth.py"), str(fa),
+                        "--seed", str(seed), "--len", "400", "--depth", "8"],
+                       check=True, stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+        fas.append(str(fa))
+    lst.write_text("".join(p + "\n" for p in fas))
+    for opts in ([], ["-r1"]):
+        ref = subprocess.run([REF_BIN, "-l", str(lst)] + opts, stdout=subprocess.PIPE,
+                             stderr=subprocess.DEVNULL).stdout
+        amd = run_stdout([cputest_bin, "-l", str(lst)] + opts, env=oracle_env)
+        assert ref == amd
+
+
+def test_quality_weights_fastq(cputest_bin, oracle_env, tmp_path):
+    """-Q: FASTQ qualities become per-base edge weights; weighted consensus,
+    MSA, FASTQ output and weighted multi-consensus must match."""
+    import random
+    rng = random.Random(31)
+    ref_seq = "".join(rng.choice("ACGT") for _ in range(500))
+    fq = tmp_path / "q.fq"
+    with open(fq, "w") as f:
+        for i in range(12):
+            out = []
+            for ch in ref_seq:
+                r = rng.random()
+                if r < 0.04: out.append(rng.choice("ACGT"))
+                elif r < 0.07: pass
+                else: out.append(ch)
+            s = "".join(out)
+            qual = "".join(chr(33 + rng.randrange(5, 40)) for _ in s)
+            f.write("@r%d\n%s\n+\n%s\n" % (i, s, qual))
+    for opts in (["-Q"], ["-Q", "-r1"], ["-Q", "-d2"], ["-Q", "-r4"], ["-Q", "-a1"]):
+        ref = subprocess.run([REF_BIN, str(fq)] + opts, stdout=subprocess.PIPE,
+                             stderr=subprocess.DEVNULL).stdout
+        amd = run_stdout([cputest_bin, str(fq)] + opts, env=oracle_env)
+        assert ref == amd, "mismatch opts=%r" % (opts,)
