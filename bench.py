@@ -106,6 +106,17 @@ def run_step(lib, para, sets, n_threads):
     del keep
 
 
+def _cpu_model():
+    try:
+        with open("/proc/cpuinfo") as f:
+            for line in f:
+                if line.startswith("model name"):
+                    return line.split(":", 1)[1].strip()
+    except OSError:
+        pass
+    return "unknown"
+
+
 def cpu_baseline_leg(depth, qlen):
     """Time the unmodified reference binary on one set (bounded sample)."""
     ref_bin = os.path.join(ROOT, "oracle", "_ref", "abpoa")
@@ -122,7 +133,8 @@ def cpu_baseline_leg(depth, qlen):
                        stderr=subprocess.DEVNULL)
         dt = time.monotonic() - t0
     return {"value": 1.0 / dt, "unit": "sets/s", "cores": 1, "kind": "reference",
-            "sample": "1 set (%d reads x %d bp), reference binary, cold arena" % (depth, qlen)}
+            "sample": "1 set (%d reads x %d bp), reference binary, cold arena, host: %s"
+                      % (depth, qlen, _cpu_model())}
 
 
 def main():
