@@ -23,7 +23,7 @@ MTX = os.path.join(ROOT, "tests", "golden", "BLOSUM62.mtx")
 COMP = {"A": "T", "C": "G", "G": "C", "T": "A"}
 
 
-def gen(path, rng, length, depth, n_hap=1, aa=False, rc_frac=0.0):
+def gen(path, rng, length, depth, n_hap=1, aa=False, rc_frac=0.0, fastq=False):
     alpha = "ARNDCQEGHILKMFPSTWYV" if aa else "ACGT"
     base = "".join(rng.choice(alpha) for _ in range(length))
     haps = [base]
@@ -48,7 +48,11 @@ def gen(path, rng, length, depth, n_hap=1, aa=False, rc_frac=0.0):
             s = "".join(out)
             if not aa and i > 0 and rng.random() < rc_frac:
                 s = "".join(COMP[c] for c in reversed(s))
-            f.write(">r%d\n%s\n" % (i, s))
+            if fastq:
+                qual = "".join(chr(33 + rng.randrange(5, 40)) for _ in s)
+                f.write("@r%d\n%s\n+\n%s\n" % (i, s, qual))
+            else:
+                f.write(">r%d\n%s\n" % (i, s))
 
 
 def draw_opts(rng, aa, has_rc, length):
@@ -101,9 +105,12 @@ def main():
             depth = rng.choice([6, 10, 16, 24])
             n_hap = rng.choice([1, 1, 2, 3])
             has_rc = (not aa) and rng.random() < 0.15
-            fa = os.path.join(td, "c%d.fa" % c)
-            gen(fa, rng, length, depth, n_hap, aa, 0.5 if has_rc else 0.0)
+            fastq = (not aa) and rng.random() < 0.15
+            fa = os.path.join(td, "c%d.%s" % (c, "fq" if fastq else "fa"))
+            gen(fa, rng, length, depth, n_hap, aa, 0.5 if has_rc else 0.0, fastq)
             opts = draw_opts(rng, aa, has_rc, length)
+            if fastq and rng.random() < 0.7:
+                opts.append("-Q")
             if aa:
                 opts = ["-c", "-t", MTX] + [o for o in opts if o not in ("-m1", "-m2", "-s")]
             # incremental restore on a fraction of cases
