@@ -1,0 +1,41 @@
+/* Flat-array graph fold core (see abamd_fold_core.c). Device-layout graph
+ * mutation, bit-exact with the pointer-based abamd_graph.c fold. */
+#ifndef ABAMD_FOLD_CORE_H
+#define ABAMD_FOLD_CORE_H
+
+#include <stdint.h>
+
+typedef struct {
+    int node_n, node_cap;
+    uint8_t *base;               /* [node_cap] */
+    int *n_read, *n_span_read;   /* [node_cap] */
+    /* append-only per-direction edge pools; per-node head/tail chains keep
+     * the pointer graph's append order */
+    int edge_n_in, edge_n_out, edge_cap;
+    int *in_head, *in_tail, *out_head, *out_tail;   /* [node_cap] */
+    int *in_to, *in_w, *in_next;                    /* [edge_cap] */
+    int *out_to, *out_w, *out_next;                 /* [edge_cap] */
+    uint64_t *rid_pool; int rid_n;  /* per-OUT-edge read-id bitsets (words) */
+    /* aligned (mismatch-column) groups as linked lists */
+    int aln_n, aln_cap;
+    int *aln_head;                                  /* [node_cap] */
+    int *aln_id, *aln_next;                         /* [aln_cap] */
+} flat_graph_t;
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+void abamd_flat_init(flat_graph_t *fg, int node_cap, int edge_cap, int aln_cap, int rid_n);
+void abamd_flat_free(flat_graph_t *fg);
+void abamd_flat_sort_adjacency(flat_graph_t *fg);
+void abamd_flat_apply_alignment(flat_graph_t *fg, int beg_node_id, int end_node_id,
+                                const uint8_t *seq, const int *weight, int seq_l,
+                                int *qpos_to_node_id, int n_cigar, const abpoa_cigar_t *cig,
+                                int read_id, int add_read_id, int inc_both_ends);
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif

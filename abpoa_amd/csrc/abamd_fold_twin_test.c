@@ -1,0 +1,113 @@
+/* Fold-twin test driver: replays every read of a FASTA through BOTH the
+ * pointer-based graph fold (abamd_graph.c) and the flat-array device-layout
+ * core (abamd_fold_core.c), comparing the full structure after every read:
+ * node bases, per-node in/out adjacency (order, ids, weights), per-out-edge
+ * read-id bitsets, aligned-group lists, n_read and qpos->node maps.
+ *
+ * Built against gpu_stub.c so the aligner is the injected oracle
+ * (ABPOA_AMD_TEST_ALIGNER_SO). Prints "twin OK (<reads> reads, <nodes>
+ * nodes)" on success; aborts on the first structural difference. */
+#include <stdio.h>
+#include <stdlib.h>
+#include <string.h>
+#include "abpoa_amd.h"
+#include "abamd_util.h"
+#include "abamd_fold_core.h"
+
+typedef struct abamd_fx_t abamd_fx_t;
+abpoa_seq_t *abamd_seq_new(void);
+void abamd_seq_destroy(abpoa_seq_t *abs);
+abamd_fx_t *abamd_fx_open(const char *fn);
+void abamd_fx_close(abamd_fx_t *x);
+int abamd_read_seq(abpoa_seq_t *abs, abamd_fx_t *x);
+
+static void die(const char *what, int read_i, int node) {
+    fprintf(stderr, "TWIN MISMATCH after read %d at node %d: %s\n", read_i, node, what);
+    exit(1);
+}
+
+static void compare(abpoa_graph_t *g, flat_graph_t *fg, int read_i, int rid_n) {
+    int i, j, e, a;
+    if (g->node_n != fg->node_n) die("node_n", read_i, -1);
+    for (i = 0; i < g->node_n; ++i) {
+        abpoa_node_t *v = &g->node[i];
+        if (v->base != fg->base[i]) die("base", read_i, i);
+        if (v->n_read != fg->n_read[i]) die("n_read", read_i, i);
+        /* out edges in order */
+        for (j = 0, e = fg->out_head[i]; j < v->out_edge_n; ++j, e = fg->out_next[e]) {
+            if (e == -1) die("out edge count (flat short)", read_i, i);
+            if (fg->out_to[e] != v->out_id[j]) die("out id order", read_i, i);
+            if (fg->out_w[e] != v->out_edge_weight[j]) die("out weight", read_i, i);
+            if (rid_n > 0 && v->read_ids_n > 0) {
+                int k;
+                for (k = 0; k < rid_n; ++k)
+                    if (fg->rid_pool[(size_t)e * rid_n + k] != v->read_ids[j][k])
+                        die("read-id bitset", read_i, i);
+            }
+        }
+        if (e != -1) die("out edge count (flat long)", read_i, i);
+        /* in edges in order */
+        for (j = 0, e = fg->in_head[i]; j < v->in_edge_n; ++j, e = fg->in_next[e]) {
+            if (e == -1) die("in edge count (flat short)", read_i, i);
+            if (fg->in_to[e] != v->in_id[j]) die("in id order", read_i, i);
+            if (fg->in_w[e] != v->in_edge_weight[j]) die("in weight", read_i, i);
+        }
+        if (e != -1) die("in edge count (flat long)", read_i, i);
+        /* aligned groups in order */
+        for (j = 0, a = fg->aln_head[i]; j < v->aligned_node_n; ++j, a = fg->aln_next[a]) {
+            if (a == -1) die("aligned count (flat short)", read_i, i);
+            if (fg->aln_id[a] != v->aligned_node_id[j]) die("aligned order", read_i, i);
+        }
+        if (a != -1) die("aligned count (flat long)", read_i, i);
+    }
+}
+
+int main(int argc, char **argv) {
+    if (argc < 2) { fprintf(stderr, "usage: %s reads.fa [-r1]\n", argv[0]); return 2; }
+    abpoa_para_t *abpt = abpoa_init_para();
+    if (argc > 2 && strcmp(argv[2], "-r1") == 0) abpt->out_msa = 1; /* turn read-id tracking on */
+    abpoa_post_set_para(abpt);
+    abpoa_t *ab = abpoa_init();
+    abpoa_reset(ab, abpt, 1024);
+
+    abpoa_seq_t *abs = abamd_seq_new();
+    abamd_fx_t *fx = abamd_fx_open(argv[1]);
+    int n_seq = abamd_read_seq(abs, fx);
+    abamd_fx_close(fx);
+    ab->abs->n_seq = n_seq;
+
+    int total_len = 0, max_len = 0, i, j;
+    for (i = 0; i < n_seq; ++i) {
+        total_len += abs->seq[i].l;
+        if (abs->seq[i].l > max_len) max_len = abs->seq[i].l;
+    }
+    int rid_n = abpt->use_read_ids ? 1 + ((n_seq - 1) >> 6) : 0;
+    flat_graph_t fg;
+    abamd_flat_init(&fg, total_len + 2, 4 * total_len + 64, 16 * n_seq + 1024, rid_n);
+
+    int *w = (int*)abamd_malloc((size_t)max_len * sizeof(int));
+    int *qmap_live = (int*)abamd_malloc((size_t)max_len * sizeof(int));
+    int *qmap_flat = (int*)abamd_malloc((size_t)max_len * sizeof(int));
+    for (i = 0; i < max_len; ++i) w[i] = 1;
+
+    uint8_t *codes = (uint8_t*)abamd_malloc((size_t)max_len);
+    for (i = 0; i < n_seq; ++i) {
+        int qlen = abs->seq[i].l;
+        for (j = 0; j < qlen; ++j) codes[j] = (uint8_t)ab_amd_char26_table[(int)abs->seq[i].s[j]];
+        abpoa_res_t res; memset(&res, 0, sizeof(res));
+        abpoa_align_sequence_to_graph(ab, abpt, codes, qlen, &res);
+        abamd_flat_apply_alignment(&fg, ABPOA_SRC_NODE_ID, ABPOA_SINK_NODE_ID, codes, w, qlen,
+                                   qmap_flat, res.n_cigar, res.graph_cigar, i,
+                                   abpt->use_read_ids, 1);
+        abamd_flat_sort_adjacency(&fg); /* the live path sorts inside its topo pass */
+        abpoa_add_graph_alignment(ab, abpt, codes, w, qlen, qmap_live, res, i, n_seq, 1);
+        if (res.n_cigar) free(res.graph_cigar);
+        for (j = 0; j < qlen; ++j)
+            if (qmap_live[j] != qmap_flat[j]) die("qpos_to_node_id", i, j);
+        compare(ab->abg, &fg, i, rid_n);
+    }
+    printf("twin OK (%d reads, %d nodes, %d out-edges, %d aligned entries)\n",
+           n_seq, fg.node_n, fg.edge_n_out, fg.aln_n);
+    abamd_flat_free(&fg);
+    return 0;
+}
