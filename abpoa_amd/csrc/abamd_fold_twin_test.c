@@ -83,7 +83,7 @@ int main(int argc, char **argv) {
     }
     int rid_n = abpt->use_read_ids ? 1 + ((n_seq - 1) >> 6) : 0;
     flat_graph_t fg;
-    abamd_flat_init(&fg, total_len + 2, 4 * total_len + 64, 16 * n_seq + 1024, rid_n);
+    abamd_flat_init(&fg, total_len + 2, 4 * total_len + 64, 8 * total_len + 1024, rid_n);
 
     int *w = (int*)abamd_malloc((size_t)max_len * sizeof(int));
     int *qmap_live = (int*)abamd_malloc((size_t)max_len * sizeof(int));
