@@ -168,6 +168,72 @@ void abamd_flat_sort_adjacency(flat_graph_t *fg) {
     }
 }
 
+/* Derived passes over the flat adjacency, matching abamd_graph.c's
+ * abpoa_BFS_set_node_index (:292-316) and abpoa_BFS_set_node_remain
+ * (:320-348) exactly: Kahn BFS with whole-aligned-group queue entry, then
+ * the reverse max-weight-out-edge remain BFS. Outputs into caller arrays
+ * sized node_n; scratch = 2*node_n ints (degree + queue). */
+void abamd_flat_topo_index(const flat_graph_t *fg, int *index_to_node_id,
+                           int *node_id_to_index, int *scratch) {
+    int i, e, a, cur, index = 0;
+    int *in_deg = scratch, *q = scratch + fg->node_n;
+    int qh = 0, qt = 0;
+    for (i = 0; i < fg->node_n; ++i) {
+        int d = 0;
+        for (e = fg->in_head[i]; e != -1; e = fg->in_next[e]) ++d;
+        in_deg[i] = d;
+    }
+    q[qt++] = 0; /* SRC */
+    while (qh < qt) {
+        cur = q[qh++];
+        index_to_node_id[index] = cur;
+        node_id_to_index[cur] = index++;
+        if (cur == 1 /* SINK */) return;
+        for (e = fg->out_head[cur]; e != -1; e = fg->out_next[e]) {
+            int out = fg->out_to[e];
+            if (--in_deg[out] == 0) {
+                int ready = 1;
+                for (a = fg->aln_head[out]; a != -1; a = fg->aln_next[a])
+                    if (in_deg[fg->aln_id[a]] != 0) { ready = 0; break; }
+                if (!ready) continue;
+                q[qt++] = out;
+                for (a = fg->aln_head[out]; a != -1; a = fg->aln_next[a])
+                    q[qt++] = fg->aln_id[a];
+            }
+        }
+    }
+    abamd_fatal("abamd_flat_topo_index", "graph is not a connected DAG");
+}
+
+void abamd_flat_remain(const flat_graph_t *fg, int *max_remain, int *scratch) {
+    int i, e, cur;
+    int *out_deg = scratch, *q = scratch + fg->node_n;
+    int qh = 0, qt = 0;
+    for (i = 0; i < fg->node_n; ++i) {
+        int d = 0;
+        for (e = fg->out_head[i]; e != -1; e = fg->out_next[e]) ++d;
+        out_deg[i] = d;
+        max_remain[i] = 0;
+    }
+    q[qt++] = 1; /* SINK */
+    max_remain[1] = -1;
+    while (qh < qt) {
+        cur = q[qh++];
+        if (cur != 1) {
+            int max_w = -1, max_id = -1;
+            for (e = fg->out_head[cur]; e != -1; e = fg->out_next[e])
+                if (fg->out_w[e] > max_w) { max_w = fg->out_w[e]; max_id = fg->out_to[e]; }
+            max_remain[cur] = max_remain[max_id] + 1;
+        }
+        if (cur == 0 /* SRC */) return;
+        for (e = fg->in_head[cur]; e != -1; e = fg->in_next[e]) {
+            int in = fg->in_to[e];
+            if (--out_deg[in] == 0) q[qt++] = in;
+        }
+    }
+    abamd_fatal("abamd_flat_remain", "graph is not a connected DAG");
+}
+
 /* same walk as abpoa_add_subgraph_alignment minus n_span/topo (those are
  * derived passes). qpos_to_node_id is filled identically. */
 void abamd_flat_apply_alignment(flat_graph_t *fg, int beg_node_id, int end_node_id,

@@ -105,6 +105,23 @@ int main(int argc, char **argv) {
         for (j = 0; j < qlen; ++j)
             if (qmap_live[j] != qmap_flat[j]) die("qpos_to_node_id", i, j);
         compare(ab->abg, &fg, i, rid_n);
+        {   /* derived passes: topo index + remain must match too */
+            int n = fg.node_n;
+            int *i2n = (int*)abamd_malloc((size_t)n * sizeof(int));
+            int *n2i = (int*)abamd_malloc((size_t)n * sizeof(int));
+            int *rem = (int*)abamd_malloc((size_t)n * sizeof(int));
+            int *scr = (int*)abamd_malloc((size_t)2 * n * sizeof(int));
+            abamd_flat_topo_index(&fg, i2n, n2i, scr);
+            abamd_flat_remain(&fg, rem, scr);
+            abpoa_graph_t *g = ab->abg;
+            for (j = 0; j < n; ++j) {
+                if (i2n[j] != g->index_to_node_id[j]) die("topo index order", i, j);
+                if (n2i[j] != g->node_id_to_index[j]) die("node->index", i, j);
+                if (g->node_id_to_max_remain && rem[j] != g->node_id_to_max_remain[j])
+                    die("max_remain", i, j);
+            }
+            free(i2n); free(n2i); free(rem); free(scr);
+        }
     }
     printf("twin OK (%d reads, %d nodes, %d out-edges, %d aligned entries)\n",
            n_seq, fg.node_n, fg.edge_n_out, fg.aln_n);
