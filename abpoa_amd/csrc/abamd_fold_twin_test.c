@@ -99,26 +99,35 @@ int main(int argc, char **argv) {
         abamd_flat_apply_alignment(&fg, ABPOA_SRC_NODE_ID, ABPOA_SINK_NODE_ID, codes, w, qlen,
                                    qmap_flat, res.n_cigar, res.graph_cigar, i,
                                    abpt->use_read_ids, 1);
-        abamd_flat_sort_adjacency(&fg); /* the live path sorts inside its topo pass */
         abpoa_add_graph_alignment(ab, abpt, codes, w, qlen, qmap_live, res, i, n_seq, 1);
         if (res.n_cigar) free(res.graph_cigar);
         for (j = 0; j < qlen; ++j)
             if (qmap_live[j] != qmap_flat[j]) die("qpos_to_node_id", i, j);
-        compare(ab->abg, &fg, i, rid_n);
-        {   /* derived passes: topo index + remain must match too */
+        {   /* derived passes in the live order (abpoa_topological_sort):
+             * index BFS on the pre-sort adjacency, THEN the weight sort,
+             * then remain BFS — only after that does the structural
+             * compare see the same post-sort adjacency the live graph has */
             int n = fg.node_n;
             int *i2n = (int*)abamd_malloc((size_t)n * sizeof(int));
             int *n2i = (int*)abamd_malloc((size_t)n * sizeof(int));
             int *rem = (int*)abamd_malloc((size_t)n * sizeof(int));
             int *scr = (int*)abamd_malloc((size_t)2 * n * sizeof(int));
             abamd_flat_topo_index(&fg, i2n, n2i, scr);
+            abamd_flat_sort_adjacency(&fg);
             abamd_flat_remain(&fg, rem, scr);
+            compare(ab->abg, &fg, i, rid_n);
             abpoa_graph_t *g = ab->abg;
+            /* the BFS returns when the SINK pops (reference quirk): nodes
+             * still queued keep stale indices, and the DP only reads the
+             * topo prefix 0..sink_index — compare exactly that */
             for (j = 0; j < n; ++j) {
+                int id;
                 if (i2n[j] != g->index_to_node_id[j]) die("topo index order", i, j);
-                if (n2i[j] != g->node_id_to_index[j]) die("node->index", i, j);
-                if (g->node_id_to_max_remain && rem[j] != g->node_id_to_max_remain[j])
-                    die("max_remain", i, j);
+                id = i2n[j];
+                if (n2i[id] != g->node_id_to_index[id]) die("node->index", i, id);
+                if (g->node_id_to_max_remain && rem[id] != g->node_id_to_max_remain[id])
+                    die("max_remain", i, id);
+                if (id == ABPOA_SINK_NODE_ID) break;
             }
             free(i2n); free(n2i); free(rem); free(scr);
         }
