@@ -8,6 +8,8 @@ the sequential mutation order bit-exactly on CPU."""
 import os
 import subprocess
 
+import pytest
+
 from conftest import ROOT, GOLDEN, ORACLE_SO
 
 TWIN = os.path.join(ROOT, "abpoa_amd", "csrc", "abpoa_amd_foldtwin")
@@ -35,3 +37,16 @@ def test_twin_synthetic(tmp_path):
                        check=True, stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
         _run(fa)
         _run(fa, ["-r1"])
+
+
+@pytest.mark.gpu
+def test_device_fold_kernel(tmp_path):
+    """abamd_fold_kernel on a real GPU: one launch per read mutates the
+    device-resident flat graph; the full state (chains, pools, read-id
+    bitsets, topo index, remain) must equal the host twin after every read."""
+    bin_ = os.path.join(ROOT, "abpoa_amd", "csrc", "abpoa_amd_foldgpu")
+    for fa in ("seq.fa", "fgt.fa"):
+        out = subprocess.run([bin_, os.path.join(GOLDEN, fa)],
+                             stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+        assert out.returncode == 0, out.stderr.decode()[-500:]
+        assert b"device fold OK" in out.stdout
