@@ -33,6 +33,7 @@ static void compare(abpoa_graph_t *g, flat_graph_t *fg, int read_i, int rid_n) {
         abpoa_node_t *v = &g->node[i];
         if (v->base != fg->base[i]) die("base", read_i, i);
         if (v->n_read != fg->n_read[i]) die("n_read", read_i, i);
+        if (v->n_span_read != fg->n_span_read[i]) die("n_span_read", read_i, i);
         /* out edges in order */
         for (j = 0, e = fg->out_head[i]; j < v->out_edge_n; ++j, e = fg->out_next[e]) {
             if (e == -1) die("out edge count (flat short)", read_i, i);
@@ -115,6 +116,7 @@ int main(int argc, char **argv) {
             abamd_flat_topo_index(&fg, i2n, n2i, scr);
             abamd_flat_sort_adjacency(&fg);
             abamd_flat_remain(&fg, rem, scr);
+            abamd_flat_update_n_span(&fg, i2n, n2i, 1);
             compare(ab->abg, &fg, i, rid_n);
             abpoa_graph_t *g = ab->abg;
             /* the BFS returns when the SINK pops (reference quirk): nodes

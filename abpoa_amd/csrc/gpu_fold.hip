@@ -68,5 +68,7 @@ extern "C" __global__ void abamd_fold_kernel(fold_job_t *jobs, int n_jobs) {
                               job->scratch);
         abamd_flat_sort_adjacency(&job->g);
         abamd_flat_remain(&job->g, job->max_remain, job->scratch);
+        abamd_flat_update_n_span(&job->g, job->index_to_node_id,
+                                 job->node_id_to_index, 1);
     }
 }
