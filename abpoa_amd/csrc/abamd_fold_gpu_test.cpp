@@ -140,6 +140,7 @@ int main(int argc, char **argv) {
         abamd_flat_topo_index(&hg, h_i2n, h_n2i, h_scr);
         abamd_flat_sort_adjacency(&hg);
         abamd_flat_remain(&hg, h_rem, h_scr);
+        abamd_flat_update_n_span(&hg, h_i2n, h_n2i, 1);
 
         /* device fold */
         if (res.n_cigar)
@@ -175,6 +176,7 @@ int main(int argc, char **argv) {
             if (memcmp(buf, hptr, (size_t)(count) * sizeof(T))) die(what, i); } while (0)
         CMP_ARR(dgh.base, hg.base, n, uint8_t, "base");
         CMP_ARR(dgh.n_read, hg.n_read, n, int, "n_read");
+        CMP_ARR(dgh.n_span_read, hg.n_span_read, n, int, "n_span_read");
         CMP_ARR(dgh.in_head, hg.in_head, n, int, "in_head");
         CMP_ARR(dgh.out_head, hg.out_head, n, int, "out_head");
         CMP_ARR(dgh.aln_head, hg.aln_head, n, int, "aln_head");
