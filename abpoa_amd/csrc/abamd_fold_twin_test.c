@@ -118,6 +118,17 @@ int main(int argc, char **argv) {
             abamd_flat_remain(&fg, rem, scr);
             abamd_flat_update_n_span(&fg, i2n, n2i, 1);
             compare(ab->abg, &fg, i, rid_n);
+            if (abpt->out_msa) { /* msa rank over the post-sort adjacency;
+                 * the live array only exists when MSA output is on */
+                int *mr = (int*)abamd_malloc((size_t)n * sizeof(int));
+                int k;
+                abamd_flat_msa_rank(&fg, mr, scr);
+                ab->abg->is_set_msa_rank = 0;
+                abamd_set_msa_rank(ab->abg, ABPOA_SRC_NODE_ID, ABPOA_SINK_NODE_ID);
+                for (k = 0; k < n; ++k)
+                    if (mr[k] != ab->abg->node_id_to_msa_rank[k]) die("msa rank", i, k);
+                free(mr);
+            }
             abpoa_graph_t *g = ab->abg;
             /* the BFS returns when the SINK pops (reference quirk): nodes
              * still queued keep stale indices, and the DP only reads the
