@@ -39,6 +39,7 @@ typedef struct {
     int *qpos_to_node_id;
     int read_id, add_read_id;
     int *index_to_node_id, *node_id_to_index, *max_remain, *scratch;
+    int *msa_rank;
 } fold_job_t;
 
 extern "C" __global__ void abamd_fold_kernel(fold_job_t *jobs, int n_jobs);
@@ -76,6 +77,7 @@ int main(int argc, char **argv) {
     int *h_i2n = (int*)abamd_malloc((size_t)node_cap * sizeof(int));
     int *h_n2i = (int*)abamd_malloc((size_t)node_cap * sizeof(int));
     int *h_rem = (int*)abamd_malloc((size_t)node_cap * sizeof(int));
+    int *h_msa = (int*)abamd_malloc((size_t)node_cap * sizeof(int));
     int *h_scr = (int*)abamd_malloc((size_t)2 * node_cap * sizeof(int));
     int *h_qmap = (int*)abamd_malloc((size_t)max_len * sizeof(int));
 
@@ -115,6 +117,8 @@ int main(int argc, char **argv) {
     DALLOC(d_qmap, max_len, int);
     DALLOC(d_i2n, node_cap, int); DALLOC(d_n2i, node_cap, int);
     DALLOC(d_rem, node_cap, int); DALLOC(d_scr, 2 * (size_t)node_cap, int);
+    int *d_msa_rank;
+    DALLOC(d_msa_rank, node_cap, int);
     fold_job_t *d_job;
     DALLOC(d_job, 1, fold_job_t);
 
@@ -141,6 +145,7 @@ int main(int argc, char **argv) {
         abamd_flat_sort_adjacency(&hg);
         abamd_flat_remain(&hg, h_rem, h_scr);
         abamd_flat_update_n_span(&hg, h_i2n, h_n2i, 1);
+        abamd_flat_msa_rank(&hg, h_msa, h_scr);
 
         /* device fold */
         if (res.n_cigar)
@@ -155,6 +160,7 @@ int main(int argc, char **argv) {
         job.read_id = i; job.add_read_id = 1;
         job.index_to_node_id = d_i2n; job.node_id_to_index = d_n2i;
         job.max_remain = d_rem; job.scratch = d_scr;
+        job.msa_rank = d_msa_rank;
         HIP_CHECK(hipMemcpy(d_job, &job, sizeof(job), hipMemcpyHostToDevice));
         hipLaunchKernelGGL(abamd_fold_kernel, dim3(1), dim3(64), 0, 0, d_job, 1);
         HIP_CHECK(hipGetLastError());
@@ -194,6 +200,14 @@ int main(int argc, char **argv) {
         CMP_POOL(dgh.aln_next, hg.aln_next, hg.aln_n, "aln_next");
         HIP_CHECK(hipMemcpy(bufr, dgh.rid_pool, (size_t)hg.edge_n_out * rid_n * sizeof(uint64_t), hipMemcpyDeviceToHost));
         if (memcmp(bufr, hg.rid_pool, (size_t)hg.edge_n_out * rid_n * sizeof(uint64_t))) die("rid_pool", i);
+        {   /* msa rank: every node reachable by the rank DFS */
+            HIP_CHECK(hipMemcpy(buf, d_msa_rank, (size_t)n * sizeof(int), hipMemcpyDeviceToHost));
+            int k, sink_idx = h_n2i[ABPOA_SINK_NODE_ID];
+            for (k = 0; k <= sink_idx; ++k) {
+                int id = h_i2n[k];
+                if (buf[id] != h_msa[id]) die("msa rank", i);
+            }
+        }
         {   /* remain over the topo prefix */
             HIP_CHECK(hipMemcpy(buf, d_rem, (size_t)n * sizeof(int), hipMemcpyDeviceToHost));
             int k, sink_idx = h_n2i[ABPOA_SINK_NODE_ID];

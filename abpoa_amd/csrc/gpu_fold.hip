@@ -53,6 +53,7 @@ typedef struct {
     int *qpos_to_node_id;
     int read_id, add_read_id;
     int *index_to_node_id, *node_id_to_index, *max_remain, *scratch;
+    int *msa_rank; /* optional; skipped when NULL */
 } fold_job_t;
 
 extern "C" __global__ void abamd_fold_kernel(fold_job_t *jobs, int n_jobs) {
@@ -70,5 +71,7 @@ extern "C" __global__ void abamd_fold_kernel(fold_job_t *jobs, int n_jobs) {
         abamd_flat_remain(&job->g, job->max_remain, job->scratch);
         abamd_flat_update_n_span(&job->g, job->index_to_node_id,
                                  job->node_id_to_index, 1);
+        if (job->msa_rank)
+            abamd_flat_msa_rank(&job->g, job->msa_rank, job->scratch);
     }
 }
