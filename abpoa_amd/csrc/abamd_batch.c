@@ -151,7 +151,12 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
          * concurrently, so each launch gets just under a third of the
          * usable HBM (abamd_gpu_free_mem counts our own held buffers) */
         int64_t free_b = abamd_gpu_free_mem();
-        mem_gb = free_b > 0 ? (double)free_b * 0.28 / 1e9 : 48.0;
+        int np = n_sets >= 3 ? 3 : (n_sets >= 1 ? n_sets : 1);
+        {
+            const char *gs = getenv("ABPOA_AMD_GROUPS");
+            if (gs && *gs) { int g = atoi(gs); if (g >= 1 && g <= 6) np = g; }
+        }
+        mem_gb = free_b > 0 ? (double)free_b * 0.84 / np / 1e9 : 48.0;
     }
     const double budget_bytes = mem_gb * 1e9;
     abamd_gpu_set_arena_cap((uint64_t)budget_bytes);
@@ -180,10 +185,12 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
      * on the dedicated slot 3. */
     int n_groups = n_sets >= 3 ? 3 : n_sets;
     {
+        /* up to 6 pipeline groups (device slots 0..5; 7 = retry/big). 3 is
+         * the measured default; deeper overlap is a round-2 tuning knob */
         const char *gs = getenv("ABPOA_AMD_GROUPS");
         if (gs && *gs) {
             int g = atoi(gs);
-            if (g >= 1 && g <= 3 && g <= n_sets) n_groups = g;
+            if (g >= 1 && g <= 6 && g <= n_sets) n_groups = g;
         }
     }
     int *grp_of = (int*)abamd_malloc((size_t)n_sets * sizeof(int));
@@ -191,16 +198,15 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
 
     /* item = (round r >= 1, group g); g-major within a round */
     long n_items = (long)(max_reads - 1) * n_groups;
-    abamd_batch_job_t *slot_jobs[3];
-    slot_jobs[0] = (abamd_batch_job_t*)abamd_malloc((size_t)n_sets * sizeof(abamd_batch_job_t));
-    slot_jobs[1] = (abamd_batch_job_t*)abamd_malloc((size_t)n_sets * sizeof(abamd_batch_job_t));
-    slot_jobs[2] = (abamd_batch_job_t*)abamd_malloc((size_t)n_sets * sizeof(abamd_batch_job_t));
-    int slot_nj[3] = {0, 0, 0};
+    abamd_batch_job_t *slot_jobs[6];
+    for (i = 0; i < 6; ++i)
+        slot_jobs[i] = (abamd_batch_job_t*)abamd_malloc((size_t)n_sets * sizeof(abamd_batch_job_t));
+    int slot_nj[6] = {0};
 
     #define ITEM_R(it) (1 + (int)((it) / n_groups))
     #define ITEM_G(it) ((int)((it) % n_groups))
 
-    int slot_big[3] = {0, 0, 0};
+    int slot_big[6] = {0};
 
     /* arena bytes one job will demand (matches gpu_align.cpp's reservation,
      * including the measured-cells tightening) */
@@ -258,7 +264,7 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
                 if (_take > 0 && _acc + _e > budget_bytes) break; \
                 _acc += _e; ++_take; \
             } \
-            abamd_gpu_align_batch_slot(&slot_jobs[slot][_done], _take, 3); \
+            abamd_gpu_align_batch_slot(&slot_jobs[slot][_done], _take, 7); \
             _done += _take; \
         } \
     } while (0)
@@ -315,8 +321,9 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
         }
         FOLD_ITEM(n_items - 1);
     }
-    free(slot_jobs[2]);
-    free(grp_of); free(slot_jobs[0]); free(slot_jobs[1]);
+
+    free(grp_of);
+    for (i = 0; i < 6; ++i) free(slot_jobs[i]);
 
     /* consensus on host threads, then emit callbacks in order */
     double tc0 = abamd_realtime();

@@ -149,9 +149,9 @@ struct SlotDev {
 
 struct GpuCtx {
     bool init = false;
-    SlotDev dev[4];
-    PinnedBuf stage2[4];              /* per-slot pinned H2D staging (slot 3 = retry/big) */
-    std::vector<HostBuf> jb_bufs2[4]; /* per-slot per-job pack buffers */
+    SlotDev dev[8];
+    PinnedBuf stage2[8];              /* per-slot pinned H2D staging (slot 7 = retry/big) */
+    std::vector<HostBuf> jb_bufs2[8]; /* per-slot per-job pack buffers */
     HostBuf hb;
     std::vector<uint64_t> hcig;
     void ensure_init() {
@@ -390,7 +390,7 @@ struct PendingBatch {
     size_t total = 0;
     bool active = false;
 };
-thread_local PendingBatch g_slots[4];
+thread_local PendingBatch g_slots[8];
 }
 
 static int batch_launch(GpuCtx &C, PendingBatch &P); /* fwd */
@@ -657,7 +657,7 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
         g_launches += 1;
     }
     g_gpu_ns += now_ns() - t_gpu0;
-    if (slot == 3) {
+    if (slot == 7) {
         /* retry slot itself overflowed: escalate reservations in place */
         for (int attempt = 0;; ++attempt) {
             bool overflow = false;
@@ -671,7 +671,7 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
                 fprintf(stderr, "[abpoa_amd] arena overflow persists after %d retries\n", attempt);
                 exit(EXIT_FAILURE);
             }
-            prepare_internal(batch, n_jobs, 3, floor_est.data());
+            prepare_internal(batch, n_jobs, 7, floor_est.data());
             batch_launch(C, PB);
             HIP_CHECK(hipStreamSynchronize(D.stream));
             HIP_CHECK(hipMemcpy(PB.hres.data(), D.results.p, (size_t)n_jobs * sizeof(abamd_gpu_res_t), hipMemcpyDeviceToHost));
@@ -686,7 +686,7 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
     C.hcig.clear();
     for (int i = 0; i < n_jobs; ++i) {
         abamd_gpu_res_t &R = PB.hres[i];
-        if (slot != 3 && R.status == ABAMD_JOB_ARENA_OVERFLOW)
+        if (slot != 7 && R.status == ABAMD_JOB_ARENA_OVERFLOW)
             continue; /* re-run and unpacked by the retry sub-batch */
         if (R.status != ABAMD_JOB_OK) {
             fprintf(stderr, "[abpoa_amd] GPU job %d failed with status %d\n", i, R.status);
@@ -720,7 +720,7 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
      * (slot 2) with doubled reservations — a whole-batch relaunch costs a
      * full latency-bound kernel pass. The sub-batch repacks from the graphs
      * (the pinned stage may already hold another slot's data). */
-    if (slot != 3) {
+    if (slot != 7) {
         std::vector<BatchJob> failed;
         std::vector<int64_t> floors;
         for (int i = 0; i < n_jobs; ++i) {
@@ -733,9 +733,9 @@ extern "C" int abamd_gpu_batch_finish_slot(int slot) {
         }
         if (!failed.empty()) {
             g_retry_jobs += failed.size();
-            prepare_internal(failed.data(), (int)failed.size(), 3, floors.data());
-            int r2 = abamd_gpu_batch_launch(3);
-            if (!r2) abamd_gpu_batch_finish_slot(3);
+            prepare_internal(failed.data(), (int)failed.size(), 7, floors.data());
+            int r2 = abamd_gpu_batch_launch(7);
+            if (!r2) abamd_gpu_batch_finish_slot(7);
         }
     }
     PB.active = false;

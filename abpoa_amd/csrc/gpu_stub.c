@@ -39,8 +39,8 @@ int abamd_gpu_align_batch(abamd_batch_job_t *batch, int n_jobs) {
 void abpoa_amd_get_stats2(uint64_t *alg_bytes) { if (alg_bytes) *alg_bytes = 0; }
 void abamd_timing_report(const char *tag) { (void)tag; }
 
-static abamd_batch_job_t *stub_slot_batch[4];
-static int stub_slot_n[4];
+static abamd_batch_job_t *stub_slot_batch[8];
+static int stub_slot_n[8];
 int abamd_gpu_batch_prepare(abamd_batch_job_t *batch, int n_jobs, int slot) {
     stub_slot_batch[slot] = batch; stub_slot_n[slot] = n_jobs; return 0;
 }
