@@ -103,9 +103,17 @@ static void segs_grow(segs_t *g) {
 
 /* one MSA-FASTA row into the graph (abpoa_fa_parse_seq, :573-608) */
 static int fa_parse_seq(abpoa_graph_t *g, abpoa_seq_t *abs, rstr_t *seq, rstr_t *name,
-                        int add_read_id, int p_i, int p_n, int **rank2node_id) {
-    if (*rank2node_id == 0)
+                        int add_read_id, int p_i, int p_n, int **rank2node_id,
+                        int *rank2cap) {
+    if (*rank2node_id == 0) {
         *rank2node_id = (int*)abamd_calloc((size_t)seq->l, sizeof(int));
+        *rank2cap = seq->l;
+    }
+    /* a valid MSA has equal-width rows; a longer row would walk off the
+     * column map (the reference reads out of bounds here) */
+    if (seq->l > *rank2cap)
+        abamd_fatal("abpoa_restore_graph",
+                    "MSA rows of unequal length (%d > %d) in restore file", seq->l, *rank2cap);
     char *s = seq->s;
     int read_ids_n = 1 + ((p_n - 1) >> 6);
     int i, cur_id, aln_id, last_id = ABPOA_SRC_NODE_ID;
@@ -255,7 +263,7 @@ abpoa_t *abpoa_restore_graph(abpoa_t *ab, abpoa_para_t *abpt) {
     segs_t segs; memset(&segs, 0, sizeof(segs)); sm_init(&segs.h);
     strmap_t in_map, out_map; sm_init(&in_map); sm_init(&out_map);
     int add_read_id = abpt->use_read_ids;
-    int p_i = -1, is_fa = 0, *rank2node_id = 0;
+    int p_i = -1, is_fa = 0, *rank2node_id = 0, rank2cap = 0;
     long line_n = 0;
     abpoa_graph_t *g = ab->abg;
     abpoa_seq_t *abs = ab->abs;
@@ -283,7 +291,7 @@ abpoa_t *abpoa_restore_graph(abpoa_t *ab, abpoa_para_t *abpt) {
             if (sl > 0 && s[0] == '>') {
                 if (segs.seq[segs.n].l > 0) {
                     fa_parse_seq(g, abs, &segs.seq[segs.n], &segs.name[segs.n],
-                                 add_read_id, p_i, p_i + 1, &rank2node_id);
+                                 add_read_id, p_i, p_i + 1, &rank2node_id, &rank2cap);
                     segs.n++;
                 }
                 int e = 1;
@@ -313,7 +321,7 @@ abpoa_t *abpoa_restore_graph(abpoa_t *ab, abpoa_para_t *abpt) {
     }
     if (is_fa && segs.seq[segs.n].l > 0) {
         fa_parse_seq(g, abs, &segs.seq[segs.n], &segs.name[segs.n],
-                     add_read_id, p_i, p_i + 1, &rank2node_id);
+                     add_read_id, p_i, p_i + 1, &rank2node_id, &rank2cap);
         segs.n++;
     }
     free(buf);

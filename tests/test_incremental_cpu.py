@@ -39,3 +39,17 @@ def test_incremental_restore(tmp_path, restore_fmt, opts):
                          stdout=subprocess.PIPE, stderr=subprocess.DEVNULL).stdout
     amd = run_stdout([CPUTEST_BIN, "-i", restore, new] + opts, env=env)
     assert ref == amd
+
+
+def test_restore_rejects_unequal_msa_rows(tmp_path):
+    """A longer-than-first MSA row in a restore file must fail cleanly (the
+    reference reads out of bounds here; we refuse with an error instead)."""
+    bad = tmp_path / "bad.fa"
+    bad.write_text(">a\nACGT\n>b\nAC-T-EXTRA-LONGER-ROW\n")
+    new = tmp_path / "new.fa"
+    new.write_text(">r0\nACGTACGT\n")
+    env = dict(os.environ)
+    env["ABPOA_AMD_TEST_ALIGNER_SO"] = ORACLE_SO
+    r = subprocess.run([CPUTEST_BIN, "-i", str(bad), str(new)], env=env,
+                       stdout=subprocess.DEVNULL, stderr=subprocess.PIPE)
+    assert r.returncode == 1 and b"unequal length" in r.stderr
