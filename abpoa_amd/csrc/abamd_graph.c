@@ -329,7 +329,10 @@ void abpoa_BFS_set_node_remain(abpoa_graph_t *g, int src_id, int sink_id) {
     g->node_id_to_max_remain[sink_id] = -1;
     while (q_pop(&q, &cur)) {
         if (cur != sink_id) {
-            int max_w = -1, max_id = -1;
+            /* INT32_MIN sentinel, not the reference's -1: all-negative edge
+             * weights (possible only via malformed FASTQ qualities) would
+             * leave max_id at -1 and index out of bounds */
+            int max_w = INT32_MIN, max_id = -1;
             for (i = 0; i < g->node[cur].out_edge_n; ++i) {
                 if (g->node[cur].out_edge_weight[i] > max_w) {
                     max_w = g->node[cur].out_edge_weight[i];
@@ -482,7 +485,8 @@ static void add_first_sequence(abpoa_graph_t *g, abpoa_para_t *abpt, const uint8
         g->node[cur].n_span_read = g->node[last].n_span_read;
         last = cur;
     }
-    abpoa_add_graph_edge(g, last, ABPOA_SINK_NODE_ID, 0, weight[len-1], add_read_id, add_read_weight, read_id, read_ids_n, tot_read_n);
+    /* len == 0 would read weight[-1] (the reference does); use 0 instead */
+    abpoa_add_graph_edge(g, last, ABPOA_SINK_NODE_ID, 0, len > 0 ? weight[len-1] : 0, add_read_id, add_read_weight, read_id, read_ids_n, tot_read_n);
     g->is_called_cons = g->is_set_msa_rank = g->is_topological_sorted = 0;
     abpoa_topological_sort(g, abpt);
     update_n_span(g, ABPOA_SRC_NODE_ID, ABPOA_SINK_NODE_ID, 1);
@@ -551,7 +555,7 @@ int abpoa_add_subgraph_alignment(abpoa_t *ab, abpoa_para_t *abpt, int beg_node_i
             }
         } /* ABPOA_CDEL: consumes nothing on the query side */
     }
-    abpoa_add_graph_edge(g, last_id, end_node_id, 1 - last_new, weight[seq_l - 1], add_read_id, add_read_weight, read_id, read_ids_n, tot_read_n);
+    abpoa_add_graph_edge(g, last_id, end_node_id, 1 - last_new, seq_l > 0 ? weight[seq_l - 1] : 0, add_read_id, add_read_weight, read_id, read_ids_n, tot_read_n);
     g->is_called_cons = g->is_set_msa_rank = g->is_topological_sorted = 0;
     abpoa_topological_sort(g, abpt);
     update_n_span(g, beg_node_id, end_node_id, inc_both_ends);

@@ -25,15 +25,21 @@ void abpoa_set_mat_from_file(abpoa_para_t *p, char *mat_fn) {
     if (!fp) abamd_fatal("abpoa_set_mat_from_file", "cannot open scoring matrix '%s'", mat_fn);
     char line[1024];
     int *order = (int*)abamd_malloc((size_t)p->m * sizeof(int));
-    int first = 1, i;
+    int first = 1, i, n_hdr = 0;
     while (fgets(line, sizeof(line), fp)) {
         if (line[0] == '#') continue;
         if (first) {
             int n = 0;
             for (i = 0; line[i]; ++i) {
                 if (isspace((unsigned char)line[i])) continue;
-                order[n++] = ab_amd_char26_table[(int)line[i]];
+                if (n >= p->m)
+                    abamd_fatal("abpoa_set_mat_from_file", "too many residues in matrix header");
+                order[n] = ab_amd_char26_table[(int)line[i]];
+                if (order[n] >= p->m)
+                    abamd_fatal("abpoa_set_mat_from_file", "unknown residue '%c' in matrix header", line[i]);
+                n++;
             }
+            n_hdr = n;
             first = 0;
         } else {
             char *s = line, *end; int row = -1, n = 0, is_base = 1;
@@ -44,7 +50,7 @@ void abpoa_set_mat_from_file(abpoa_para_t *p, char *mat_fn) {
                     if (row >= p->m) abamd_fatal("abpoa_set_mat_from_file", "unknown residue '%c'", *s);
                     is_base = 0; ++s;
                 } else {
-                    if (n == p->m) abamd_fatal("abpoa_set_mat_from_file", "too many scores in a row");
+                    if (n >= n_hdr) abamd_fatal("abpoa_set_mat_from_file", "more scores than header residues");
                     long v = strtol(s, &end, 10);
                     s = end;
                     p->mat[row * p->m + order[n]] = (int)v;

@@ -152,6 +152,9 @@ int abamd_fx_read(abamd_fx_t *x) {
             if (c == '>' || c == '@') { x->last_hdr = c; break; }
             if (c != '\n' && c != '\r') { gr_ungetc(&x->rd, c); break; }
         }
+        /* kseq semantics: truncated quality is a malformed record (-2);
+         * the record is dropped and reading stops */
+        if (x->qual.l < x->seq.l) return -2;
     }
     return x->seq.l;
 }
