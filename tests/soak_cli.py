@@ -94,7 +94,9 @@ def main():
     base = int(sys.argv[2]) if len(sys.argv) > 2 else 9000
     binary = sys.argv[3] if len(sys.argv) > 3 else CPU
     env = dict(os.environ)
-    if binary == CPU:
+    # inject the oracle for any CPU-test build; the product GPU CLI (named
+    # exactly abpoa_amd) must run its own aligner
+    if os.path.basename(binary) != "abpoa_amd":
         env["ABPOA_AMD_TEST_ALIGNER_SO"] = ORACLE
     fails = 0
     with tempfile.TemporaryDirectory() as td:
