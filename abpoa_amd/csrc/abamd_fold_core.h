@@ -35,6 +35,11 @@ void abamd_flat_remain(const flat_graph_t *fg, int *max_remain, int *scratch);
 void abamd_flat_update_n_span(flat_graph_t *fg, const int *index_to_node_id,
                               const int *node_id_to_index, int inc_both_ends);
 void abamd_flat_msa_rank(const flat_graph_t *fg, int *msa_rank_out, int *scratch);
+int abamd_flat_build_rows(const flat_graph_t *fg, const int *index_to_node_id,
+                          const int *node_id_to_index, const int *max_remain,
+                          int use_remain, uint8_t *row_base, int *row_node_id,
+                          int *pre_off, int *out_off, int *remain,
+                          int *pre_idx, int *out_idx);
 void abamd_flat_apply_alignment(flat_graph_t *fg, int beg_node_id, int end_node_id,
                                 const uint8_t *seq, const int *weight, int seq_l,
                                 int *qpos_to_node_id, int n_cigar, const abpoa_cigar_t *cig,

@@ -129,6 +129,43 @@ int main(int argc, char **argv) {
                     if (mr[k] != ab->abg->node_id_to_msa_rank[k]) die("msa rank", i, k);
                 free(mr);
             }
+            {   /* DP-row CSR: flat materializer vs the pointer graph,
+                 * mirroring pack_job's whole-graph loops (gpu_align.cpp) */
+                abpoa_graph_t *pg = ab->abg;
+                int nr = pg->node_id_to_index[ABPOA_SINK_NODE_ID] + 1;
+                int ecap = fg.edge_n_out + 4;
+                uint8_t *fb = (uint8_t*)abamd_malloc((size_t)nr);
+                int *fni = (int*)abamd_malloc((size_t)nr * sizeof(int));
+                int *fpo = (int*)abamd_malloc((size_t)(nr + 1) * sizeof(int));
+                int *foo = (int*)abamd_malloc((size_t)(nr + 1) * sizeof(int));
+                int *frm = (int*)abamd_malloc((size_t)nr * sizeof(int));
+                int *fpi = (int*)abamd_malloc((size_t)ecap * sizeof(int));
+                int *foi = (int*)abamd_malloc((size_t)ecap * sizeof(int));
+                int flat_nr = abamd_flat_build_rows(&fg, i2n, n2i, rem, 1,
+                                                    fb, fni, fpo, foo, frm, fpi, foi);
+                if (flat_nr != nr) die("csr n_rows", i, flat_nr);
+                int r, e2, np = 0, no = 0;
+                for (r = 0; r < nr; ++r) {
+                    int nid = pg->index_to_node_id[r];
+                    if (fb[r] != pg->node[nid].base) die("csr base", i, r);
+                    if (fni[r] != nid) die("csr node id", i, r);
+                    if (frm[r] != pg->node_id_to_max_remain[nid]) die("csr remain", i, r);
+                    if (fpo[r] != np || foo[r] != no) die("csr offsets", i, r);
+                    if (r > 0)
+                        for (e2 = 0; e2 < pg->node[nid].in_edge_n; ++e2) {
+                            if (fpi[np] != pg->node_id_to_index[pg->node[nid].in_id[e2]])
+                                die("csr pre idx", i, r);
+                            ++np;
+                        }
+                    for (e2 = 0; e2 < pg->node[nid].out_edge_n; ++e2) {
+                        if (foi[no] != pg->node_id_to_index[pg->node[nid].out_id[e2]])
+                            die("csr out idx", i, r);
+                        ++no;
+                    }
+                }
+                if (fpo[nr] != np || foo[nr] != no) die("csr totals", i, nr);
+                free(fb); free(fni); free(fpo); free(foo); free(frm); free(fpi); free(foi);
+            }
             abpoa_graph_t *g = ab->abg;
             /* the BFS returns when the SINK pops (reference quirk): nodes
              * still queued keep stale indices, and the DP only reads the
