@@ -135,9 +135,6 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
         sets[i].n_seqs = n_seqs[i];
         if (n_seqs[i] > max_reads) max_reads = n_seqs[i];
     }
-    abamd_batch_job_t *jobs = (abamd_batch_job_t*)abamd_malloc((size_t)n_sets * sizeof(abamd_batch_job_t));
-    int *job_set = (int*)abamd_malloc((size_t)n_sets * sizeof(int));
-    pthread_t *tids = (pthread_t*)abamd_malloc((size_t)n_host_threads * sizeof(pthread_t));
 
     /* GPU memory budget per launch (arena is the dominant term); default
      * mirrors the shim's 70%-of-free-HBM reservation */
@@ -341,6 +338,6 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
         free(sets[i].weight_buf);
         abpoa_free(sets[i].ab);
     }
-    free(sets); free(jobs); free(job_set); free(tids);
+    free(sets);
     return 0;
 }

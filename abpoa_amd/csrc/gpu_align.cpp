@@ -177,7 +177,7 @@ struct JobPack {
     size_t o_ml, o_mr, o_meta;
     int64_t arena_off;      /* cells */
     size_t slab_base;       /* this job's base offset in the device slab */
-    int cigar_off;          /* entries */
+    int64_t cigar_off;      /* entries */
     int n_rows, qlen;
     int n_pre, n_out;
 };
@@ -578,11 +578,16 @@ static int batch_launch(GpuCtx &C, PendingBatch &PB) {
         }
         D.arena.ensure((size_t)arena_cells * planes * ssz, 1, g_arena_cap.load());
 
-        /* cigar buffers */
-        int cig_total = 0;
+        /* cigar buffers. Deterministic bound on backtrack entries: every
+         * M/I step consumes one query base (<= qlen total), every D step
+         * strictly decreases the topo row (<= n_rows total), plus the
+         * initial/terminal soft pushes — so qlen + n_rows + margin can
+         * never overflow (graphs with far more rows than qlen included,
+         * e.g. incremental restore + short reads). */
+        int64_t cig_total = 0;
         for (int i = 0; i < n_jobs; ++i) {
             packs[i].cigar_off = cig_total;
-            cig_total += 2 * packs[i].qlen + 1024;
+            cig_total += (int64_t)packs[i].qlen + packs[i].n_rows + 64;
         }
         D.cigars.ensure((size_t)cig_total * 8);
 
@@ -610,7 +615,7 @@ static int batch_launch(GpuCtx &C, PendingBatch &PB) {
             jb.arena = (uint8_t*)D.arena.p + (size_t)P.arena_off * planes * ssz;
             jb.arena_cap = arena_est[i];
             jb.cigar = (uint64_t*)D.cigars.p + P.cigar_off;
-            jb.cigar_cap = 2 * P.qlen + 1024;
+            jb.cigar_cap = P.qlen + P.n_rows + 64;
             PB.hjobs[i] = jb;
         }
         D.jobs.ensure((size_t)n_jobs * sizeof(abamd_gpu_job_t));

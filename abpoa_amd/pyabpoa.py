@@ -19,6 +19,20 @@ from collections import defaultdict as dd
 _libc = ctypes.CDLL(None)
 _libc.free.argtypes = [ctypes.c_void_p]
 _libc.free.restype = None
+_libc.strdup.argtypes = [ctypes.c_char_p]
+_libc.strdup.restype = ctypes.c_void_p
+
+
+def _set_cstr(para, attr, value):
+    """Assign a C-owned string to an abpoa_para_t filename field.
+
+    abpoa_free_para() (abamd_para.c:131-136, matching the reference) calls
+    free() on mat_fn/out_pog/incr_fn, so these fields must never point into
+    Python-owned ctypes buffers — they are C-malloc'd copies (libc strdup),
+    exactly like the reference pyx's malloc+strcpy. Frees any previous value.
+    """
+    _libc.free(getattr(para, attr))
+    setattr(para, attr, _libc.strdup(value) if value is not None else None)
 
 from . import lib as _load_lib
 from . import ConsT
@@ -33,7 +47,7 @@ class ParaT(ctypes.Structure):
     """Mirror of abpoa_para_t (include/abpoa_amd.h); layout verified at
     import against abpoa_init_para()'s known defaults."""
     _fields_ = [
-        ("m", ctypes.c_int), ("mat", ctypes.POINTER(ctypes.c_int)), ("mat_fn", ctypes.c_char_p),
+        ("m", ctypes.c_int), ("mat", ctypes.POINTER(ctypes.c_int)), ("mat_fn", ctypes.c_void_p),
         ("use_score_matrix", ctypes.c_int),
         ("match", ctypes.c_int), ("max_mat", ctypes.c_int), ("mismatch", ctypes.c_int),
         ("min_mis", ctypes.c_int), ("gap_open1", ctypes.c_int), ("gap_open2", ctypes.c_int),
@@ -50,7 +64,7 @@ class ParaT(ctypes.Structure):
         ("sub_aln", ctypes.c_uint8, 1), ("use_qv", ctypes.c_uint8, 1),
         ("disable_seeding", ctypes.c_uint8, 1), ("progressive_poa", ctypes.c_uint8, 1),
         ("put_gap_on_right", ctypes.c_uint8, 1), ("put_gap_at_end", ctypes.c_uint8, 1),
-        ("incr_fn", ctypes.c_char_p), ("out_pog", ctypes.c_char_p),
+        ("incr_fn", ctypes.c_void_p), ("out_pog", ctypes.c_void_p),
         ("align_mode", ctypes.c_int), ("gap_mode", ctypes.c_int),
         ("max_n_cons", ctypes.c_int), ("cons_algrm", ctypes.c_int),
         ("min_freq", ctypes.c_double),
@@ -219,8 +233,7 @@ class msa_aligner:
             if isinstance(score_matrix, str):
                 score_matrix = score_matrix.encode()
             p.use_score_matrix = 1
-            self._mat_fn = ctypes.create_string_buffer(score_matrix)
-            p.mat_fn = ctypes.cast(self._mat_fn, ctypes.c_char_p)
+            _set_cstr(p, "mat_fn", score_matrix)
         p.gap_open1, p.gap_open2 = gap_open1, gap_open2
         p.gap_ext1, p.gap_ext2 = gap_ext1, gap_ext2
         p.ret_cigar = 1
@@ -320,22 +333,20 @@ class msa_aligner:
         if out_pog:
             if isinstance(out_pog, str):
                 out_pog = out_pog.encode()
-            self._pog = ctypes.create_string_buffer(out_pog)
-            p.out_pog = ctypes.cast(self._pog, ctypes.c_char_p)
+            _set_cstr(p, "out_pog", out_pog)
         else:
-            p.out_pog = None
+            _set_cstr(p, "out_pog", None)
         L.abpoa_post_set_para(self._para)
         L.abpoa_reset(self.ab, self._para, len(seqs[0]))
         if incr_fn:
             if isinstance(incr_fn, str):
                 incr_fn = incr_fn.encode()
-            self._incr = ctypes.create_string_buffer(incr_fn)
-            p.incr_fn = ctypes.cast(self._incr, ctypes.c_char_p)
+            _set_cstr(p, "incr_fn", incr_fn)
             L.abpoa_restore_graph(self.ab, self._para)
             exist_n = self.ab.contents.abs.contents.n_seq
             tot_n += exist_n
         else:
-            p.incr_fn = None
+            _set_cstr(p, "incr_fn", None)
         self.ab.contents.abs.contents.n_seq += seq_n
         self._add_sequences(seqs, qscores, exist_n, tot_n)
         if p.out_msa:
@@ -365,13 +376,12 @@ class msa_aligner:
         if incr_fn:
             if isinstance(incr_fn, str):
                 incr_fn = incr_fn.encode()
-            self._incr = ctypes.create_string_buffer(incr_fn)
-            p.incr_fn = ctypes.cast(self._incr, ctypes.c_char_p)
+            _set_cstr(p, "incr_fn", incr_fn)
             L.abpoa_restore_graph(self.ab, self._para)
             exist_n = self.ab.contents.abs.contents.n_seq
             tot_n += exist_n
         else:
-            p.incr_fn = None
+            _set_cstr(p, "incr_fn", None)
         self.ab.contents.abs.contents.n_seq += seq_n
         self._add_sequences(seqs, qscores, exist_n, tot_n)
         return self
