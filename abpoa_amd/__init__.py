@@ -59,9 +59,11 @@ class ConsT(ctypes.Structure):
 CONS_CB = ctypes.CFUNCTYPE(None, ctypes.c_int, ctypes.POINTER(ConsT), ctypes.c_void_p)
 
 
-def msa_batch_consensus(sets, n_threads=4):
+def msa_batch_consensus(sets, n_threads=4, cons_algrm=None):
     """Run the batched GPU driver over `sets` (list of list of bytes, codes
-    0..3) and return each set's consensus as an ACGT string."""
+    0..3) and return each set's consensus as an ACGT string.
+    cons_algrm: None = library default (HB), "MF" = most-frequent (exercises
+    the per-edge read-id bitsets on the device-resident path)."""
     L = lib()
     L.abpoa_init_para.restype = ctypes.c_void_p
     L.abpoa_post_set_para.argtypes = [ctypes.c_void_p]
@@ -73,6 +75,10 @@ def msa_batch_consensus(sets, n_threads=4):
         ctypes.POINTER(ctypes.POINTER(ctypes.POINTER(ctypes.c_uint8))),
         CONS_CB, ctypes.c_void_p, ctypes.c_int]
     para = L.abpoa_init_para()
+    if cons_algrm is not None:
+        from .pyabpoa import ParaT, ABPOA_HB, ABPOA_MF
+        p = ctypes.cast(para, ctypes.POINTER(ParaT)).contents
+        p.cons_algrm = {"HB": ABPOA_HB, "MF": ABPOA_MF}[cons_algrm.upper()]
     L.abpoa_post_set_para(para)
 
     n_sets = len(sets)

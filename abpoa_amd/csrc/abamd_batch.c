@@ -113,10 +113,28 @@ static void cons_worker(void *arg, int tid, int nthr) {
     }
 }
 
+/* device-resident driver (gpu_batch_resident.cpp; stubbed out in the
+ * CPU-only test build, where this host-fold driver always runs) */
+int abamd_batch_resident_supported(const abpoa_para_t *abpt);
+int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
+                                 const int *const *seq_lens, const uint8_t *const *const *seqs,
+                                 abpoa_amd_cons_cb cb, void *user, int n_host_threads);
+
 int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
                         const int *const *seq_lens, const uint8_t *const *const *seqs,
                         abpoa_amd_cons_cb cb, void *user, int n_host_threads) {
     if (n_sets <= 0) return 0;
+    if (abamd_batch_resident_supported(abpt)) {
+        /* device-resident graphs need every read non-empty (the reference
+         * fold indexes weight[len-1]); scan is trivial vs one alignment */
+        int ok = 1, s, i;
+        for (s = 0; s < n_sets && ok; ++s)
+            for (i = 0; i < n_seqs[s]; ++i)
+                if (seq_lens[s][i] < 1) { ok = 0; break; }
+        if (ok)
+            return abpoa_amd_msa_batch_resident(abpt, n_sets, n_seqs, seq_lens, seqs,
+                                                cb, user, n_host_threads);
+    }
     {
         /* fold thread-count override: at ~250 threads the concurrent folds'
          * ~10 MB graph working sets thrash the LLC (measured ~20x per-call

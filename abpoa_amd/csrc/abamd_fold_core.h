@@ -5,7 +5,7 @@
 
 #include <stdint.h>
 
-typedef struct {
+typedef struct flat_graph_t {
     int node_n, node_cap;
     uint8_t *base;               /* [node_cap] */
     int *n_read, *n_span_read;   /* [node_cap] */
@@ -22,6 +22,11 @@ typedef struct {
     int *aln_id, *aln_next;                         /* [aln_cap] */
 } flat_graph_t;
 
+/* Device translation units (gpu_fold.hip) include this header for the
+ * flat_graph_t layout only: the function bodies there are __device__ builds
+ * of abamd_fold_core.inc, and host declarations of the same names would
+ * clash with them. */
+#ifndef ABAMD_FC_NO_HOST_DECLS
 #ifdef __cplusplus
 extern "C" {
 #endif
@@ -45,8 +50,16 @@ void abamd_flat_apply_alignment(flat_graph_t *fg, int beg_node_id, int end_node_
                                 int *qpos_to_node_id, int n_cigar, const abpoa_cigar_t *cig,
                                 int read_id, int add_read_id, int inc_both_ends);
 
+#ifdef ABPOA_AMD_H
+/* Rebuild a pointer graph from a (host copy of a) flat graph and
+ * topo-sort it — the device-resident batch driver's hand-off to the host
+ * consensus path (abamd_graph.c). `ab` must be fresh. */
+void abamd_graph_from_flat(abpoa_t *ab, const flat_graph_t *fg, abpoa_para_t *abpt, int read_ids_n);
+#endif
+
 #ifdef __cplusplus
 }
 #endif
+#endif /* ABAMD_FC_NO_HOST_DECLS */
 
 #endif

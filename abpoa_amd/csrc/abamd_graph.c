@@ -653,3 +653,93 @@ void abpoa_subgraph_nodes(abpoa_t *ab, abpoa_para_t *abpt, int inc_beg, int inc_
     *exc_beg = g->index_to_node_id[eb];
     *exc_end = g->index_to_node_id[ee];
 }
+
+/* ---------------- flat-graph import (device-resident batch path) ----------------
+ *
+ * Rebuild the pointer graph from a host copy of a device-resident flat graph
+ * (abamd_fold_core.h) after the GPU has folded every read of a set. The flat
+ * fold is CPU-twin-proven bit-equal to the pointer fold (tests/test_fold_twin,
+ * device-validated by the gpu fold test), so walking its chains in order
+ * reproduces the exact adjacency order — including the weight-sort order the
+ * device applied — and the final abpoa_topological_sort here re-derives the
+ * index arrays with the identical algorithm, leaving the graph byte-equal to
+ * what the host fold path would have produced. `ab` must be fresh
+ * (abpoa_init, untouched). read_ids_n > 0 copies per-out-edge read bitsets
+ * (fg->rid_n must equal read_ids_n). */
+#include "abamd_fold_core.h"
+
+void abamd_graph_from_flat(abpoa_t *ab, const flat_graph_t *fg, abpoa_para_t *abpt, int read_ids_n) {
+    abpoa_graph_t *g = ab->abg;
+    int id, e, n;
+    if (fg->node_n > g->node_m) {
+        int node_m = ab_round_up_pow2_32(fg->node_n);
+        g->node = (abpoa_node_t*)abamd_realloc(g->node, (size_t)node_m * sizeof(abpoa_node_t));
+        for (id = g->node_m; id < node_m; ++id) init_node(&g->node[id], id);
+        g->node_m = node_m;
+    }
+    if (read_ids_n > 0 && fg->rid_n != read_ids_n)
+        abamd_fatal("abamd_graph_from_flat", "rid_n %d != read_ids_n %d", fg->rid_n, read_ids_n);
+    for (id = 0; id < fg->node_n; ++id) {
+        abpoa_node_t *v = &g->node[id];
+        v->node_id = id;
+        v->base = fg->base[id];
+        v->n_read = fg->n_read[id];
+        v->n_span_read = fg->n_span_read[id];
+        /* in edges, chain order */
+        n = 0;
+        for (e = fg->in_head[id]; e != -1; e = fg->in_next[e]) ++n;
+        if (n > v->in_edge_m) {
+            v->in_id = (int*)abamd_realloc(v->in_id, (size_t)n * sizeof(int));
+            v->in_edge_weight = (int*)abamd_realloc(v->in_edge_weight, (size_t)n * sizeof(int));
+            v->in_edge_m = n;
+        }
+        v->in_edge_n = n;
+        n = 0;
+        for (e = fg->in_head[id]; e != -1; e = fg->in_next[e]) {
+            v->in_id[n] = fg->in_to[e];
+            v->in_edge_weight[n] = fg->in_w[e];
+            ++n;
+        }
+        /* out edges (+ read-id bitsets travelling with each edge) */
+        n = 0;
+        for (e = fg->out_head[id]; e != -1; e = fg->out_next[e]) ++n;
+        if (n > v->out_edge_m) {
+            int i, old = v->out_edge_m;
+            v->out_id = (int*)abamd_realloc(v->out_id, (size_t)n * sizeof(int));
+            v->out_edge_weight = (int*)abamd_realloc(v->out_edge_weight, (size_t)n * sizeof(int));
+            if (read_ids_n > 0 || v->read_ids) {
+                v->read_ids = (uint64_t**)abamd_realloc(v->read_ids, (size_t)n * sizeof(uint64_t*));
+                for (i = old; i < n; ++i) v->read_ids[i] = NULL;
+            }
+            v->out_edge_m = n;
+        }
+        v->out_edge_n = n;
+        n = 0;
+        for (e = fg->out_head[id]; e != -1; e = fg->out_next[e]) {
+            v->out_id[n] = fg->out_to[e];
+            v->out_edge_weight[n] = fg->out_w[e];
+            if (read_ids_n > 0) {
+                if (!v->read_ids[n])
+                    v->read_ids[n] = (uint64_t*)abamd_malloc((size_t)read_ids_n * sizeof(uint64_t));
+                memcpy(v->read_ids[n], fg->rid_pool + (size_t)e * fg->rid_n,
+                       (size_t)read_ids_n * sizeof(uint64_t));
+            }
+            ++n;
+        }
+        if (read_ids_n > 0 && v->out_edge_n > 0) v->read_ids_n = read_ids_n;
+        /* aligned (mismatch-column) group, chain order */
+        n = 0;
+        for (e = fg->aln_head[id]; e != -1; e = fg->aln_next[e]) ++n;
+        if (n > v->aligned_node_m) {
+            v->aligned_node_id = (int*)abamd_realloc(v->aligned_node_id, (size_t)n * sizeof(int));
+            v->aligned_node_m = n;
+        }
+        v->aligned_node_n = n;
+        n = 0;
+        for (e = fg->aln_head[id]; e != -1; e = fg->aln_next[e])
+            v->aligned_node_id[n++] = fg->aln_id[e];
+    }
+    g->node_n = fg->node_n;
+    g->is_topological_sorted = g->is_called_cons = g->is_set_msa_rank = 0;
+    abpoa_topological_sort(g, abpt);
+}

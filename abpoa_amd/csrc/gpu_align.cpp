@@ -63,6 +63,15 @@ extern "C" void abpoa_amd_reset_stats(void) {
     g_pack_ns = 0; g_stage_ns = 0; g_gpu_ns = 0; g_unpack_ns = 0;
 }
 
+/* shared-counter hooks for the device-resident batch driver
+ * (gpu_batch_resident.cpp), so abpoa_amd_get_stats covers both paths */
+extern "C" void abamd_stats_add_cells(uint64_t cells, uint64_t alg_bytes) {
+    g_dp_cells += cells; g_alg_bytes += alg_bytes;
+}
+extern "C" void abamd_stats_add_kernel(uint64_t ns) {
+    g_kernel_ns += ns; g_launches += 1;
+}
+
 /* ---------------- device context (per thread) ---------------- */
 
 namespace {
@@ -365,6 +374,11 @@ static void pick_width(abpoa_para_t *abpt, int qlen, int gn, int *bits, int *inf
 }
 
 } // namespace
+
+/* score-width pick, shared with the device-resident batch driver */
+extern "C" void abamd_pick_width(abpoa_para_t *abpt, int qlen, int gn, int *bits, int *inf_min) {
+    pick_width(abpt, qlen, gn, bits, inf_min);
+}
 
 /* Pipelined batch interface: begin() packs, uploads and launches without
  * waiting; finish() synchronizes and unpacks. The caller overlaps host-side

@@ -55,3 +55,15 @@ int abamd_gpu_align_batch_slot(abamd_batch_job_t *batch, int n_jobs, int slot) {
 
 int64_t abamd_gpu_free_mem(void) { return 0; }
 void abamd_gpu_set_arena_cap(uint64_t bytes) { (void)bytes; }
+
+/* device-resident batch driver: GPU build only — the CPU test build always
+ * takes the host-fold driver */
+int abamd_batch_resident_supported(const abpoa_para_t *abpt) { (void)abpt; return 0; }
+int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
+                                 const int *const *seq_lens, const uint8_t *const *const *seqs,
+                                 abpoa_amd_cons_cb cb, void *user, int n_host_threads) {
+    (void)abpt; (void)n_sets; (void)n_seqs; (void)seq_lens; (void)seqs;
+    (void)cb; (void)user; (void)n_host_threads;
+    abamd_fatal("abpoa_amd", "device-resident batch driver is not in the CPU test build");
+    return -1;
+}

@@ -19,7 +19,7 @@ def test_batch_matches_cli(tmp_path):
     import importlib
     bench = importlib.import_module("bench")
     sets = bench.gen_sets(rng, 4, depth=20, qlen=800)
-    # batched GPU consensus
+    # batched GPU consensus (device-resident driver: graphs fold on the GPU)
     cons = abpoa_amd.msa_batch_consensus(sets, n_threads=2)
     # sequential CLI on the same sets
     code2ch = "ACGT"
@@ -31,3 +31,45 @@ def test_batch_matches_cli(tmp_path):
         out = run_stdout([GPU_BIN, str(fa)]).decode()
         seq = "".join(out.splitlines()[1:])
         assert cons[i] == seq, "batch/CLI consensus mismatch on set %d" % i
+
+
+def test_resident_matches_hostfold():
+    """The device-resident batch driver (GPU fold + device-built CSR) must be
+    byte-identical to the round-1 host-fold driver on the same inputs —
+    including the MF-consensus config, which exercises the per-edge read-id
+    bitsets maintained on device."""
+    sys.path.insert(0, ROOT)
+    import abpoa_amd
+    import importlib
+    import numpy as np
+    bench = importlib.import_module("bench")
+    rng = np.random.default_rng(321)
+    sets = bench.gen_sets(rng, 6, depth=15, qlen=600)
+    for algrm in (None, "MF"):
+        os.environ.pop("ABPOA_AMD_HOST_FOLD", None)
+        resident = abpoa_amd.msa_batch_consensus(sets, n_threads=2, cons_algrm=algrm)
+        os.environ["ABPOA_AMD_HOST_FOLD"] = "1"
+        try:
+            hostfold = abpoa_amd.msa_batch_consensus(sets, n_threads=2, cons_algrm=algrm)
+        finally:
+            os.environ.pop("ABPOA_AMD_HOST_FOLD", None)
+        assert resident == hostfold, "resident/host-fold mismatch (cons_algrm=%s)" % algrm
+
+
+def test_resident_pool_expansion():
+    """Tiny ABPOA_AMD_NODE_ALPHA forces graph-pool overflows mid-run: the
+    expand + device-to-device move + refold path must still be byte-exact."""
+    sys.path.insert(0, ROOT)
+    import abpoa_amd
+    import importlib
+    import numpy as np
+    bench = importlib.import_module("bench")
+    rng = np.random.default_rng(77)
+    sets = bench.gen_sets(rng, 3, depth=20, qlen=500)
+    os.environ["ABPOA_AMD_NODE_ALPHA"] = "0.02"
+    try:
+        tight = abpoa_amd.msa_batch_consensus(sets, n_threads=2)
+    finally:
+        os.environ.pop("ABPOA_AMD_NODE_ALPHA", None)
+    roomy = abpoa_amd.msa_batch_consensus(sets, n_threads=2)
+    assert tight == roomy, "pool-expansion path changed the consensus"
