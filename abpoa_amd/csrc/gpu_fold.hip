@@ -78,6 +78,181 @@ extern "C" __global__ void abamd_fold_kernel(fold_job_t *jobs, int n_jobs) {
 }
 
 /* ------------------------------------------------------------------ */
+/* Wave-parallel variants of the order-insensitive derived passes.     */
+/*                                                                     */
+/* The mutation (apply) and the Kahn BFS stay serial on lane 0 — their */
+/* ORDER is observable (queue order defines the topo index; the fold   */
+/* order defines edge append order). Everything else is a pure         */
+/* function of the graph whose outputs the serial twin provably        */
+/* produces; these versions compute the same values with all 64 lanes  */
+/* (per-node independence, wave prefix scans), cutting the per-set     */
+/* fold latency that otherwise matches the DP kernel's whole wall.     */
+/* Byte-parity with the host path is enforced end-to-end by            */
+/* tests/test_batch_gpu.py (resident vs host-fold vs sequential CLI).  */
+/* ------------------------------------------------------------------ */
+
+#define FOLD_WAVE 64
+
+/* per-node in-degrees into in_deg[]; replaces the serial count loop of
+ * abamd_flat_topo_index so the lane-0 BFS starts from precomputed degrees */
+__device__ static void dev_par_in_deg(const flat_graph_t *fg, int *in_deg, int lane) {
+    for (int i = lane; i < fg->node_n; i += FOLD_WAVE) {
+        int d = 0;
+        for (int e = fg->in_head[i]; e != -1; e = fg->in_next[e]) ++d;
+        in_deg[i] = d;
+    }
+}
+
+/* Kahn BFS with whole-aligned-group queue entry (identical order to
+ * abamd_flat_topo_index, which the twin test pins); in_deg precomputed */
+__device__ static void dev_topo_bfs(const flat_graph_t *fg, int *index_to_node_id,
+                                    int *node_id_to_index, int *in_deg, int *q) {
+    int e, a, cur, index = 0, qh = 0, qt = 0;
+    q[qt++] = 0; /* SRC */
+    while (qh < qt) {
+        cur = q[qh++];
+        index_to_node_id[index] = cur;
+        node_id_to_index[cur] = index++;
+        if (cur == 1 /* SINK */) return;
+        for (e = fg->out_head[cur]; e != -1; e = fg->out_next[e]) {
+            int out = fg->out_to[e];
+            if (--in_deg[out] == 0) {
+                int ready = 1;
+                for (a = fg->aln_head[out]; a != -1; a = fg->aln_next[a])
+                    if (in_deg[fg->aln_id[a]] != 0) { ready = 0; break; }
+                if (!ready) continue;
+                q[qt++] = out;
+                for (a = fg->aln_head[out]; a != -1; a = fg->aln_next[a])
+                    q[qt++] = fg->aln_id[a];
+            }
+        }
+    }
+    abort(); /* not a connected DAG: driver bug */
+}
+
+/* weight-descending adjacency sort — per-node work is independent, so nodes
+ * stride across lanes; the per-node pairwise-swap pattern (tie semantics)
+ * matches abamd_flat_sort_adjacency exactly */
+__device__ static void dev_par_sort_adjacency(flat_graph_t *fg, int lane) {
+    int scratch[1024];
+    for (int i = lane; i < fg->node_n; i += FOLD_WAVE) {
+        int j, k, e, n;
+        n = 0;
+        for (e = fg->in_head[i]; e != -1; e = fg->in_next[e]) {
+            if (n >= 1024) abort();
+            scratch[n++] = e;
+        }
+        for (j = 0; j < n - 1; ++j)
+            for (k = j + 1; k < n; ++k)
+                if (fg->in_w[scratch[j]] < fg->in_w[scratch[k]]) {
+                    int t = scratch[j]; scratch[j] = scratch[k]; scratch[k] = t;
+                }
+        fg->in_head[i] = n ? scratch[0] : -1;
+        for (j = 0; j + 1 < n; ++j) fg->in_next[scratch[j]] = scratch[j + 1];
+        if (n) { fg->in_next[scratch[n - 1]] = -1; fg->in_tail[i] = scratch[n - 1]; }
+        n = 0;
+        for (e = fg->out_head[i]; e != -1; e = fg->out_next[e]) {
+            if (n >= 1024) abort();
+            scratch[n++] = e;
+        }
+        for (j = 0; j < n - 1; ++j)
+            for (k = j + 1; k < n; ++k)
+                if (fg->out_w[scratch[j]] < fg->out_w[scratch[k]]) {
+                    int t = scratch[j]; scratch[j] = scratch[k]; scratch[k] = t;
+                }
+        fg->out_head[i] = n ? scratch[0] : -1;
+        for (j = 0; j + 1 < n; ++j) fg->out_next[scratch[j]] = scratch[j + 1];
+        if (n) { fg->out_next[scratch[n - 1]] = -1; fg->out_tail[i] = scratch[n - 1]; }
+    }
+}
+
+/* max_remain by REVERSE topo-index scan: every successor has a higher index,
+ * so its value is final when a node is visited — the same pure function the
+ * twin's reverse BFS computes (max-weight out-edge, strict >, chain-order
+ * tie-break), minus the queue machinery */
+__device__ static void dev_remain_reverse(const flat_graph_t *fg, const int *index_to_node_id,
+                                          int sink_index, int *max_remain) {
+    for (int k = sink_index; k >= 0; --k) {
+        int id = index_to_node_id[k];
+        if (id == 1 /* SINK */) { max_remain[id] = -1; continue; }
+        int max_w = INT32_MIN, max_id = -1;
+        for (int e = fg->out_head[id]; e != -1; e = fg->out_next[e])
+            if (fg->out_w[e] > max_w) { max_w = fg->out_w[e]; max_id = fg->out_to[e]; }
+        max_remain[id] = max_remain[max_id] + 1;
+    }
+}
+
+/* span counts: disjoint increments across lanes */
+__device__ static void dev_par_n_span(flat_graph_t *fg, const int *index_to_node_id,
+                                      int src_index, int sink_index, int lane) {
+    for (int i = src_index + 1 + lane; i < sink_index; i += FOLD_WAVE)
+        fg->n_span_read[index_to_node_id[i]] += 1;
+    if (lane == 0) {
+        fg->n_span_read[0] += 1;
+        fg->n_span_read[1] += 1;
+    }
+}
+
+/* DP-row CSR materialization (mirrors abamd_flat_build_rows / pack_job):
+ * lane-parallel degree count, wave prefix scan into the offset arrays,
+ * lane-parallel fill. deg scratch: scratch[0..nc) = pre, scratch[nc..2nc) =
+ * out (the BFS queue is done with it by now). Returns n_rows on lane 0. */
+__device__ static int dev_par_build_rows(const flat_graph_t *fg,
+                                         const int *index_to_node_id,
+                                         const int *node_id_to_index,
+                                         const int *max_remain, int use_remain,
+                                         uint8_t *row_base, int *row_node_id,
+                                         int *pre_off, int *out_off, int *remain,
+                                         int *pre_idx, int *out_idx,
+                                         int *scratch, int node_cap, int lane) {
+    const int n_rows = node_id_to_index[1 /* SINK */] + 1;
+    int *dpre = scratch, *dout = scratch + node_cap;
+    for (int r = lane; r < n_rows; r += FOLD_WAVE) {
+        int nid = index_to_node_id[r];
+        row_base[r] = fg->base[nid];
+        row_node_id[r] = nid;
+        remain[r] = use_remain ? max_remain[nid] : 0;
+        int e, dp = 0, dq = 0;
+        if (r > 0)
+            for (e = fg->in_head[nid]; e != -1; e = fg->in_next[e]) ++dp;
+        for (e = fg->out_head[nid]; e != -1; e = fg->out_next[e]) ++dq;
+        dpre[r] = dp; dout[r] = dq;
+    }
+    __syncthreads();
+    /* wave prefix scan: lane l owns rows [l*B, (l+1)*B) */
+    const int B = (n_rows + FOLD_WAVE) / FOLD_WAVE; /* +1 slot for the totals */
+    int sum_pre = 0, sum_out = 0;
+    {
+        const int r0 = lane * B, r1 = min(r0 + B, n_rows);
+        for (int r = r0; r < r1; ++r) { sum_pre += dpre[r]; sum_out += dout[r]; }
+        /* exclusive scan of per-lane sums */
+        int ex_pre = sum_pre, ex_out = sum_out;
+        for (int s = 1; s < FOLD_WAVE; s <<= 1) {
+            int vp = __shfl_up(ex_pre, s), vo = __shfl_up(ex_out, s);
+            if (lane >= s) { ex_pre += vp; ex_out += vo; }
+        }
+        ex_pre -= sum_pre; ex_out -= sum_out; /* inclusive -> exclusive */
+        int ap = ex_pre, ao = ex_out;
+        for (int r = r0; r < r1; ++r) {
+            pre_off[r] = ap; out_off[r] = ao;
+            ap += dpre[r]; ao += dout[r];
+        }
+        if (r1 == n_rows && r0 <= n_rows) { pre_off[n_rows] = ap; out_off[n_rows] = ao; }
+    }
+    __syncthreads();
+    for (int r = lane; r < n_rows; r += FOLD_WAVE) {
+        int nid = index_to_node_id[r];
+        int e, np = pre_off[r], no = out_off[r];
+        if (r > 0)
+            for (e = fg->in_head[nid]; e != -1; e = fg->in_next[e])
+                pre_idx[np++] = node_id_to_index[fg->in_to[e]];
+        for (e = fg->out_head[nid]; e != -1; e = fg->out_next[e])
+            out_idx[no++] = node_id_to_index[fg->out_to[e]];
+    }
+    return n_rows;
+}
+
+/* ------------------------------------------------------------------ */
 /* Production per-round kernel.                                        */
 /* ------------------------------------------------------------------ */
 
@@ -151,31 +326,43 @@ void abamd_fold_round_kernel(abamd_fold_round_job_t *jobs, int n_jobs) {
     }
     __syncthreads();
 
-    if (lane != 0) return;
-
-    abamd_flat_apply_alignment(g, ABPOA_SRC_NODE_ID, ABPOA_SINK_NODE_ID,
-                               job->seq, job->weight, job->seq_l, NULL,
-                               n_cigar, job->cigar, job->read_id,
-                               job->add_read_id, 1);
-    abamd_flat_topo_index(g, job->index_to_node_id, job->node_id_to_index, job->scratch);
-    abamd_flat_sort_adjacency(g);
-    if (job->use_remain)
-        abamd_flat_remain(g, job->max_remain, job->scratch);
-    abamd_flat_update_n_span(g, job->index_to_node_id, job->node_id_to_index, 1);
-
-    int n_rows = abamd_flat_build_rows(g, job->index_to_node_id, job->node_id_to_index,
-                                       job->max_remain, job->use_remain,
-                                       job->row_base, job->row_node_id,
-                                       job->pre_off, job->out_off, job->row_remain,
-                                       job->pre_idx, job->out_idx);
-    abamd_fold_out_t *o = job->out;
-    o->status = ABAMD_FOLD_OK;
-    o->node_n = g->node_n;
-    o->edge_n_in = g->edge_n_in; o->edge_n_out = g->edge_n_out;
-    o->aln_n = g->aln_n;
-    o->n_rows = n_rows;
-    o->n_pre = job->pre_off[n_rows];
-    o->n_out = job->out_off[n_rows];
+    if (lane == 0)
+        abamd_flat_apply_alignment(g, ABPOA_SRC_NODE_ID, ABPOA_SINK_NODE_ID,
+                                   job->seq, job->weight, job->seq_l, NULL,
+                                   n_cigar, job->cigar, job->read_id,
+                                   job->add_read_id, 1);
+    __syncthreads();
+    dev_par_in_deg(g, job->scratch, lane);
+    __syncthreads();
+    if (lane == 0)
+        dev_topo_bfs(g, job->index_to_node_id, job->node_id_to_index,
+                     job->scratch, job->scratch + g->node_cap);
+    __syncthreads();
+    dev_par_sort_adjacency(g, lane);
+    __syncthreads();
+    const int sink_index = job->node_id_to_index[1 /* SINK */];
+    if (job->use_remain && lane == 0)
+        dev_remain_reverse(g, job->index_to_node_id, sink_index, job->max_remain);
+    dev_par_n_span(g, job->index_to_node_id, job->node_id_to_index[0 /* SRC */],
+                   sink_index, lane);
+    __syncthreads();
+    int n_rows = dev_par_build_rows(g, job->index_to_node_id, job->node_id_to_index,
+                                    job->max_remain, job->use_remain,
+                                    job->row_base, job->row_node_id,
+                                    job->pre_off, job->out_off, job->row_remain,
+                                    job->pre_idx, job->out_idx,
+                                    job->scratch, g->node_cap, lane);
+    __syncthreads();
+    if (lane == 0) {
+        abamd_fold_out_t *o = job->out;
+        o->status = ABAMD_FOLD_OK;
+        o->node_n = g->node_n;
+        o->edge_n_in = g->edge_n_in; o->edge_n_out = g->edge_n_out;
+        o->aln_n = g->aln_n;
+        o->n_rows = n_rows;
+        o->n_pre = job->pre_off[n_rows];
+        o->n_out = job->out_off[n_rows];
+    }
 }
 
 extern "C" void abamd_launch_fold_round(const abamd_fold_round_job_t *dev_jobs,
