@@ -77,6 +77,22 @@ extern "C" __global__ void abamd_fold_kernel(fold_job_t *jobs, int n_jobs) {
     }
 }
 
+#ifdef ABAMD_KPROF
+__device__ unsigned long long abamd_kprof_fold[16];
+extern "C" void abamd_kprof_fold_fetch(unsigned long long *out) {
+    (void)hipMemcpyFromSymbol(out, HIP_SYMBOL(abamd_kprof_fold), sizeof(abamd_kprof_fold));
+}
+extern "C" void abamd_kprof_fold_reset(void) {
+    unsigned long long z[16] = {0};
+    (void)hipMemcpyToSymbol(HIP_SYMBOL(abamd_kprof_fold), z, sizeof(z));
+}
+#define FKPROF_T(v) unsigned long long v = __builtin_readcyclecounter()
+#define FKPROF_ACC(slot, a, b) if (lane == 0) atomicAdd(&abamd_kprof_fold[slot], (b) - (a))
+#else
+#define FKPROF_T(v)
+#define FKPROF_ACC(slot, a, b)
+#endif
+
 /* ------------------------------------------------------------------ */
 /* Wave-parallel variants of the order-insensitive derived passes.     */
 /*                                                                     */
@@ -261,7 +277,12 @@ void abamd_fold_round_kernel(abamd_fold_round_job_t *jobs, int n_jobs) {
     const int j = blockIdx.x;
     if (j >= n_jobs) return;
     abamd_fold_round_job_t *job = &jobs[j];
-    flat_graph_t *g = &job->g;
+    /* by-VALUE graph descriptor: pool pointers and counters live in
+     * registers — through job->g every chain step would re-load the pool
+     * base pointer from global memory (stores to the pools could alias the
+     * jobs array in the compiler's view) */
+    flat_graph_t gl = job->g;
+    flat_graph_t *g = &gl;
     const int lane = threadIdx.x;
 
     int n_cigar = 0;
@@ -326,26 +347,47 @@ void abamd_fold_round_kernel(abamd_fold_round_job_t *jobs, int n_jobs) {
     }
     __syncthreads();
 
-    if (lane == 0)
+    FKPROF_T(ft0);
+    if (lane == 0) {
         abamd_flat_apply_alignment(g, ABPOA_SRC_NODE_ID, ABPOA_SINK_NODE_ID,
                                    job->seq, job->weight, job->seq_l, NULL,
                                    n_cigar, job->cigar, job->read_id,
                                    job->add_read_id, 1);
+        /* publish the mutated counters so every lane's local copy agrees */
+        job->g.node_n = gl.node_n;
+        job->g.edge_n_in = gl.edge_n_in;
+        job->g.edge_n_out = gl.edge_n_out;
+        job->g.aln_n = gl.aln_n;
+    }
     __syncthreads();
+    gl.node_n = job->g.node_n;
+    gl.edge_n_in = job->g.edge_n_in;
+    gl.edge_n_out = job->g.edge_n_out;
+    gl.aln_n = job->g.aln_n;
+    FKPROF_T(ft1);
+    FKPROF_ACC(1, ft0, ft1);
     dev_par_in_deg(g, job->scratch, lane);
     __syncthreads();
+    FKPROF_T(ft2);
+    FKPROF_ACC(2, ft1, ft2);
     if (lane == 0)
         dev_topo_bfs(g, job->index_to_node_id, job->node_id_to_index,
                      job->scratch, job->scratch + g->node_cap);
     __syncthreads();
+    FKPROF_T(ft3);
+    FKPROF_ACC(3, ft2, ft3);
     dev_par_sort_adjacency(g, lane);
     __syncthreads();
+    FKPROF_T(ft4);
+    FKPROF_ACC(4, ft3, ft4);
     const int sink_index = job->node_id_to_index[1 /* SINK */];
     if (job->use_remain && lane == 0)
         dev_remain_reverse(g, job->index_to_node_id, sink_index, job->max_remain);
     dev_par_n_span(g, job->index_to_node_id, job->node_id_to_index[0 /* SRC */],
                    sink_index, lane);
     __syncthreads();
+    FKPROF_T(ft5);
+    FKPROF_ACC(5, ft4, ft5);
     int n_rows = dev_par_build_rows(g, job->index_to_node_id, job->node_id_to_index,
                                     job->max_remain, job->use_remain,
                                     job->row_base, job->row_node_id,
@@ -353,6 +395,15 @@ void abamd_fold_round_kernel(abamd_fold_round_job_t *jobs, int n_jobs) {
                                     job->pre_idx, job->out_idx,
                                     job->scratch, g->node_cap, lane);
     __syncthreads();
+#ifdef ABAMD_KPROF
+    {
+        unsigned long long ft6 = __builtin_readcyclecounter();
+        if (lane == 0) {
+            atomicAdd(&abamd_kprof_fold[6], ft6 - ft5);
+            atomicAdd(&abamd_kprof_fold[0], 1ull);
+        }
+    }
+#endif
     if (lane == 0) {
         abamd_fold_out_t *o = job->out;
         o->status = ABAMD_FOLD_OK;

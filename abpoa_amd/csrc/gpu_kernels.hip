@@ -31,6 +31,25 @@
 #include <hip/hip_runtime.h>
 #include "gpu_core.h"
 
+/* Compile-time phase profiler (make kprof): accumulates s_memtime cycles per
+ * row-loop phase of the convex kernel across all jobs. Diagnostic build
+ * only — never part of the product library. */
+#ifdef ABAMD_KPROF
+__device__ unsigned long long abamd_kprof_acc[16];
+extern "C" void abamd_kprof_fetch(unsigned long long *out) {
+    (void)hipMemcpyFromSymbol(out, HIP_SYMBOL(abamd_kprof_acc), sizeof(abamd_kprof_acc));
+}
+extern "C" void abamd_kprof_reset(void) {
+    unsigned long long z[16] = {0};
+    (void)hipMemcpyToSymbol(HIP_SYMBOL(abamd_kprof_acc), z, sizeof(z));
+}
+#define KPROF_T(v) unsigned long long v = __builtin_readcyclecounter()
+#define KPROF_ACC(slot, val) if (lane == 0) atomicAdd(&abamd_kprof_acc[slot], (unsigned long long)(val))
+#else
+#define KPROF_T(v)
+#define KPROF_ACC(slot, val)
+#endif
+
 #define WAVE 64
 #define JOBS_PER_BLOCK 4
 /* LDS previous-row cache width (cells). Default bands are ~230-450 cells
@@ -116,7 +135,11 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
     }
     __syncthreads();
     if (jid >= n_jobs) return;
-    const abamd_gpu_job_t &jb = jobs[jid];
+    /* by VALUE: every field lives in registers (uniform loads become
+       s_loads into SGPRs) — a reference would re-load fields from global
+       memory inside the row loop because arena/steering stores could alias
+       the jobs array in the compiler's view */
+    const abamd_gpu_job_t jb = jobs[jid];
     abamd_gpu_res_t *res = &results[jid];
     abamd_row_meta_t *__restrict__ meta = (abamd_row_meta_t*)jb.row_meta;
 
@@ -192,23 +215,68 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         buf_cur ^= 1;
     }
 
-    /* ---- main row loop ---- */
+    /* ---- main row loop ----
+     * Per-row scalars (CSR offsets, remain, band steering, first
+     * predecessor) roll ONE ROW AHEAD: their loads issue during the
+     * previous row's chunk work, hiding the dependent-load latency that
+     * dominated round 1 (76% WAIT_ANY, profiles/r01_sq_stalls.txt). The
+     * common band push r -> r+1 is carried in registers; max_left/right
+     * only round-trip memory for out-edges that skip rows, whose values
+     * are final before the prefetch that reads them (pushes from rows
+     * < r land before row r issues row r+1's loads; row r's own push to
+     * r+1 never touches memory). */
+    int cur_pk0 = jb.pre_off[1];
+    int cur_pk1 = jb.pre_off[2 <= n_rows ? 2 : n_rows];
+    int cur_oo0 = jb.out_off[1];
+    int cur_oo1 = jb.out_off[2 <= n_rows ? 2 : n_rows];
+    int cur_remain = jb.max_remain[1];
+    int cur_ml_mem = jb.max_left[1];
+    int cur_mr_mem = jb.max_right[1];
+    int cur_pidx0 = jb.pre_idx[cur_pk0];
+    int cur_ps0 = jb.pre_ps[cur_pk0];
+    int push_ml = 0x7fffffff, push_mr = -0x7fffffff;
+#ifdef ABAMD_KPROF
+    unsigned long long kp_band = 0, kp_chunk = 0, kp_epi = 0, kp_rows = 0, kp_chunks = 0;
+#endif
     for (int r = 1; r < n_rows - 1; ++r) {
-        const int pk0 = jb.pre_off[r], pk1 = jb.pre_off[r + 1];
+        KPROF_T(kt0);
+        const int pk0 = cur_pk0, pk1 = cur_pk1;
+        const int oo0 = cur_oo0, oo1 = cur_oo1;
+        const int row_remain = cur_remain;
+        const int pidx0 = cur_pidx0;
+        const S ps0 = (S)cur_ps0;
+        const int ml_eff = cur_ml_mem < push_ml ? cur_ml_mem : push_ml;
+        const int mr_eff = cur_mr_mem > push_mr ? cur_mr_mem : push_mr;
+        {   /* prefetch row r+1's scalars (independent loads) */
+            const int nr = r + 1;
+            cur_pk0 = pk1;
+            cur_pk1 = jb.pre_off[nr + 1];
+            cur_oo0 = oo1;
+            cur_oo1 = jb.out_off[nr + 1];
+            cur_remain = jb.max_remain[nr];
+            cur_ml_mem = jb.max_left[nr];
+            cur_mr_mem = jb.max_right[nr];
+            cur_pidx0 = jb.pre_idx[cur_pk0];
+            cur_ps0 = jb.pre_ps[cur_pk0];
+        }
         int beg, end;
         {
-            int mr = jb.max_remain[r] - end_remain - 1;
+            int mr = row_remain - end_remain - 1;
             if (jb.banded) {
-                int ml = jb.max_left[r], mrr = jb.max_right[r];
-                int lo = ml < qlen - mr ? ml : qlen - mr;
+                int lo = ml_eff < qlen - mr ? ml_eff : qlen - mr;
                 beg = lo - w; if (beg < 0) beg = 0;
-                int hi = mrr > qlen - mr ? mrr : qlen - mr;
+                int hi = mr_eff > qlen - mr ? mr_eff : qlen - mr;
                 end = hi + w; if (end > qlen) end = qlen;
-                int min_pre_beg = 0x7fffffff;
-                for (int k = pk0; k < pk1; ++k) {
-                    const int pidx = jb.pre_idx[k];
-                    int pb = (prev_ok && pidx == prev_row) ? prev_beg : meta[pidx].beg;
-                    if (pb < min_pre_beg) min_pre_beg = pb;
+                int min_pre_beg;
+                if (pk1 - pk0 == 1) {
+                    min_pre_beg = (prev_ok && pidx0 == prev_row) ? prev_beg : meta[pidx0].beg;
+                } else {
+                    min_pre_beg = 0x7fffffff;
+                    for (int k = pk0; k < pk1; ++k) {
+                        const int pidx = jb.pre_idx[k];
+                        int pb = (prev_ok && pidx == prev_row) ? prev_beg : meta[pidx].beg;
+                        if (pb < min_pre_beg) min_pre_beg = pb;
+                    }
                 }
                 if (beg < min_pre_beg) beg = min_pre_beg;
             } else { beg = 0; end = qlen; }
@@ -225,14 +293,33 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         S *cw = &prev_lds[wid][buf_cur][0];          /* written for the next row */
         const S *cr = &prev_lds[wid][buf_cur ^ 1][0]; /* previous row's planes */
 
+        /* the overwhelmingly common row shape (in-degree 1, predecessor =
+         * previous row, cached in LDS): zero global loads in the gather */
+        const bool fast1 = (pk1 - pk0 == 1) && prev_ok && (pidx0 == prev_row);
+
         S carry_h = inf_min, f1c = inf_min, f2c = inf_min;
         S lmax = inf_min; int lleft = -1, lright = -1;
 
+        KPROF_T(kt1);
         for (int cs = beg; cs <= end; cs += WAVE) {
             const int j = cs + lane;
             const bool act = j <= end;
             S h = inf_min, e1v = inf_min, e2v = inf_min;
-            for (int k = pk0; k < pk1; ++k) {
+            if (fast1) {
+                if (act) {
+                    if (local_mode && j == 0) { if (ps0 > h) h = ps0; }
+                    if (j - 1 >= prev_beg && j - 1 <= prev_end) {
+                        S v = (S)(cr[j - 1 - prev_beg] + ps0);
+                        if (v > h) h = v;
+                    }
+                    if (j >= prev_beg && j <= prev_end) {
+                        S v1 = (S)(cr[BMAX + j - prev_beg] + ps0);
+                        S v2 = (S)(cr[2 * BMAX + j - prev_beg] + ps0);
+                        if (v1 > e1v) e1v = v1;
+                        if (v2 > e2v) e2v = v2;
+                    }
+                }
+            } else for (int k = pk0; k < pk1; ++k) {
                 const int p = jb.pre_idx[k];
                 const S ps = (S)jb.pre_ps[k];
                 if (prev_ok && p == prev_row) {
@@ -312,6 +399,7 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
             }
         }
 
+        KPROF_T(kt2);
         if (cache_fits) { prev_ok = 1; prev_row = r; prev_beg = beg; prev_end = end; }
         else prev_ok = 0;
         buf_cur ^= 1;
@@ -337,23 +425,52 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
             } else if (extend_mode) {
                 if (mv > run_best) {
                     run_best = mv; run_best_i = r; run_best_j = rr;
-                    run_best_remain = jb.max_remain[r];
+                    run_best_remain = row_remain;
                 } else if (jb.zdrop > 0) {
-                    int delta = run_best_remain - jb.max_remain[r];
+                    int delta = run_best_remain - row_remain;
                     int dd = delta - (rr - run_best_j); if (dd < 0) dd = -dd;
                     if (run_best - mv > jb.zdrop + jb.e1 * dd) zdropped = 1;
                 }
             }
+            push_ml = 0x7fffffff; push_mr = -0x7fffffff;
             if (!zdropped && jb.banded) {
-                for (int k = jb.out_off[r] + lane; k < jb.out_off[r + 1]; k += WAVE) {
+                /* serial scan (mean out-degree ~1.05): the common edge
+                 * r -> r+1 updates the register push; skip-edges go to
+                 * memory (their readers prefetch strictly later) */
+                for (int k = oo0; k < oo1; ++k) {
                     int o = jb.out_idx[k];
-                    if (rr + 1 > jb.max_right[o]) jb.max_right[o] = rr + 1;
-                    if (ll + 1 < jb.max_left[o]) jb.max_left[o] = ll + 1;
+                    if (o == r + 1) {
+                        if (rr + 1 > push_mr) push_mr = rr + 1;
+                        if (ll + 1 < push_ml) push_ml = ll + 1;
+                    } else if (lane == 0) {
+                        if (rr + 1 > jb.max_right[o]) jb.max_right[o] = rr + 1;
+                        if (ll + 1 < jb.max_left[o]) jb.max_left[o] = ll + 1;
+                    }
                 }
             }
             if (zdropped) break;
+        } else {
+            push_ml = 0x7fffffff; push_mr = -0x7fffffff;
         }
+#ifdef ABAMD_KPROF
+        {
+            unsigned long long kt3 = __builtin_readcyclecounter();
+            kp_band += kt1 - kt0; kp_chunk += kt2 - kt1; kp_epi += kt3 - kt2;
+            kp_rows += 1; kp_chunks += (unsigned long long)((end - beg) / WAVE + 1);
+        }
+#endif
     }
+#ifdef ABAMD_KPROF
+    if (lane == 0) {
+        atomicAdd(&abamd_kprof_acc[0], kp_rows);
+        atomicAdd(&abamd_kprof_acc[1], kp_chunks);
+        atomicAdd(&abamd_kprof_acc[2], kp_band);
+        atomicAdd(&abamd_kprof_acc[3], kp_chunk);
+        atomicAdd(&abamd_kprof_acc[4], kp_epi);
+        atomicAdd(&abamd_kprof_acc[7], 1ull);
+    }
+#endif
+    KPROF_T(kt_bt0);
 
     if (lane == 0) res->cells = used;
 
@@ -528,6 +645,9 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         res->node_e = jb.row_node_id[best_i]; res->query_e = best_j - 1;
         res->node_s = jb.row_node_id[start_i]; res->query_s = start_j - 1;
     }
+#ifdef ABAMD_KPROF
+    atomicAdd(&abamd_kprof_acc[5], __builtin_readcyclecounter() - kt_bt0);
+#endif
 }
 
 extern "C" void abamd_launch_cg_i16(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
@@ -570,7 +690,11 @@ void ag_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
     }
     __syncthreads();
     if (jid >= n_jobs) return;
-    const abamd_gpu_job_t &jb = jobs[jid];
+    /* by VALUE: every field lives in registers (uniform loads become
+       s_loads into SGPRs) — a reference would re-load fields from global
+       memory inside the row loop because arena/steering stores could alias
+       the jobs array in the compiler's view */
+    const abamd_gpu_job_t jb = jobs[jid];
     abamd_gpu_res_t *res = &results[jid];
     abamd_row_meta_t *__restrict__ meta = (abamd_row_meta_t*)jb.row_meta;
 
@@ -937,7 +1061,11 @@ void lg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
     }
     __syncthreads();
     if (jid >= n_jobs) return;
-    const abamd_gpu_job_t &jb = jobs[jid];
+    /* by VALUE: every field lives in registers (uniform loads become
+       s_loads into SGPRs) — a reference would re-load fields from global
+       memory inside the row loop because arena/steering stores could alias
+       the jobs array in the compiler's view */
+    const abamd_gpu_job_t jb = jobs[jid];
     abamd_gpu_res_t *res = &results[jid];
     abamd_row_meta_t *__restrict__ meta = (abamd_row_meta_t*)jb.row_meta;
 
