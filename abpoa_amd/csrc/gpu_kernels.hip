@@ -97,6 +97,35 @@ __device__ __forceinline__ S scan_maxplus(S f, int e, S neutral, int lane) {
     return f;
 }
 
+/* Wave-uniform max/min reductions: 4 DPP row_shr steps collect each 16-lane
+ * row's reduction into its last lane, 4 readlanes + scalar ops finish — the
+ * result is identical to a 6-step __shfl_xor tree but costs ~4 VALU + 4
+ * SALU-ish ops instead of 18 DS-unit permutes (measured ~3.3k cycles per row
+ * in the epilogue before this). bound_ctrl=false keeps the lane's own value
+ * when the DPP source is out of range: the identity for max/min. */
+__device__ __forceinline__ int wave_red_max_i32(int v) {
+    int t;
+    t = __builtin_amdgcn_update_dpp(v, v, 0x111, 0xf, 0xf, false); v = v > t ? v : t;
+    t = __builtin_amdgcn_update_dpp(v, v, 0x112, 0xf, 0xf, false); v = v > t ? v : t;
+    t = __builtin_amdgcn_update_dpp(v, v, 0x114, 0xf, 0xf, false); v = v > t ? v : t;
+    t = __builtin_amdgcn_update_dpp(v, v, 0x118, 0xf, 0xf, false); v = v > t ? v : t;
+    int r0 = __builtin_amdgcn_readlane(v, 15), r1 = __builtin_amdgcn_readlane(v, 31);
+    int r2 = __builtin_amdgcn_readlane(v, 47), r3 = __builtin_amdgcn_readlane(v, 63);
+    r0 = r0 > r1 ? r0 : r1; r2 = r2 > r3 ? r2 : r3;
+    return r0 > r2 ? r0 : r2;
+}
+__device__ __forceinline__ int wave_red_min_i32(int v) {
+    int t;
+    t = __builtin_amdgcn_update_dpp(v, v, 0x111, 0xf, 0xf, false); v = v < t ? v : t;
+    t = __builtin_amdgcn_update_dpp(v, v, 0x112, 0xf, 0xf, false); v = v < t ? v : t;
+    t = __builtin_amdgcn_update_dpp(v, v, 0x114, 0xf, 0xf, false); v = v < t ? v : t;
+    t = __builtin_amdgcn_update_dpp(v, v, 0x118, 0xf, 0xf, false); v = v < t ? v : t;
+    int r0 = __builtin_amdgcn_readlane(v, 15), r1 = __builtin_amdgcn_readlane(v, 31);
+    int r2 = __builtin_amdgcn_readlane(v, 47), r3 = __builtin_amdgcn_readlane(v, 63);
+    r0 = r0 < r1 ? r0 : r1; r2 = r2 < r3 ? r2 : r3;
+    return r0 < r2 ? r0 : r2;
+}
+
 /* device push_cigar, matching abpoa_align.h:54-73 (run-length merge for I/S/H) */
 __device__ static int dev_push_cigar(uint64_t *cig, int *n_c, int cap, int op, int len,
                                      int node_id, int query_id, int *status) {
@@ -237,6 +266,7 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
     int push_ml = 0x7fffffff, push_mr = -0x7fffffff;
 #ifdef ABAMD_KPROF
     unsigned long long kp_band = 0, kp_chunk = 0, kp_epi = 0, kp_rows = 0, kp_chunks = 0;
+    unsigned long long kp_g = 0, kp_s = 0, kp_w = 0;
 #endif
     for (int r = 1; r < n_rows - 1; ++r) {
         KPROF_T(kt0);
@@ -297,10 +327,28 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
          * previous row, cached in LDS): zero global loads in the gather */
         const bool fast1 = (pk1 - pk0 == 1) && prev_ok && (pidx0 == prev_row);
 
+        /* per-row preload of the query scores: the per-chunk global byte
+         * load sat on the chunk critical path (kprof: the gather phase was
+         * ~4k of ~5k cycles per chunk); issuing every chunk's load at row
+         * start overlaps the latency across the whole row */
+        constexpr int QPRE = 8;
+        S qpre[QPRE];
+        {
+            const int n_ch = (int)((end - beg) / WAVE) + 1;
+            #pragma unroll
+            for (int c = 0; c < QPRE; ++c) {
+                const int jj = beg + c * WAVE + lane;
+                qpre[c] = (c < n_ch && jj >= 1 && jj <= end) ? (S)mrow[query[jj - 1]] : (S)0;
+            }
+        }
+
         S carry_h = inf_min, f1c = inf_min, f2c = inf_min;
         S lmax = inf_min; int lleft = -1, lright = -1;
 
         KPROF_T(kt1);
+#ifdef ABAMD_KPROF
+        unsigned long long kc_prev = kt1;
+#endif
         for (int cs = beg; cs <= end; cs += WAVE) {
             const int j = cs + lane;
             const bool act = j <= end;
@@ -357,7 +405,10 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                     }
                 }
             }
-            const S q = (S)((j == 0 || !act) ? 0 : mrow[query[j - 1]]);
+            const int ci = (cs - beg) / WAVE;
+            const S q = ci < QPRE ? qpre[ci]
+                                  : (S)((j == 0 || !act) ? 0 : mrow[query[j - 1]]);
+            KPROF_T(kc1);
             S hpre = (S)(h + q);
             hpre = smax(hpre, smax(e1v, e2v));
             if (!act) hpre = inf_min;
@@ -381,6 +432,7 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
             f1c = (S)__builtin_amdgcn_readlane((int)f1, WAVE - 1);
             f2c = (S)__builtin_amdgcn_readlane((int)f2, WAVE - 1);
 
+            KPROF_T(kc2);
             S hf = smax(hpre, smax(f1, f2));
             if (local_mode) hf = smax(hf, (S)0);
             S e1n = smax((S)(e1v - e1), (S)(hf - oe1));
@@ -397,6 +449,15 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 if (hf > lmax) { lmax = hf; lleft = j; lright = j; }
                 else if (hf == lmax) { lright = j; }
             }
+#ifdef ABAMD_KPROF
+            {
+                unsigned long long kc3 = __builtin_readcyclecounter();
+                kp_g += kc1 - kc_prev; /* gather + query since chunk entry */
+                kp_s += kc2 - kc1;     /* shfl + F scans + carries */
+                kp_w += kc3 - kc2;     /* H/E fold + stores + local track */
+                kc_prev = kc3;
+            }
+#endif
         }
 
         KPROF_T(kt2);
@@ -407,19 +468,11 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         /* row argmax reduce + adaptive band push (simd_abpoa_max_in_row /
          * ada_max_i); local/extend also track the running best here */
         if (jb.banded || local_mode || extend_mode) {
-            int mv = (int)lmax;
-            #pragma unroll
-            for (int sft = 32; sft >= 1; sft >>= 1) {
-                int o = __shfl_xor(mv, sft);
-                if (o > mv) mv = o;
-            }
+            int mv = wave_red_max_i32((int)lmax);
             int ll = ((int)lmax == mv && lleft >= 0) ? lleft : 0x7fffffff;
             int rr = ((int)lmax == mv && lright >= 0) ? lright : -1;
-            #pragma unroll
-            for (int sft = 32; sft >= 1; sft >>= 1) {
-                int lo = __shfl_xor(ll, sft); if (lo < ll) ll = lo;
-                int ro = __shfl_xor(rr, sft); if (ro > rr) rr = ro;
-            }
+            ll = wave_red_min_i32(ll);
+            rr = wave_red_max_i32(rr);
             if (local_mode) {
                 if (mv > run_best) { run_best = mv; run_best_i = r; run_best_j = ll; }
             } else if (extend_mode) {
@@ -468,6 +521,9 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         atomicAdd(&abamd_kprof_acc[3], kp_chunk);
         atomicAdd(&abamd_kprof_acc[4], kp_epi);
         atomicAdd(&abamd_kprof_acc[7], 1ull);
+        atomicAdd(&abamd_kprof_acc[8], kp_g);
+        atomicAdd(&abamd_kprof_acc[9], kp_s);
+        atomicAdd(&abamd_kprof_acc[10], kp_w);
     }
 #endif
     KPROF_T(kt_bt0);
@@ -862,19 +918,11 @@ void ag_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         buf_cur ^= 1;
 
         if (jb.banded || local_mode || extend_mode) {
-            int mv = (int)lmax;
-            #pragma unroll
-            for (int sft = 32; sft >= 1; sft >>= 1) {
-                int o = __shfl_xor(mv, sft);
-                if (o > mv) mv = o;
-            }
+            int mv = wave_red_max_i32((int)lmax);
             int ll = ((int)lmax == mv && lleft >= 0) ? lleft : 0x7fffffff;
             int rr = ((int)lmax == mv && lright >= 0) ? lright : -1;
-            #pragma unroll
-            for (int sft = 32; sft >= 1; sft >>= 1) {
-                int lo = __shfl_xor(ll, sft); if (lo < ll) ll = lo;
-                int ro = __shfl_xor(rr, sft); if (ro > rr) rr = ro;
-            }
+            ll = wave_red_min_i32(ll);
+            rr = wave_red_max_i32(rr);
             if (local_mode) {
                 if (mv > run_best) { run_best = mv; run_best_i = r; run_best_j = ll; }
             } else if (extend_mode) {
@@ -1211,19 +1259,11 @@ void lg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         buf_cur ^= 1;
 
         if (jb.banded || local_mode || extend_mode) {
-            int mv = (int)lmax;
-            #pragma unroll
-            for (int sft = 32; sft >= 1; sft >>= 1) {
-                int o = __shfl_xor(mv, sft);
-                if (o > mv) mv = o;
-            }
+            int mv = wave_red_max_i32((int)lmax);
             int ll = ((int)lmax == mv && lleft >= 0) ? lleft : 0x7fffffff;
             int rr = ((int)lmax == mv && lright >= 0) ? lright : -1;
-            #pragma unroll
-            for (int sft = 32; sft >= 1; sft >>= 1) {
-                int lo = __shfl_xor(ll, sft); if (lo < ll) ll = lo;
-                int ro = __shfl_xor(rr, sft); if (ro > rr) rr = ro;
-            }
+            ll = wave_red_min_i32(ll);
+            rr = wave_red_max_i32(rr);
             if (local_mode) {
                 if (mv > run_best) { run_best = mv; run_best_i = r; run_best_j = ll; }
             } else if (extend_mode) {

@@ -43,6 +43,7 @@ def main():
     f = fetch(L, "abamd_kprof_fold_fetch")
 
     rows, chunks, band, chunk, epi, bt, _, jobs = k[:8]
+    kg, ks, kw = k[8:11]
     tot = band + chunk + epi
     print("== cg DP kernel (per lane-0 job; %d jobs, %d rows, %.1f chunks/row) ==" %
           (jobs, rows, chunks / max(rows, 1)))
@@ -51,6 +52,8 @@ def main():
               (name, v, 100.0 * v / max(tot, 1), v / max(rows, 1)))
     print("  %-14s %12d cyc  (%.0f cyc/chunk in-loop)" %
           ("backtrack", bt, chunk / max(chunks, 1)))
+    for name, v in (("  gather+query", kg), ("  shfl+Fscan", ks), ("  fold+stores", kw)):
+        print("  %-14s %12d cyc  (%.0f cyc/chunk)" % (name, v, v / max(chunks, 1)))
 
     fjobs, fapply, findeg, fbfs, fsort, fremspan, fbuild = f[:7]
     ftot = fapply + findeg + fbfs + fsort + fremspan + fbuild
