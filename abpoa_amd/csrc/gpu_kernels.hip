@@ -706,16 +706,601 @@ void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
 #endif
 }
 
+/* ------------------------------------------------------------------ */
+/* Multi-wave convex kernel: ONE JOB PER 512-THREAD BLOCK.              */
+/*                                                                      */
+/* The one-wave kernel above is instruction-ISSUE bound: a lone wave    */
+/* on a SIMD issues one VALU op per 4 cycles, and a ~350-instruction    */
+/* chunk body makes a typical 340-cell row cost ~21k cycles (kprof).    */
+/* Here 8 waves cover the whole band at once: the M/E gather, H fold,   */
+/* stores and argmax are embarrassingly lane-parallel, and the F        */
+/* (insertion) max-plus chain splits into per-wave local scans plus a   */
+/* cross-wave carry recurrence over the 8 published wave edges — the    */
+/* regrouped decays are bit-identical because wrapping addition is      */
+/* associative and the inf_min headroom (+512*ext_max, the same bound   */
+/* the one-wave scan relies on) keeps every comparison un-wrapped.      */
+/* Three block barriers per row replace ~5 serial chunk iterations.     */
+/* ------------------------------------------------------------------ */
+#define MWAVES 8
+#define MWT (WAVE * MWAVES)
+#define BMW 512 /* prev-row LDS cache width (cells); wider bands re-read the arena */
+
+template <typename S>
+__global__ __launch_bounds__(MWT, 1)
+void cg_global_mw_kernel(const abamd_gpu_job_t *__restrict__ jobs,
+                         abamd_gpu_res_t *__restrict__ results, int n_jobs) {
+    const int jid = blockIdx.x;
+    if (jid >= n_jobs) return;
+    const int tid = threadIdx.x;
+    const int wv = tid / WAVE, lane = tid % WAVE;
+
+    __shared__ int mat_lds[27 * 27];
+    __shared__ S prev_lds[2][3 * BMW];
+    __shared__ S sc_h[MWAVES], sc_f1[MWAVES], sc_f2[MWAVES];
+    __shared__ int sc_red[3 * MWAVES];
+    __shared__ S sc_carry[3];
+
+    const abamd_gpu_job_t jb = jobs[jid];
+    abamd_gpu_res_t *res = &results[jid];
+    abamd_row_meta_t *__restrict__ meta = (abamd_row_meta_t*)jb.row_meta;
+
+    const int qlen = jb.qlen, n_rows = jb.n_rows, w = jb.w, m = jb.m;
+    const S inf_min = (S)jb.inf_min;
+    const S e1 = (S)jb.e1, e2 = (S)jb.e2;
+    const S oe1 = (S)jb.oe1, oe2 = (S)jb.oe2;
+    const int local_mode = jb.align_mode == 1, extend_mode = jb.align_mode == 2;
+    int32_t run_best = jb.inf_min;
+    int run_best_i = 0, run_best_j = 0, run_best_remain = jb.max_remain[0];
+    int zdropped = 0;
+    const int end_remain = jb.max_remain[n_rows - 1];
+    S *arena = (S*)jb.arena;
+    const uint8_t *__restrict__ query = jb.query;
+
+    for (int i = tid; i < m * m; i += MWT) mat_lds[i] = jb.mat[i];
+    if (tid == 0) { res->status = ABAMD_JOB_OK; res->n_cigar = 0; }
+
+    /* adaptive band state init (abpoa_topological_sort:347-353 + first_dp) */
+    for (int i = tid; i < n_rows; i += MWT) {
+        jb.max_left[i] = jb.node_n_init;
+        jb.max_right[i] = 0;
+    }
+    if (tid == 0) { jb.max_left[0] = 0; jb.max_right[0] = 0; }
+    for (int k = jb.out_off[0] + tid; k < jb.out_off[1]; k += MWT) {
+        int o = jb.out_idx[k];
+        jb.max_left[o] = 1; jb.max_right[o] = 1;
+    }
+    __syncthreads();
+
+    int buf_cur = 0;
+    int prev_ok = 0, prev_row = -1, prev_beg = 0, prev_end = -1;
+
+    /* ---- first row (simd_abpoa_cg_first_dp) ---- */
+    int64_t used;
+    {
+        int mr = jb.max_remain[0] - end_remain - 1;
+        int end0;
+        if (jb.banded) {
+            int t = jb.max_right[0] > qlen - mr ? jb.max_right[0] : qlen - mr;
+            end0 = (qlen < t + w) ? qlen : t + w;
+        } else end0 = qlen;
+        if (tid == 0) { meta[0].beg = 0; meta[0].end = end0; meta[0].off = 0; }
+        int64_t bw = end0 + 1;
+        used = bw;
+        S *H = arena, *E1 = arena + bw, *E2 = arena + 2 * bw, *F1 = arena + 3 * bw, *F2 = arena + 4 * bw;
+        S *c = &prev_lds[buf_cur][0];
+        const int fits = end0 + 1 <= BMW;
+        for (int j = tid; j <= end0; j += MWT) {
+            S hv, e1v2, e2v2;
+            if (local_mode) {
+                hv = 0; e1v2 = 0; e2v2 = 0;
+                F1[j] = 0; F2[j] = 0;
+            } else if (j == 0) {
+                hv = 0; e1v2 = (S)(0 - oe1); e2v2 = (S)(0 - oe2);
+                F1[0] = inf_min; F2[0] = inf_min;
+            } else {
+                S f1 = (S)(-(jb.o1 + jb.e1 * j));
+                S f2 = (S)(-(jb.o2 + jb.e2 * j));
+                F1[j] = f1; F2[j] = f2;
+                hv = smax(f1, f2);
+                e1v2 = inf_min; e2v2 = inf_min;
+            }
+            H[j] = hv; E1[j] = e1v2; E2[j] = e2v2;
+            if (fits) {
+                c[j] = hv; c[BMW + j] = e1v2; c[2 * BMW + j] = e2v2;
+            }
+        }
+        if (fits) { prev_ok = 1; prev_row = 0; prev_beg = 0; prev_end = end0; }
+        buf_cur ^= 1;
+        __syncthreads();
+    }
+
+    /* ---- main row loop (row scalars rolled one row ahead; the common
+     * band push r -> r+1 carried in registers — every thread keeps the
+     * same copy, and skip-edge pushes are idempotent same-value writes
+     * done by every thread so each thread's later prefetch of the pushed
+     * row is self-ordered) ---- */
+    int cur_pk0 = jb.pre_off[1];
+    int cur_pk1 = jb.pre_off[2 <= n_rows ? 2 : n_rows];
+    int cur_oo0 = jb.out_off[1];
+    int cur_oo1 = jb.out_off[2 <= n_rows ? 2 : n_rows];
+    int cur_remain = jb.max_remain[1];
+    int cur_ml_mem = jb.max_left[1];
+    int cur_mr_mem = jb.max_right[1];
+    int cur_pidx0 = jb.pre_idx[cur_pk0];
+    int cur_ps0 = jb.pre_ps[cur_pk0];
+    int push_ml = 0x7fffffff, push_mr = -0x7fffffff;
+#ifdef ABAMD_KPROF
+    unsigned long long mw_a = 0, mw_b = 0, mw_c = 0, mw_d = 0, mw_rows = 0;
+#endif
+    for (int r = 1; r < n_rows - 1; ++r) {
+        KPROF_T(mt0);
+        const int pk0 = cur_pk0, pk1 = cur_pk1;
+        const int oo0 = cur_oo0, oo1 = cur_oo1;
+        const int row_remain = cur_remain;
+        const int pidx0 = cur_pidx0;
+        const S ps0 = (S)cur_ps0;
+        const int ml_eff = cur_ml_mem < push_ml ? cur_ml_mem : push_ml;
+        const int mr_eff = cur_mr_mem > push_mr ? cur_mr_mem : push_mr;
+        {
+            const int nr = r + 1;
+            cur_pk0 = pk1;
+            cur_pk1 = jb.pre_off[nr + 1];
+            cur_oo0 = oo1;
+            cur_oo1 = jb.out_off[nr + 1];
+            cur_remain = jb.max_remain[nr];
+            cur_ml_mem = jb.max_left[nr];
+            cur_mr_mem = jb.max_right[nr];
+            cur_pidx0 = jb.pre_idx[cur_pk0];
+            cur_ps0 = jb.pre_ps[cur_pk0];
+        }
+        int beg, end;
+        {
+            int mr = row_remain - end_remain - 1;
+            if (jb.banded) {
+                int lo = ml_eff < qlen - mr ? ml_eff : qlen - mr;
+                beg = lo - w; if (beg < 0) beg = 0;
+                int hi = mr_eff > qlen - mr ? mr_eff : qlen - mr;
+                end = hi + w; if (end > qlen) end = qlen;
+                int min_pre_beg;
+                if (pk1 - pk0 == 1) {
+                    min_pre_beg = (prev_ok && pidx0 == prev_row) ? prev_beg : meta[pidx0].beg;
+                } else {
+                    min_pre_beg = 0x7fffffff;
+                    for (int k = pk0; k < pk1; ++k) {
+                        const int pidx = jb.pre_idx[k];
+                        int pb = (prev_ok && pidx == prev_row) ? prev_beg : meta[pidx].beg;
+                        if (pb < min_pre_beg) min_pre_beg = pb;
+                    }
+                }
+                if (beg < min_pre_beg) beg = min_pre_beg;
+            } else { beg = 0; end = qlen; }
+        }
+        const int64_t bw = end - beg + 1;
+        if (used + bw > jb.arena_cap) { if (tid == 0) res->status = ABAMD_JOB_ARENA_OVERFLOW; return; }
+        const int64_t off = used;
+        if (tid == 0) { meta[r].beg = beg; meta[r].end = end; meta[r].off = off; }
+        used += bw;
+        S *H = arena + off * 5, *E1r = H + bw, *E2r = E1r + bw, *F1r = E2r + bw, *F2r = F1r + bw;
+        const uint8_t base = jb.row_base[r];
+        const int *mrow = &mat_lds[base * m];
+        const int cache_fits = bw <= BMW;
+        S *cw = &prev_lds[buf_cur][0];
+        const S *cr = &prev_lds[buf_cur ^ 1][0];
+        const bool fast1 = (pk1 - pk0 == 1) && prev_ok && (pidx0 == prev_row);
+
+        S carry_h = inf_min, f1c = inf_min, f2c = inf_min; /* superchunk carries */
+        S lmax = inf_min; int lleft = -1, lright = -1;
+
+        for (int ss = beg; ss <= end; ss += MWT) {
+            const int j = ss + tid;
+            const bool act = j <= end;
+            S h = inf_min, e1v = inf_min, e2v = inf_min;
+            if (fast1) {
+                if (act) {
+                    if (local_mode && j == 0) { if (ps0 > h) h = ps0; }
+                    if (j - 1 >= prev_beg && j - 1 <= prev_end) {
+                        S v = (S)(cr[j - 1 - prev_beg] + ps0);
+                        if (v > h) h = v;
+                    }
+                    if (j >= prev_beg && j <= prev_end) {
+                        S v1 = (S)(cr[BMW + j - prev_beg] + ps0);
+                        S v2 = (S)(cr[2 * BMW + j - prev_beg] + ps0);
+                        if (v1 > e1v) e1v = v1;
+                        if (v2 > e2v) e2v = v2;
+                    }
+                }
+            } else for (int k = pk0; k < pk1; ++k) {
+                const int p = jb.pre_idx[k];
+                const S ps = (S)jb.pre_ps[k];
+                if (prev_ok && p == prev_row) {
+                    if (act) {
+                        if (local_mode && j == 0) { if (ps > h) h = ps; }
+                        if (j - 1 >= prev_beg && j - 1 <= prev_end) {
+                            S v = (S)(cr[j - 1 - prev_beg] + ps);
+                            if (v > h) h = v;
+                        }
+                        if (j >= prev_beg && j <= prev_end) {
+                            S v1 = (S)(cr[BMW + j - prev_beg] + ps);
+                            S v2 = (S)(cr[2 * BMW + j - prev_beg] + ps);
+                            if (v1 > e1v) e1v = v1;
+                            if (v2 > e2v) e2v = v2;
+                        }
+                    }
+                    continue;
+                }
+                const abamd_row_meta_t pm = meta[p];
+                const int64_t pbw = pm.end - pm.beg + 1;
+                const S *__restrict__ pH = arena + pm.off * 5;
+                const S *__restrict__ pE1 = pH + pbw;
+                const S *__restrict__ pE2 = pE1 + pbw;
+                if (act) {
+                    if (local_mode && j == 0) { if (ps > h) h = ps; }
+                    if (j - 1 >= pm.beg && j - 1 <= pm.end) {
+                        S v = (S)(pH[j - 1 - pm.beg] + ps);
+                        if (v > h) h = v;
+                    }
+                    if (j >= pm.beg && j <= pm.end) {
+                        S v1 = (S)(pE1[j - pm.beg] + ps);
+                        S v2 = (S)(pE2[j - pm.beg] + ps);
+                        if (v1 > e1v) e1v = v1;
+                        if (v2 > e2v) e2v = v2;
+                    }
+                }
+            }
+            const S q = (S)((j == 0 || !act) ? 0 : mrow[query[j - 1]]);
+            S hpre = (S)(h + q);
+            hpre = smax(hpre, smax(e1v, e2v));
+            if (!act) hpre = inf_min;
+
+            /* wave edges of hpre, then per-wave local F scans */
+            KPROF_T(mt1);
+            if (lane == WAVE - 1) sc_h[wv] = hpre;
+            __syncthreads();
+            S hshift = (S)__shfl_up((int)hpre, 1);
+            S c1, c2;
+            if (lane == 0) {
+                if (wv == 0) {
+                    if (ss == beg) { c1 = (S)(inf_min - oe1); c2 = (S)(inf_min - oe2); }
+                    else {
+                        c1 = smax((S)(carry_h - oe1), (S)(f1c - e1));
+                        c2 = smax((S)(carry_h - oe2), (S)(f2c - e2));
+                    }
+                } else {
+                    S hp = sc_h[wv - 1];
+                    c1 = (S)(hp - oe1);
+                    c2 = (S)(hp - oe2);
+                }
+            } else {
+                c1 = (S)(hshift - oe1);
+                c2 = (S)(hshift - oe2);
+            }
+            S L1 = scan_maxplus(c1, jb.e1, inf_min, lane);
+            S L2 = scan_maxplus(c2, jb.e2, inf_min, lane);
+            if (lane == WAVE - 1) { sc_f1[wv] = L1; sc_f2[wv] = L2; }
+            __syncthreads();
+            /* cross-wave carry: C = true F at the last lane of wave wv-1
+             * (max-plus with linear decay is associative; wrapping adds
+             * regroup exactly) */
+            S f1 = L1, f2 = L2;
+            if (wv > 0) {
+                S C1 = sc_f1[0], C2 = sc_f2[0];
+                for (int ww = 1; ww < wv; ++ww) {
+                    C1 = smax(sc_f1[ww], (S)(C1 - (S)(WAVE * jb.e1)));
+                    C2 = smax(sc_f2[ww], (S)(C2 - (S)(WAVE * jb.e2)));
+                }
+                f1 = smax(L1, (S)(C1 - (S)((lane + 1) * jb.e1)));
+                f2 = smax(L2, (S)(C2 - (S)((lane + 1) * jb.e2)));
+            }
+
+            S hf = smax(hpre, smax(f1, f2));
+            if (local_mode) hf = smax(hf, (S)0);
+            S e1n = smax((S)(e1v - e1), (S)(hf - oe1));
+            S e2n = smax((S)(e2v - e2), (S)(hf - oe2));
+            if (local_mode) { e1n = smax(e1n, (S)0); e2n = smax(e2n, (S)0); }
+            if (act) {
+                H[j - beg] = hf; E1r[j - beg] = e1n; E2r[j - beg] = e2n;
+                F1r[j - beg] = f1; F2r[j - beg] = f2;
+                if (cache_fits) {
+                    cw[j - beg] = hf;
+                    cw[BMW + j - beg] = e1n;
+                    cw[2 * BMW + j - beg] = e2n;
+                }
+                if (hf > lmax) { lmax = hf; lleft = j; lright = j; }
+                else if (hf == lmax) { lright = j; }
+            }
+            KPROF_T(mt2);
+            const bool more = ss + MWT <= end;
+            if (more && j == ss + MWT - 1) {
+                sc_carry[0] = hpre; sc_carry[1] = f1; sc_carry[2] = f2;
+            }
+            __syncthreads(); /* also orders cw writes before the next row's reads */
+            if (more) { carry_h = sc_carry[0]; f1c = sc_carry[1]; f2c = sc_carry[2]; }
+#ifdef ABAMD_KPROF
+            {
+                unsigned long long mt3 = __builtin_readcyclecounter();
+                mw_a += mt1 - mt0;   /* band + prefetch + gather + hpre */
+                mw_b += mt2 - mt1;   /* B1 + scans + B2 + carries + fold + stores */
+                mw_c += mt3 - mt2;   /* carry publish + B3 */
+            }
+#endif
+        }
+
+        if (cache_fits) { prev_ok = 1; prev_row = r; prev_beg = beg; prev_end = end; }
+        else prev_ok = 0;
+        buf_cur ^= 1;
+
+        if (jb.banded || local_mode || extend_mode) {
+            /* per-wave argmax, then a cross-wave combine every thread
+             * repeats from the 8 published triples (uniform result) */
+            int mvw = wave_red_max_i32((int)lmax);
+            int llw = ((int)lmax == mvw && lleft >= 0) ? lleft : 0x7fffffff;
+            int rrw = ((int)lmax == mvw && lright >= 0) ? lright : -1;
+            llw = wave_red_min_i32(llw);
+            rrw = wave_red_max_i32(rrw);
+            if (lane == 0) {
+                sc_red[wv] = mvw;
+                sc_red[MWAVES + wv] = llw;
+                sc_red[2 * MWAVES + wv] = rrw;
+            }
+            __syncthreads();
+            int mv = sc_red[0], ll = sc_red[MWAVES], rr = sc_red[2 * MWAVES];
+            #pragma unroll
+            for (int ww = 1; ww < MWAVES; ++ww) {
+                int m2 = sc_red[ww];
+                if (m2 > mv) { mv = m2; ll = sc_red[MWAVES + ww]; rr = sc_red[2 * MWAVES + ww]; }
+                else if (m2 == mv) {
+                    if (sc_red[MWAVES + ww] < ll) ll = sc_red[MWAVES + ww];
+                    if (sc_red[2 * MWAVES + ww] > rr) rr = sc_red[2 * MWAVES + ww];
+                }
+            }
+            if (local_mode) {
+                if (mv > run_best) { run_best = mv; run_best_i = r; run_best_j = ll; }
+            } else if (extend_mode) {
+                if (mv > run_best) {
+                    run_best = mv; run_best_i = r; run_best_j = rr;
+                    run_best_remain = row_remain;
+                } else if (jb.zdrop > 0) {
+                    int delta = run_best_remain - row_remain;
+                    int dd = delta - (rr - run_best_j); if (dd < 0) dd = -dd;
+                    if (run_best - mv > jb.zdrop + jb.e1 * dd) zdropped = 1;
+                }
+            }
+            push_ml = 0x7fffffff; push_mr = -0x7fffffff;
+            if (!zdropped && jb.banded) {
+                for (int k = oo0; k < oo1; ++k) {
+                    int o = jb.out_idx[k];
+                    if (o == r + 1) {
+                        if (rr + 1 > push_mr) push_mr = rr + 1;
+                        if (ll + 1 < push_ml) push_ml = ll + 1;
+                    } else {
+                        /* every thread performs the same idempotent update so
+                         * its own later prefetch of row o is self-ordered */
+                        if (rr + 1 > jb.max_right[o]) jb.max_right[o] = rr + 1;
+                        if (ll + 1 < jb.max_left[o]) jb.max_left[o] = ll + 1;
+                    }
+                }
+            }
+            if (zdropped) break;
+        } else {
+            push_ml = 0x7fffffff; push_mr = -0x7fffffff;
+        }
+#ifdef ABAMD_KPROF
+        {
+            unsigned long long mt4 = __builtin_readcyclecounter();
+            mw_d += mt4 - mt0; mw_rows += 1; /* mt0 re-read below start */
+        }
+#endif
+    }
+#ifdef ABAMD_KPROF
+    if (tid == 0) {
+        atomicAdd(&abamd_kprof_acc[11], mw_a);
+        atomicAdd(&abamd_kprof_acc[12], mw_b);
+        atomicAdd(&abamd_kprof_acc[13], mw_c);
+        atomicAdd(&abamd_kprof_acc[14], mw_d);
+        atomicAdd(&abamd_kprof_acc[15], mw_rows);
+    }
+#endif
+
+    __syncthreads();
+    if (tid == 0) res->cells = used;
+
+    /* ---- final best + backtrack (thread 0; simd_abpoa_cg_backtrack) ---- */
+    if (tid != 0) return;
+
+    int32_t best_score = run_best;
+    int best_i = run_best_i, best_j = run_best_j;
+    if (jb.align_mode == 0) {
+        best_score = jb.inf_min; best_i = 0; best_j = 0;
+        for (int k = jb.pre_off[n_rows - 1]; k < jb.pre_off[n_rows]; ++k) {
+            const int p = jb.pre_idx[k];
+            const abamd_row_meta_t pm = meta[p];
+            int e = pm.end < qlen ? pm.end : qlen;
+            const S *pH = arena + pm.off * 5;
+            int32_t sc = (e >= pm.beg) ? (int32_t)pH[e - pm.beg] : jb.inf_min;
+            if (sc > best_score) { best_score = sc; best_i = p; best_j = e; }
+        }
+    }
+    res->best_score = best_score;
+    res->best_i = best_i; res->best_j = best_j;
+    if (!jb.ret_cigar) return;
+    {
+        int bi = best_i, bj = best_j, start_i = best_i, start_j = best_j;
+        int cur_op = 0x1f, n_c = 0, status = ABAMD_JOB_OK;
+        int look_end = jb.put_gap_at_end, put_right = jb.put_gap_on_right;
+        int n_aln = 0, n_matched = 0;
+        uint64_t *cig = jb.cigar;
+        int id = jb.row_node_id[bi];
+        if (best_j < qlen) dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, qlen - best_j, -1, qlen - 1, &status);
+        while (bi > 0 && bj > 0 && status == ABAMD_JOB_OK) {
+            const abamd_row_meta_t bm = meta[bi];
+            const int rb = bm.beg, re = bm.end;
+            const int64_t bw = re - rb + 1;
+            const S *H = arena + bm.off * 5;
+            const S *E1r = H + bw, *E2r = E1r + bw, *F1r = E2r + bw, *F2r = F1r + bw;
+            const S Hj = (bj >= rb && bj <= re) ? H[bj - rb] : inf_min;
+            const S Hjm1 = (bj - 1 >= rb && bj - 1 <= re) ? H[bj - 1 - rb] : inf_min;
+            const S E1j = (bj >= rb && bj <= re) ? E1r[bj - rb] : inf_min;
+            const S E2j = (bj >= rb && bj <= re) ? E2r[bj - rb] : inf_min;
+            const S F1j = (bj >= rb && bj <= re) ? F1r[bj - rb] : inf_min;
+            const S F2j = (bj >= rb && bj <= re) ? F2r[bj - rb] : inf_min;
+            const S F1jm1 = (bj - 1 >= rb && bj - 1 <= re) ? F1r[bj - 1 - rb] : inf_min;
+            const S F2jm1 = (bj - 1 >= rb && bj - 1 <= re) ? F2r[bj - 1 - rb] : inf_min;
+            if (local_mode && Hj == 0) break;
+            start_i = bi; start_j = bj;
+            const int pq0 = jb.pre_off[bi], pq1 = jb.pre_off[bi + 1];
+            const S s = (S)mat_lds[m * jb.row_base[bi] + query[bj - 1]];
+            const int is_match = jb.row_base[bi] == query[bj - 1];
+            int hit = 0;
+            if (put_right == 0 && look_end == 0 && (cur_op & 0x1)) {
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
+                    const S *pH = arena + pm.off * 5;
+                    if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
+                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
+                        bi = p; --bj; id = jb.row_node_id[bi]; hit = 1;
+                        cur_op = 0x1f; ++n_aln; n_matched += is_match;
+                        break;
+                    }
+                }
+            }
+            if (!hit && (cur_op & 0x6)) { /* deletion */
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj < pm.beg || bj > pm.end) continue;
+                    const int poffc = bj - pm.beg;
+                    const int64_t pbw = pm.end - pm.beg + 1;
+                    const S *pH = arena + pm.off * 5;
+                    const S *pE1 = pH + pbw, *pE2 = pE1 + pbw;
+                    if (cur_op & 0x2) {
+                        if (cur_op & 0x1) {
+                            if (Hj == (S)(pE1[poffc] + ps)) {
+                                cur_op = ((S)(pH[poffc] - oe1) == pE1[poffc]) ? (0x1 | 0x18) : 0x2;
+                                hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                                bi = p; id = jb.row_node_id[bi];
+                                if (look_end) look_end = 0;
+                                break;
+                            }
+                        } else {
+                            if (E1j == (S)(pE1[poffc] - e1 + ps)) {
+                                cur_op = ((S)(pH[poffc] - oe1) == pE1[poffc]) ? (0x1 | 0x18) : 0x2;
+                                hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                                bi = p; id = jb.row_node_id[bi];
+                                if (look_end) look_end = 0;
+                                break;
+                            }
+                        }
+                    }
+                    if (cur_op & 0x4) {
+                        if (cur_op & 0x1) {
+                            if (Hj == (S)(pE2[poffc] + ps)) {
+                                cur_op = ((S)(pH[poffc] - oe2) == pE2[poffc]) ? (0x1 | 0x18) : 0x4;
+                                hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                                bi = p; id = jb.row_node_id[bi];
+                                if (look_end) look_end = 0;
+                                break;
+                            }
+                        } else {
+                            if (E2j == (S)(pE2[poffc] - e2 + ps)) {
+                                cur_op = ((S)(pH[poffc] - oe2) == pE2[poffc]) ? (0x1 | 0x18) : 0x4;
+                                hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                                bi = p; id = jb.row_node_id[bi];
+                                if (look_end) look_end = 0;
+                                break;
+                            }
+                        }
+                    }
+                }
+            }
+            if (!hit && (cur_op & 0x18)) { /* insertion */
+                if (cur_op & 0x8) {
+                    if (cur_op & 0x1) {
+                        if (Hj == F1j) {
+                            if ((S)(Hjm1 - oe1) == F1j) { cur_op = 0x1 | 0x6; hit = 1; }
+                            else if ((S)(F1jm1 - e1) == F1j) { cur_op = 0x8; hit = 1; }
+                        }
+                    } else {
+                        if ((S)(Hjm1 - oe1) == F1j) { cur_op = 0x1 | 0x6; hit = 1; }
+                        else if ((S)(F1jm1 - e1) == F1j) { cur_op = 0x8; hit = 1; }
+                    }
+                }
+                if (!hit && (cur_op & 0x10)) {
+                    if (cur_op & 0x1) {
+                        if (Hj == F2j) {
+                            if ((S)(Hjm1 - oe2) == F2j) { cur_op = 0x1 | 0x6; hit = 1; }
+                            else if ((S)(F2jm1 - e2) == F2j) { cur_op = 0x10; hit = 1; }
+                        }
+                    } else {
+                        if ((S)(Hjm1 - oe2) == F2j) { cur_op = 0x1 | 0x6; hit = 1; }
+                        else if ((S)(F2jm1 - e2) == F2j) { cur_op = 0x10; hit = 1; }
+                    }
+                }
+                if (hit) {
+                    dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, 1, id, bj - 1, &status);
+                    --bj;
+                    if (look_end) look_end = 0;
+                    ++n_aln;
+                }
+            }
+            if (!hit && (cur_op & 0x1)) {
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
+                    const S *pH = arena + pm.off * 5;
+                    if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
+                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
+                        bi = p; --bj; id = jb.row_node_id[bi]; hit = 1;
+                        cur_op = 0x1f; ++n_aln; n_matched += is_match;
+                        look_end = 0;
+                        break;
+                    }
+                }
+            }
+            if (!hit) { status = ABAMD_JOB_BT_DEAD_END; break; }
+        }
+        if (status == ABAMD_JOB_OK && bj > 0)
+            dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, bj, -1, bj - 1, &status);
+        res->status = status;
+        res->n_cigar = n_c;
+        res->n_aln_bases = n_aln;
+        res->n_matched_bases = n_matched;
+        res->node_e = jb.row_node_id[best_i]; res->query_e = best_j - 1;
+        res->node_s = jb.row_node_id[start_i]; res->query_s = start_j - 1;
+    }
+}
+
+static int use_single_wave_kernel(void) {
+    static int v = -1;
+    if (v < 0) v = getenv("ABPOA_AMD_SW_KERNEL") != nullptr;
+    return v;
+}
+
 extern "C" void abamd_launch_cg_i16(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
                                     int n_jobs, void *stream) {
-    int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
-    hipLaunchKernelGGL((cg_global_kernel<int16_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
+    if (use_single_wave_kernel()) {
+        int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
+        hipLaunchKernelGGL((cg_global_kernel<int16_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
+                           (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
+        return;
+    }
+    hipLaunchKernelGGL((cg_global_mw_kernel<int16_t>), dim3(n_jobs), dim3(MWT), 0,
                        (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
 }
 extern "C" void abamd_launch_cg_i32(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
                                     int n_jobs, void *stream) {
-    int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
-    hipLaunchKernelGGL((cg_global_kernel<int32_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
+    if (use_single_wave_kernel()) {
+        int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
+        hipLaunchKernelGGL((cg_global_kernel<int32_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
+                           (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
+        return;
+    }
+    hipLaunchKernelGGL((cg_global_mw_kernel<int32_t>), dim3(n_jobs), dim3(MWT), 0,
                        (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
 }
 

@@ -54,6 +54,13 @@ def main():
           ("backtrack", bt, chunk / max(chunks, 1)))
     for name, v in (("  gather+query", kg), ("  shfl+Fscan", ks), ("  fold+stores", kw)):
         print("  %-14s %12d cyc  (%.0f cyc/chunk)" % (name, v, v / max(chunks, 1)))
+    mwa, mwb, mwc, mwtot, mwrows = k[11], k[12], k[13], k[14], k[15]
+    if mwrows:
+        print("== cg MULTI-WAVE kernel (%d rows, %.0f cyc/row total) ==" %
+              (mwrows, mwtot / mwrows))
+        for name, v in (("gather+hpre", mwa), ("B1+scan+B2+fold", mwb),
+                        ("carry+B3", mwc), ("epilogue", mwtot - mwa - mwb - mwc)):
+            print("  %-16s %12d cyc  (%.0f cyc/row)" % (name, v, v / mwrows))
 
     fjobs, fapply, findeg, fbfs, fsort, fremspan, fbuild = f[:7]
     ftot = fapply + findeg + fbfs + fsort + fremspan + fbuild
