@@ -51,10 +51,13 @@ void abamd_flat_apply_alignment(flat_graph_t *fg, int beg_node_id, int end_node_
                                 int read_id, int add_read_id, int inc_both_ends);
 
 #ifdef ABPOA_AMD_H
-/* Rebuild a pointer graph from a (host copy of a) flat graph and
- * topo-sort it — the device-resident batch driver's hand-off to the host
- * consensus path (abamd_graph.c). `ab` must be fresh. */
-void abamd_graph_from_flat(abpoa_t *ab, const flat_graph_t *fg, abpoa_para_t *abpt, int read_ids_n);
+/* Rebuild a pointer graph from a (host copy of a) flat graph and topo-sort
+ * it — the device-resident batch driver's hand-off to the host consensus
+ * path (abamd_graph.c). `ab` must be fresh. All node arrays are carved from
+ * ONE returned slab; call abamd_graph_arena_release(ab, slab) before
+ * abpoa_free(ab). */
+void *abamd_graph_from_flat(abpoa_t *ab, const flat_graph_t *fg, abpoa_para_t *abpt, int read_ids_n);
+void abamd_graph_arena_release(abpoa_t *ab, void *slab);
 #endif
 
 #ifdef __cplusplus

@@ -660,86 +660,112 @@ void abpoa_subgraph_nodes(abpoa_t *ab, abpoa_para_t *abpt, int inc_beg, int inc_
  * (abamd_fold_core.h) after the GPU has folded every read of a set. The flat
  * fold is CPU-twin-proven bit-equal to the pointer fold (tests/test_fold_twin,
  * device-validated by the gpu fold test), so walking its chains in order
- * reproduces the exact adjacency order — including the weight-sort order the
- * device applied — and the final abpoa_topological_sort here re-derives the
+ * reproduces the exact adjacency order - including the weight-sort order the
+ * device applied - and the final abpoa_topological_sort here re-derives the
  * index arrays with the identical algorithm, leaving the graph byte-equal to
- * what the host fold path would have produced. `ab` must be fresh
- * (abpoa_init, untouched). read_ids_n > 0 copies per-out-edge read bitsets
- * (fg->rid_n must equal read_ids_n). */
+ * what the host fold path would have produced.
+ *
+ * ALL node-level arrays are carved from ONE malloc per set (returned to the
+ * caller): with ~50k nodes x 4 arrays x 1000 sets, per-array malloc would be
+ * ~200M allocator calls and serializes a 200-thread consensus pool on the
+ * allocator lock (measured: 12 s of a 34 s step). The caller must NOT
+ * abpoa_free() the graph while the slab is live - it releases the slab with
+ * abamd_graph_arena_release(), which restores a normally-allocated empty
+ * graph first. `ab` must be fresh (abpoa_init, untouched). read_ids_n > 0
+ * copies per-out-edge read bitsets (fg->rid_n must equal read_ids_n). */
 #include "abamd_fold_core.h"
 
-void abamd_graph_from_flat(abpoa_t *ab, const flat_graph_t *fg, abpoa_para_t *abpt, int read_ids_n) {
+void *abamd_graph_from_flat(abpoa_t *ab, const flat_graph_t *fg, abpoa_para_t *abpt, int read_ids_n) {
     abpoa_graph_t *g = ab->abg;
-    int id, e, n;
-    if (fg->node_n > g->node_m) {
-        int node_m = ab_round_up_pow2_32(fg->node_n);
-        g->node = (abpoa_node_t*)abamd_realloc(g->node, (size_t)node_m * sizeof(abpoa_node_t));
-        for (id = g->node_m; id < node_m; ++id) init_node(&g->node[id], id);
-        g->node_m = node_m;
-    }
+    int id, e;
     if (read_ids_n > 0 && fg->rid_n != read_ids_n)
         abamd_fatal("abamd_graph_from_flat", "rid_n %d != read_ids_n %d", fg->rid_n, read_ids_n);
+    const size_t nn = (size_t)fg->node_n;
+    const size_t n_in = (size_t)fg->edge_n_in, n_out = (size_t)fg->edge_n_out;
+    const size_t n_aln = (size_t)fg->aln_n;
+    size_t sz_nodes = nn * sizeof(abpoa_node_t);
+    size_t sz_in = 2 * n_in * sizeof(int);
+    size_t sz_out = 2 * n_out * sizeof(int);
+    size_t sz_aln = n_aln * sizeof(int);
+    size_t sz_ridp = read_ids_n > 0 ? n_out * sizeof(uint64_t*) : 0;
+    size_t sz_rid = read_ids_n > 0 ? n_out * (size_t)read_ids_n * sizeof(uint64_t) : 0;
+    size_t a = 0;
+    #define AL16(x) (((x) + 15) & ~(size_t)15)
+    size_t o_nodes = 0;                 a = AL16(sz_nodes);
+    size_t o_in = a;                    a += AL16(sz_in);
+    size_t o_out = a;                   a += AL16(sz_out);
+    size_t o_aln = a;                   a += AL16(sz_aln);
+    size_t o_ridp = a;                  a += AL16(sz_ridp);
+    size_t o_rid = a;                   a += AL16(sz_rid);
+    uint8_t *slab = (uint8_t*)abamd_malloc(a ? a : 16);
+    abpoa_node_t *nodes = (abpoa_node_t*)(slab + o_nodes);
+    int *in_pool = (int*)(slab + o_in);
+    int *out_pool = (int*)(slab + o_out);
+    int *aln_pool = (int*)(slab + o_aln);
+    uint64_t **ridp_pool = (uint64_t**)(slab + o_ridp);
+    uint64_t *rid_pool = (uint64_t*)(slab + o_rid);
+    size_t c_in = 0, c_out = 0, c_aln = 0;
     for (id = 0; id < fg->node_n; ++id) {
-        abpoa_node_t *v = &g->node[id];
+        abpoa_node_t *v = &nodes[id];
+        memset(v, 0, sizeof(*v));
         v->node_id = id;
         v->base = fg->base[id];
         v->n_read = fg->n_read[id];
         v->n_span_read = fg->n_span_read[id];
         /* in edges, chain order */
-        n = 0;
-        for (e = fg->in_head[id]; e != -1; e = fg->in_next[e]) ++n;
-        if (n > v->in_edge_m) {
-            v->in_id = (int*)abamd_realloc(v->in_id, (size_t)n * sizeof(int));
-            v->in_edge_weight = (int*)abamd_realloc(v->in_edge_weight, (size_t)n * sizeof(int));
-            v->in_edge_m = n;
-        }
-        v->in_edge_n = n;
-        n = 0;
-        for (e = fg->in_head[id]; e != -1; e = fg->in_next[e]) {
-            v->in_id[n] = fg->in_to[e];
-            v->in_edge_weight[n] = fg->in_w[e];
-            ++n;
-        }
+        v->in_id = in_pool + c_in * 2; /* ids then weights, contiguous per node */
+        int cnt = 0;
+        for (e = fg->in_head[id]; e != -1; e = fg->in_next[e]) v->in_id[cnt++] = fg->in_to[e];
+        v->in_edge_weight = v->in_id + cnt;
+        cnt = 0;
+        for (e = fg->in_head[id]; e != -1; e = fg->in_next[e]) v->in_edge_weight[cnt++] = fg->in_w[e];
+        v->in_edge_n = v->in_edge_m = cnt;
+        c_in += (size_t)cnt;
         /* out edges (+ read-id bitsets travelling with each edge) */
-        n = 0;
-        for (e = fg->out_head[id]; e != -1; e = fg->out_next[e]) ++n;
-        if (n > v->out_edge_m) {
-            int i, old = v->out_edge_m;
-            v->out_id = (int*)abamd_realloc(v->out_id, (size_t)n * sizeof(int));
-            v->out_edge_weight = (int*)abamd_realloc(v->out_edge_weight, (size_t)n * sizeof(int));
-            if (read_ids_n > 0 || v->read_ids) {
-                v->read_ids = (uint64_t**)abamd_realloc(v->read_ids, (size_t)n * sizeof(uint64_t*));
-                for (i = old; i < n; ++i) v->read_ids[i] = NULL;
-            }
-            v->out_edge_m = n;
-        }
-        v->out_edge_n = n;
-        n = 0;
-        for (e = fg->out_head[id]; e != -1; e = fg->out_next[e]) {
-            v->out_id[n] = fg->out_to[e];
-            v->out_edge_weight[n] = fg->out_w[e];
-            if (read_ids_n > 0) {
-                if (!v->read_ids[n])
-                    v->read_ids[n] = (uint64_t*)abamd_malloc((size_t)read_ids_n * sizeof(uint64_t));
-                memcpy(v->read_ids[n], fg->rid_pool + (size_t)e * fg->rid_n,
+        v->out_id = out_pool + c_out * 2;
+        cnt = 0;
+        for (e = fg->out_head[id]; e != -1; e = fg->out_next[e]) v->out_id[cnt++] = fg->out_to[e];
+        v->out_edge_weight = v->out_id + cnt;
+        cnt = 0;
+        for (e = fg->out_head[id]; e != -1; e = fg->out_next[e]) v->out_edge_weight[cnt++] = fg->out_w[e];
+        if (read_ids_n > 0 && cnt > 0) {
+            v->read_ids = ridp_pool + c_out;
+            int k = 0;
+            for (e = fg->out_head[id]; e != -1; e = fg->out_next[e]) {
+                v->read_ids[k] = rid_pool + (c_out + (size_t)k) * read_ids_n;
+                memcpy(v->read_ids[k], fg->rid_pool + (size_t)e * fg->rid_n,
                        (size_t)read_ids_n * sizeof(uint64_t));
+                ++k;
             }
-            ++n;
+            v->read_ids_n = read_ids_n;
         }
-        if (read_ids_n > 0 && v->out_edge_n > 0) v->read_ids_n = read_ids_n;
+        v->out_edge_n = v->out_edge_m = cnt;
+        c_out += (size_t)cnt;
         /* aligned (mismatch-column) group, chain order */
-        n = 0;
-        for (e = fg->aln_head[id]; e != -1; e = fg->aln_next[e]) ++n;
-        if (n > v->aligned_node_m) {
-            v->aligned_node_id = (int*)abamd_realloc(v->aligned_node_id, (size_t)n * sizeof(int));
-            v->aligned_node_m = n;
-        }
-        v->aligned_node_n = n;
-        n = 0;
-        for (e = fg->aln_head[id]; e != -1; e = fg->aln_next[e])
-            v->aligned_node_id[n++] = fg->aln_id[e];
+        v->aligned_node_id = aln_pool + c_aln;
+        cnt = 0;
+        for (e = fg->aln_head[id]; e != -1; e = fg->aln_next[e]) v->aligned_node_id[cnt++] = fg->aln_id[e];
+        v->aligned_node_n = v->aligned_node_m = cnt;
+        c_aln += (size_t)cnt;
     }
+    /* swap the slab-carved node array in; the original (tiny) array is freed
+     * here and restored by abamd_graph_arena_release */
+    for (id = 0; id < g->node_m; ++id) free_node_arrays(&g->node[id]);
+    free(g->node);
+    g->node = nodes;
     g->node_n = fg->node_n;
+    g->node_m = fg->node_n;
     g->is_topological_sorted = g->is_called_cons = g->is_set_msa_rank = 0;
     abpoa_topological_sort(g, abpt);
+    return slab;
+}
+
+/* undo abamd_graph_from_flat's slab carve so abpoa_free() is valid again */
+void abamd_graph_arena_release(abpoa_t *ab, void *slab) {
+    abpoa_graph_t *g = ab->abg;
+    g->node = (abpoa_node_t*)abamd_calloc(2, sizeof(abpoa_node_t));
+    init_node(&g->node[0], ABPOA_SRC_NODE_ID);
+    init_node(&g->node[1], ABPOA_SINK_NODE_ID);
+    g->node_n = g->node_m = 2;
+    free(slab);
 }

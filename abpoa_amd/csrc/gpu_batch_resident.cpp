@@ -778,7 +778,10 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
             RHIP_CHECK(hipMemGetInfo(&free_b, &total_b));
             uint64_t arena_held = 0;
             for (int i = 0; i < 8; ++i) arena_held += C.slot[i].arena.cap + C.slot[i].cigars.cap;
-            mem_gb = (double)(free_b + arena_held) * 0.84 / n_groups / 1e9;
+            /* n_groups pipeline slots PLUS the retry/big slot 7 each hold an
+             * arena; splitting only across the pipeline slots made slot 7's
+             * first big-item allocation churn tens of GB of hipMalloc/hipFree */
+            mem_gb = (double)(free_b + arena_held) * 0.84 / (n_groups + 1) / 1e9;
         }
         B.budget_bytes = mem_gb * 1e9;
     }
@@ -987,17 +990,19 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
         Batch *B;
         std::vector<flat_graph_t> *hflat;
         std::vector<abpoa_t*> abs;
+        std::vector<void*> slabs;
         abpoa_para_t *abpt;
     } cc;
     cc.B = &B; cc.hflat = &hflat; cc.abpt = abpt;
     cc.abs.assign(n_sets, nullptr);
+    cc.slabs.assign(n_sets, nullptr);
     auto cons_worker = [](void *p, int, int) {
         ConsCtx &c = *(ConsCtx*)p;
         for (;;) {
             int s = c.next.fetch_add(1);
             if (s >= c.B->n_sets) break;
             abpoa_t *ab = abpoa_init();
-            abamd_graph_from_flat(ab, &(*c.hflat)[s], c.abpt, (*c.hflat)[s].rid_n);
+            c.slabs[s] = abamd_graph_from_flat(ab, &(*c.hflat)[s], c.abpt, (*c.hflat)[s].rid_n);
             ab->abs->n_seq = c.B->sets[s].n_seqs;
             abpoa_generate_consensus(ab, c.abpt);
             c.abs[s] = ab;
@@ -1007,6 +1012,7 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
     if (cb)
         for (int s = 0; s < n_sets; ++s) cb(s, cc.abs[s]->abc, user);
     for (int s = 0; s < n_sets; ++s) {
+        abamd_graph_arena_release(cc.abs[s], cc.slabs[s]);
         abpoa_free(cc.abs[s]);
         flat_graph_t &h = hflat[s];
         free(h.base); free(h.n_read); free(h.n_span_read);

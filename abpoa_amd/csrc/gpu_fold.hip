@@ -127,9 +127,22 @@ __device__ static void dev_topo_bfs(const flat_graph_t *fg, int *index_to_node_i
     q[qt++] = 0; /* SRC */
     while (qh < qt) {
         cur = q[qh++];
+    chain:
         index_to_node_id[index] = cur;
         node_id_to_index[cur] = index++;
         if (cur == 1 /* SINK */) return;
+        /* chain fast path: when the queue is otherwise drained and cur has a
+         * single out-edge whose target becomes ready, the general code would
+         * push exactly that node and dequeue it next — skip the queue */
+        e = fg->out_head[cur];
+        if (qh == qt && e != -1 && fg->out_next[e] == -1) {
+            int out = fg->out_to[e];
+            if (in_deg[out] == 1 && fg->aln_head[out] == -1) {
+                in_deg[out] = 0;
+                cur = out;
+                goto chain;
+            }
+        }
         for (e = fg->out_head[cur]; e != -1; e = fg->out_next[e]) {
             int out = fg->out_to[e];
             if (--in_deg[out] == 0) {
@@ -182,19 +195,44 @@ __device__ static void dev_par_sort_adjacency(flat_graph_t *fg, int lane) {
     }
 }
 
-/* max_remain by REVERSE topo-index scan: every successor has a higher index,
- * so its value is final when a node is visited — the same pure function the
- * twin's reverse BFS computes (max-weight out-edge, strict >, chain-order
- * tie-break), minus the queue machinery */
-__device__ static void dev_remain_reverse(const flat_graph_t *fg, const int *index_to_node_id,
-                                          int sink_index, int *max_remain) {
-    for (int k = sink_index; k >= 0; --k) {
-        int id = index_to_node_id[k];
-        if (id == 1 /* SINK */) { max_remain[id] = -1; continue; }
-        int max_w = INT32_MIN, max_id = -1;
-        for (int e = fg->out_head[id]; e != -1; e = fg->out_next[e])
-            if (fg->out_w[e] > max_w) { max_w = fg->out_w[e]; max_id = fg->out_to[e]; }
-        max_remain[id] = max_remain[max_id] + 1;
+/* max_remain: every successor has a HIGHER topo index, so scanning indices
+ * high->low makes each node's value a pure function of already-final
+ * successors — the same values the twin's reverse BFS computes (max-weight
+ * out-edge, strict >, chain-order tie-break). Wave-parallel: 64 indices per
+ * chunk; the (parallel) chain walks find each node's max-weight successor,
+ * then a shuffle wavefront resolves intra-chunk dependence (successor lanes
+ * are strictly higher, so it converges in <= 64 steps; out-of-chunk
+ * successors were finalized by earlier chunks). */
+__device__ static void dev_par_remain(const flat_graph_t *fg, const int *index_to_node_id,
+                                      const int *node_id_to_index, int sink_index,
+                                      int *max_remain, int lane) {
+    for (int base = (sink_index / FOLD_WAVE) * FOLD_WAVE; base >= 0; base -= FOLD_WAVE) {
+        const int k = base + lane;
+        int id = -1, src = 0, val = 0;
+        bool have = true;
+        if (k <= sink_index) {
+            id = index_to_node_id[k];
+            if (id == 1 /* SINK */) val = -1;
+            else {
+                int max_w = INT32_MIN, max_id = -1;
+                for (int e = fg->out_head[id]; e != -1; e = fg->out_next[e])
+                    if (fg->out_w[e] > max_w) { max_w = fg->out_w[e]; max_id = fg->out_to[e]; }
+                const int sidx = node_id_to_index[max_id];
+                if (sidx > base + (FOLD_WAVE - 1)) val = max_remain[max_id] + 1;
+                else { src = sidx - base; have = false; }
+            }
+        }
+        unsigned long long done = __ballot(have ? 1 : 0);
+        int guard = 0;
+        while (done != ~0ull) {
+            const int sval = __shfl(val, src);
+            if (!have && ((done >> src) & 1)) { val = sval + 1; have = true; }
+            done = __ballot(have ? 1 : 0);
+            /* each step resolves at least the highest unresolved lane
+             * (successor indices are strictly higher); > 64 steps = bug */
+            if (++guard > FOLD_WAVE) abort();
+        }
+        if (k <= sink_index) max_remain[id] = val;
     }
 }
 
@@ -381,8 +419,10 @@ void abamd_fold_round_kernel(abamd_fold_round_job_t *jobs, int n_jobs) {
     FKPROF_T(ft4);
     FKPROF_ACC(4, ft3, ft4);
     const int sink_index = job->node_id_to_index[1 /* SINK */];
-    if (job->use_remain && lane == 0)
-        dev_remain_reverse(g, job->index_to_node_id, sink_index, job->max_remain);
+    if (job->use_remain)
+        dev_par_remain(g, job->index_to_node_id, job->node_id_to_index, sink_index,
+                       job->max_remain, lane);
+    __syncthreads();
     dev_par_n_span(g, job->index_to_node_id, job->node_id_to_index[0 /* SRC */],
                    sink_index, lane);
     __syncthreads();
