@@ -63,6 +63,7 @@ inline uint64_t now_ns() {
 }
 
 uint64_t g_held; /* bytes held by this driver's reusable device buffers */
+double g_alloc_s; long g_alloc_n; /* hipMalloc churn diagnostics */
 
 struct DevBuf {
     void *p = nullptr;
@@ -80,8 +81,11 @@ struct DevBuf {
             want = n > limit ? n : limit;
         }
         if (want < 4096) want = 4096;
+        double t0 = abamd_realtime();
         if (p) RHIP_CHECK(hipFree(p));
         RHIP_CHECK(hipMalloc(&p, want));
+        g_alloc_s += abamd_realtime() - t0;
+        g_alloc_n += 1;
         g_held += want - cap;
         cap = want;
     }
@@ -251,6 +255,8 @@ struct Batch {
     double budget_bytes = 0;
     uint64_t retry_jobs = 0, pool_expands = 0;
     double t_host_build = 0, t_finish = 0;
+    double t_big_build = 0, t_big_gpu = 0;
+    long big_chunks = 0;
 };
 
 int64_t job_est_cells(const Batch &B, const SetState &S, int qlen) {
@@ -524,9 +530,10 @@ void settle_fold(Batch &B, Slot &S, int i) {
 
 void finish_slot(Batch &B, Slot &S);
 
-/* DP arena overflow: retry the failed jobs on the dedicated slot 7 with
+/* DP arena overflow: retry the failed jobs on the dedicated slot 5 with
  * doubled reservations (their folds were skipped — the retry launch carries
- * its own fold pass, so graph state catches up there). */
+ * its own fold pass, so graph state catches up there; slots 6/7 belong to
+ * the big-item double buffer and may be in flight while this runs). */
 void retry_failed(Batch &B, Slot &S) {
     const abamd_gpu_res_t *res = (const abamd_gpu_res_t*)S.h_read.p;
     std::vector<int> failed;
@@ -542,7 +549,8 @@ void retry_failed(Batch &B, Slot &S) {
     }
     if (failed.empty()) return;
     B.retry_jobs += failed.size();
-    Slot &R = g_ctx.slot[7];
+    Slot &R = g_ctx.slot[5];
+    R.ensure_init();
     for (int attempt = 0;; ++attempt) {
         if (attempt > 8) {
             fprintf(stderr, "[abpoa_amd] arena overflow persists after %d retries\n", attempt);
@@ -762,10 +770,11 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
     /* ---- pipeline-group plan + per-launch arena budget ---- */
     int n_groups = n_sets >= 3 ? 3 : n_sets;
     {
+        /* slots 0..3 pipeline, 5 retry, 6/7 big-item double buffer */
         const char *gs = getenv("ABPOA_AMD_GROUPS");
         if (gs && *gs) {
             int gg = atoi(gs);
-            if (gg >= 1 && gg <= 6 && gg <= n_sets) n_groups = gg;
+            if (gg >= 1 && gg <= 4 && gg <= n_sets) n_groups = gg;
         }
     }
     B.n_groups = n_groups;
@@ -882,9 +891,18 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
                 out_list.push_back(s);
     };
 
-    /* launch one item; oversized items run synchronously in budget-bounded
-     * chunks on slot 7 (never clobbering an in-flight pipeline slot) */
+    /* launch one item; oversized items run in budget-bounded chunks
+     * double-buffered across slots 6/7 (chunks hold disjoint sets, so
+     * chunk k+1's kernels overlap chunk k's drain; pipeline slots are
+     * never clobbered) */
     std::vector<int> list, chunk;
+    auto job_bytes = [&](int set_idx, int r) -> double {
+        SetState &st = B.sets[set_idx];
+        const int qlen = st.seq_lens[r];
+        int bits, im;
+        abamd_pick_width(abpt, qlen, st.n_rows, &bits, &im);
+        return (double)job_est_cells(B, st, qlen) * B.planes * (bits == 16 ? 2 : 4);
+    };
     auto launch_item = [&](long it, int slot_i) {
         double tb0 = abamd_realtime();
         const int r = 1 + (int)(it / n_groups);
@@ -892,30 +910,38 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
         Slot &S = C.slot[slot_i];
         S.ensure_init();
         if (list.empty()) { S.active = false; B.t_host_build += abamd_realtime() - tb0; return; }
-        /* arena demand vs budget */
-        const size_t ssz0 = 2; /* estimate at int16; widths re-checked in build */
         double demand = 0;
-        for (int idx : list)
-            demand += (double)job_est_cells(B, B.sets[idx], B.sets[idx].seq_lens[r]) * B.planes * ssz0;
+        for (int idx : list) demand += job_bytes(idx, r);
         if (demand > B.budget_bytes && list.size() > 1) {
-            /* big item: sequential chunks on slot 7 */
+            Slot *bs[2] = {&C.slot[6], &C.slot[7]};
+            bs[0]->ensure_init(); bs[1]->ensure_init();
+            int par = 0;
             size_t done = 0;
             while (done < list.size()) {
                 double acc = 0;
                 chunk.clear();
                 while (done < list.size()) {
-                    double e = (double)job_est_cells(B, B.sets[list[done]], B.sets[list[done]].seq_lens[r]) * B.planes * ssz0;
+                    double e = job_bytes(list[done], r);
                     if (!chunk.empty() && acc + e > B.budget_bytes) break;
                     acc += e;
                     chunk.push_back(list[done]);
                     ++done;
                 }
-                Slot &R = C.slot[7];
-                R.ensure_init();
+                Slot &R = *bs[par];
+                double tb = abamd_realtime();
+                if (R.active) finish_slot(B, R); /* drain the older chunk */
+                double tg = abamd_realtime();
                 build_jobs(B, R, chunk, r, nullptr);
                 launch_slot(B, R);
-                finish_slot(B, R);
+                B.t_big_build += abamd_realtime() - tg;
+                B.t_big_gpu += tg - tb;
+                B.big_chunks += 1;
+                par ^= 1;
             }
+            double tw = abamd_realtime();
+            if (bs[par]->active) finish_slot(B, *bs[par]);
+            if (bs[par ^ 1]->active) finish_slot(B, *bs[par ^ 1]);
+            B.t_big_gpu += abamd_realtime() - tw;
             S.active = false;
             B.t_host_build += abamd_realtime() - tb0;
             return;
@@ -1028,12 +1054,16 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
     if (getenv("ABPOA_AMD_TIMING")) {
         double busy = 0, dpms = 0, foldms = 0;
         abpoa_amd_get_gpu_spans(&busy, &dpms, &foldms);
-        fprintf(stderr, "[abamd timing resident] setup %.2fs rounds %.2fs (host build %.2fs finish-wait %.2fs) "
+        fprintf(stderr, "[abamd timing resident] setup %.2fs rounds %.2fs (host build %.2fs finish-wait %.2fs "
+                        "big: %ld chunks build %.2fs gpu %.2fs) "
                         "cons %.2fs (dl %.2fs) | dp-kernel-sum %.2fs fold-kernel-sum %.2fs gpu-busy %.2fs "
                         "retry_jobs %llu pool_expands %llu\n",
-                t_setup, t_rounds, B.t_host_build, B.t_finish, t_cons, t_dl,
+                t_setup, t_rounds, B.t_host_build, B.t_finish,
+                B.big_chunks, B.t_big_build, B.t_big_gpu, t_cons, t_dl,
                 dpms / 1e3, foldms / 1e3, busy / 1e3,
                 (unsigned long long)B.retry_jobs, (unsigned long long)B.pool_expands);
+        fprintf(stderr, "[abamd timing resident] hipMalloc %ld calls %.2fs\n",
+                g_alloc_n, g_alloc_s);
     }
     return 0;
 }
