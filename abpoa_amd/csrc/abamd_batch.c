@@ -233,7 +233,7 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
         if (_alt > _ms) _ms = _alt; \
         int _oe1 = abpt->gap_open1 + abpt->gap_ext1, _oe2 = abpt->gap_open2 + abpt->gap_ext2; \
         int _ssz = (_ms <= 32767 - abpt->min_mis - _oe1 - _oe2) ? 2 : 4; \
-        int _pl = abpt->gap_mode == ABPOA_CONVEX_GAP ? 5 : abpt->gap_mode == ABPOA_AFFINE_GAP ? 3 : 1; \
+        int _pl = abpt->gap_mode == ABPOA_CONVEX_GAP ? 3 : abpt->gap_mode == ABPOA_AFFINE_GAP ? 3 : 1; \
         double _cells = (double)_gn * (2.0 * _w + 160.0); \
         if ((J)->est_cells_hint > 0) { \
             double _t = (double)(J)->est_cells_hint * 1.5 + _q; \

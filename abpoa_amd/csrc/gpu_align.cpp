@@ -434,7 +434,8 @@ static int prepare_internal(BatchJob *batch, int n_jobs, int slot, const int64_t
     GpuCtx &C = g_ctx;
     C.ensure_init();
     abpoa_para_t *abpt = batch[0].abpt;
-    const int planes = abpt->gap_mode == ABPOA_CONVEX_GAP ? 5
+    /* convex stores 3 planes (H,E1,E2): F recomputed at backtrack */
+    const int planes = abpt->gap_mode == ABPOA_CONVEX_GAP ? 3
                      : abpt->gap_mode == ABPOA_AFFINE_GAP ? 3 : 1;
 
     uint64_t t_pack0 = now_ns();

@@ -679,7 +679,7 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
     B.abpt = abpt;
     B.n_sets = n_sets;
     B.use_remain = (abpt->wb >= 0 || abpt->zdrop > 0);
-    B.planes = abpt->gap_mode == ABPOA_CONVEX_GAP ? 5
+    B.planes = abpt->gap_mode == ABPOA_CONVEX_GAP ? 3 /* F recomputed at backtrack */
              : abpt->gap_mode == ABPOA_AFFINE_GAP ? 3 : 1;
     B.sets.resize(n_sets);
 
@@ -787,10 +787,10 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
             RHIP_CHECK(hipMemGetInfo(&free_b, &total_b));
             uint64_t arena_held = 0;
             for (int i = 0; i < 8; ++i) arena_held += C.slot[i].arena.cap + C.slot[i].cigars.cap;
-            /* n_groups pipeline slots PLUS the retry/big slot 7 each hold an
-             * arena; splitting only across the pipeline slots made slot 7's
-             * first big-item allocation churn tens of GB of hipMalloc/hipFree */
-            mem_gb = (double)(free_b + arena_held) * 0.84 / (n_groups + 1) / 1e9;
+            /* every arena-holding slot can reach the cap at once: n_groups
+             * pipeline slots + the big-item double-buffer pair + the retry
+             * slot — splitting across fewer shares OOMs a full-HBM run */
+            mem_gb = (double)(free_b + arena_held) * 0.84 / (n_groups + 3) / 1e9;
         }
         B.budget_bytes = mem_gb * 1e9;
     }

@@ -141,572 +141,6 @@ __device__ static int dev_push_cigar(uint64_t *cig, int *n_c, int cap, int op, i
 }
 
 /* ------------------------------------------------------------------ */
-/* The DP + backtrack kernel (global alignment, convex gaps).          */
-/* ------------------------------------------------------------------ */
-template <typename S>
-__global__ __launch_bounds__(WAVE * JOBS_PER_BLOCK, 7)
-void cg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
-                      abamd_gpu_res_t *__restrict__ results, int n_jobs) {
-    const int wid = threadIdx.x / WAVE;
-    const int jid = blockIdx.x * JOBS_PER_BLOCK + wid;
-    const int lane = threadIdx.x % WAVE;
-
-    __shared__ int mat_lds[27 * 27];
-    /* per-job double-buffered previous-row cache: H,E1,E2 planes */
-    constexpr int BMAX = BmaxOf<S>::v;
-    __shared__ S prev_lds[JOBS_PER_BLOCK][2][3 * BMAX];
-    {
-        /* every job of a batch shares one scoring matrix (gpu_align.cpp) */
-        const int m0 = jobs[0].m;
-        const int *mat0 = jobs[0].mat;
-        for (int i = threadIdx.x; i < m0 * m0; i += WAVE * JOBS_PER_BLOCK)
-            mat_lds[i] = mat0[i];
-    }
-    __syncthreads();
-    if (jid >= n_jobs) return;
-    /* by VALUE: every field lives in registers (uniform loads become
-       s_loads into SGPRs) — a reference would re-load fields from global
-       memory inside the row loop because arena/steering stores could alias
-       the jobs array in the compiler's view */
-    const abamd_gpu_job_t jb = jobs[jid];
-    abamd_gpu_res_t *res = &results[jid];
-    abamd_row_meta_t *__restrict__ meta = (abamd_row_meta_t*)jb.row_meta;
-
-    const int qlen = jb.qlen, n_rows = jb.n_rows, w = jb.w, m = jb.m;
-    const S inf_min = (S)jb.inf_min;
-    const S o1 = (S)jb.o1, o2 = (S)jb.o2, e1 = (S)jb.e1, e2 = (S)jb.e2;
-    const S oe1 = (S)jb.oe1, oe2 = (S)jb.oe2;
-    (void)o1; (void)o2;
-    const int local_mode = jb.align_mode == 1, extend_mode = jb.align_mode == 2;
-    int32_t run_best = jb.inf_min;      /* local/extend running best */
-    int run_best_i = 0, run_best_j = 0, run_best_remain = jb.max_remain[0];
-    int zdropped = 0;
-    const int end_remain = jb.max_remain[n_rows - 1];
-    S *arena = (S*)jb.arena;
-    const uint8_t *__restrict__ query = jb.query;
-
-    if (lane == 0) {
-        res->status = ABAMD_JOB_OK;
-        res->n_cigar = 0;
-    }
-
-    /* adaptive band state init (abpoa_topological_sort:347-353 + first_dp) */
-    for (int i = lane; i < n_rows; i += WAVE) {
-        jb.max_left[i] = jb.node_n_init;
-        jb.max_right[i] = 0;
-    }
-    if (lane == 0) { jb.max_left[0] = 0; jb.max_right[0] = 0; }
-    for (int k = jb.out_off[0] + lane; k < jb.out_off[1]; k += WAVE) {
-        int o = jb.out_idx[k];
-        jb.max_left[o] = 1; jb.max_right[o] = 1;
-    }
-
-    /* LDS cache state: which buffer holds the previous row's planes */
-    int buf_cur = 0;
-    int prev_ok = 0, prev_row = -1, prev_beg = 0, prev_end = -1;
-
-    /* ---- first row (simd_abpoa_cg_first_dp) ---- */
-    int64_t used;
-    {
-        int mr = jb.max_remain[0] - end_remain - 1;
-        int end0;
-        if (jb.banded) {
-            int t = jb.max_right[0] > qlen - mr ? jb.max_right[0] : qlen - mr;
-            end0 = (qlen < t + w) ? qlen : t + w;
-        } else end0 = qlen;
-        if (lane == 0) { meta[0].beg = 0; meta[0].end = end0; meta[0].off = 0; }
-        int64_t bw = end0 + 1;
-        used = bw;
-        S *H = arena, *E1 = arena + bw, *E2 = arena + 2 * bw, *F1 = arena + 3 * bw, *F2 = arena + 4 * bw;
-        S *c = &prev_lds[wid][buf_cur][0];
-        const int fits = end0 + 1 <= BMAX;
-        for (int j = lane; j <= end0; j += WAVE) {
-            S hv, e1v2, e2v2;
-            if (local_mode) {
-                hv = 0; e1v2 = 0; e2v2 = 0;
-                F1[j] = 0; F2[j] = 0;
-            } else if (j == 0) {
-                hv = 0; e1v2 = (S)(0 - oe1); e2v2 = (S)(0 - oe2);
-                F1[0] = inf_min; F2[0] = inf_min;
-            } else {
-                S f1 = (S)(-(jb.o1 + jb.e1 * j));
-                S f2 = (S)(-(jb.o2 + jb.e2 * j));
-                F1[j] = f1; F2[j] = f2;
-                hv = smax(f1, f2);
-                e1v2 = inf_min; e2v2 = inf_min;
-            }
-            H[j] = hv; E1[j] = e1v2; E2[j] = e2v2;
-            if (fits) {
-                c[j] = hv; c[BMAX + j] = e1v2; c[2 * BMAX + j] = e2v2;
-            }
-        }
-        if (fits) { prev_ok = 1; prev_row = 0; prev_beg = 0; prev_end = end0; }
-        buf_cur ^= 1;
-    }
-
-    /* ---- main row loop ----
-     * Per-row scalars (CSR offsets, remain, band steering, first
-     * predecessor) roll ONE ROW AHEAD: their loads issue during the
-     * previous row's chunk work, hiding the dependent-load latency that
-     * dominated round 1 (76% WAIT_ANY, profiles/r01_sq_stalls.txt). The
-     * common band push r -> r+1 is carried in registers; max_left/right
-     * only round-trip memory for out-edges that skip rows, whose values
-     * are final before the prefetch that reads them (pushes from rows
-     * < r land before row r issues row r+1's loads; row r's own push to
-     * r+1 never touches memory). */
-    int cur_pk0 = jb.pre_off[1];
-    int cur_pk1 = jb.pre_off[2 <= n_rows ? 2 : n_rows];
-    int cur_oo0 = jb.out_off[1];
-    int cur_oo1 = jb.out_off[2 <= n_rows ? 2 : n_rows];
-    int cur_remain = jb.max_remain[1];
-    int cur_ml_mem = jb.max_left[1];
-    int cur_mr_mem = jb.max_right[1];
-    int cur_pidx0 = jb.pre_idx[cur_pk0];
-    int cur_ps0 = jb.pre_ps[cur_pk0];
-    int push_ml = 0x7fffffff, push_mr = -0x7fffffff;
-#ifdef ABAMD_KPROF
-    unsigned long long kp_band = 0, kp_chunk = 0, kp_epi = 0, kp_rows = 0, kp_chunks = 0;
-    unsigned long long kp_g = 0, kp_s = 0, kp_w = 0;
-#endif
-    for (int r = 1; r < n_rows - 1; ++r) {
-        KPROF_T(kt0);
-        const int pk0 = cur_pk0, pk1 = cur_pk1;
-        const int oo0 = cur_oo0, oo1 = cur_oo1;
-        const int row_remain = cur_remain;
-        const int pidx0 = cur_pidx0;
-        const S ps0 = (S)cur_ps0;
-        const int ml_eff = cur_ml_mem < push_ml ? cur_ml_mem : push_ml;
-        const int mr_eff = cur_mr_mem > push_mr ? cur_mr_mem : push_mr;
-        {   /* prefetch row r+1's scalars (independent loads) */
-            const int nr = r + 1;
-            cur_pk0 = pk1;
-            cur_pk1 = jb.pre_off[nr + 1];
-            cur_oo0 = oo1;
-            cur_oo1 = jb.out_off[nr + 1];
-            cur_remain = jb.max_remain[nr];
-            cur_ml_mem = jb.max_left[nr];
-            cur_mr_mem = jb.max_right[nr];
-            cur_pidx0 = jb.pre_idx[cur_pk0];
-            cur_ps0 = jb.pre_ps[cur_pk0];
-        }
-        int beg, end;
-        {
-            int mr = row_remain - end_remain - 1;
-            if (jb.banded) {
-                int lo = ml_eff < qlen - mr ? ml_eff : qlen - mr;
-                beg = lo - w; if (beg < 0) beg = 0;
-                int hi = mr_eff > qlen - mr ? mr_eff : qlen - mr;
-                end = hi + w; if (end > qlen) end = qlen;
-                int min_pre_beg;
-                if (pk1 - pk0 == 1) {
-                    min_pre_beg = (prev_ok && pidx0 == prev_row) ? prev_beg : meta[pidx0].beg;
-                } else {
-                    min_pre_beg = 0x7fffffff;
-                    for (int k = pk0; k < pk1; ++k) {
-                        const int pidx = jb.pre_idx[k];
-                        int pb = (prev_ok && pidx == prev_row) ? prev_beg : meta[pidx].beg;
-                        if (pb < min_pre_beg) min_pre_beg = pb;
-                    }
-                }
-                if (beg < min_pre_beg) beg = min_pre_beg;
-            } else { beg = 0; end = qlen; }
-        }
-        const int64_t bw = end - beg + 1;
-        if (used + bw > jb.arena_cap) { if (lane == 0) res->status = ABAMD_JOB_ARENA_OVERFLOW; return; }
-        const int64_t off = used;
-        if (lane == 0) { meta[r].beg = beg; meta[r].end = end; meta[r].off = off; }
-        used += bw;
-        S *H = arena + off * 5, *E1r = H + bw, *E2r = E1r + bw, *F1r = E2r + bw, *F2r = F1r + bw;
-        const uint8_t base = jb.row_base[r];
-        const int *mrow = &mat_lds[base * m];
-        const int cache_fits = bw <= BMAX;
-        S *cw = &prev_lds[wid][buf_cur][0];          /* written for the next row */
-        const S *cr = &prev_lds[wid][buf_cur ^ 1][0]; /* previous row's planes */
-
-        /* the overwhelmingly common row shape (in-degree 1, predecessor =
-         * previous row, cached in LDS): zero global loads in the gather */
-        const bool fast1 = (pk1 - pk0 == 1) && prev_ok && (pidx0 == prev_row);
-
-        /* per-row preload of the query scores: the per-chunk global byte
-         * load sat on the chunk critical path (kprof: the gather phase was
-         * ~4k of ~5k cycles per chunk); issuing every chunk's load at row
-         * start overlaps the latency across the whole row */
-        constexpr int QPRE = 8;
-        S qpre[QPRE];
-        {
-            const int n_ch = (int)((end - beg) / WAVE) + 1;
-            #pragma unroll
-            for (int c = 0; c < QPRE; ++c) {
-                const int jj = beg + c * WAVE + lane;
-                qpre[c] = (c < n_ch && jj >= 1 && jj <= end) ? (S)mrow[query[jj - 1]] : (S)0;
-            }
-        }
-
-        S carry_h = inf_min, f1c = inf_min, f2c = inf_min;
-        S lmax = inf_min; int lleft = -1, lright = -1;
-
-        KPROF_T(kt1);
-#ifdef ABAMD_KPROF
-        unsigned long long kc_prev = kt1;
-#endif
-        for (int cs = beg; cs <= end; cs += WAVE) {
-            const int j = cs + lane;
-            const bool act = j <= end;
-            S h = inf_min, e1v = inf_min, e2v = inf_min;
-            if (fast1) {
-                if (act) {
-                    if (local_mode && j == 0) { if (ps0 > h) h = ps0; }
-                    if (j - 1 >= prev_beg && j - 1 <= prev_end) {
-                        S v = (S)(cr[j - 1 - prev_beg] + ps0);
-                        if (v > h) h = v;
-                    }
-                    if (j >= prev_beg && j <= prev_end) {
-                        S v1 = (S)(cr[BMAX + j - prev_beg] + ps0);
-                        S v2 = (S)(cr[2 * BMAX + j - prev_beg] + ps0);
-                        if (v1 > e1v) e1v = v1;
-                        if (v2 > e2v) e2v = v2;
-                    }
-                }
-            } else for (int k = pk0; k < pk1; ++k) {
-                const int p = jb.pre_idx[k];
-                const S ps = (S)jb.pre_ps[k];
-                if (prev_ok && p == prev_row) {
-                    if (act) { /* LDS fast path */
-                        if (local_mode && j == 0) { if (ps > h) h = ps; }
-                        if (j - 1 >= prev_beg && j - 1 <= prev_end) {
-                            S v = (S)(cr[j - 1 - prev_beg] + ps);
-                            if (v > h) h = v;
-                        }
-                        if (j >= prev_beg && j <= prev_end) {
-                            S v1 = (S)(cr[BMAX + j - prev_beg] + ps);
-                            S v2 = (S)(cr[2 * BMAX + j - prev_beg] + ps);
-                            if (v1 > e1v) e1v = v1;
-                            if (v2 > e2v) e2v = v2;
-                        }
-                    }
-                    continue;
-                }
-                const abamd_row_meta_t pm = meta[p];
-                const int64_t pbw = pm.end - pm.beg + 1;
-                const S *__restrict__ pH = arena + pm.off * 5;
-                const S *__restrict__ pE1 = pH + pbw;
-                const S *__restrict__ pE2 = pE1 + pbw;
-                if (act) {
-                    if (local_mode && j == 0) { if (ps > h) h = ps; }
-                    if (j - 1 >= pm.beg && j - 1 <= pm.end) {
-                        S v = (S)(pH[j - 1 - pm.beg] + ps);
-                        if (v > h) h = v;
-                    }
-                    if (j >= pm.beg && j <= pm.end) {
-                        S v1 = (S)(pE1[j - pm.beg] + ps);
-                        S v2 = (S)(pE2[j - pm.beg] + ps);
-                        if (v1 > e1v) e1v = v1;
-                        if (v2 > e2v) e2v = v2;
-                    }
-                }
-            }
-            const int ci = (cs - beg) / WAVE;
-            const S q = ci < QPRE ? qpre[ci]
-                                  : (S)((j == 0 || !act) ? 0 : mrow[query[j - 1]]);
-            KPROF_T(kc1);
-            S hpre = (S)(h + q);
-            hpre = smax(hpre, smax(e1v, e2v));
-            if (!act) hpre = inf_min;
-
-            /* F scan: candidates c[j] = max(Hpre[j-1]-oe, chain) */
-            S hshift = (S)__shfl_up((int)hpre, 1);
-            S c1, c2;
-            if (lane == 0) {
-                if (cs == beg) { c1 = (S)(inf_min - oe1); c2 = (S)(inf_min - oe2); }
-                else {
-                    c1 = smax((S)(carry_h - oe1), (S)(f1c - e1));
-                    c2 = smax((S)(carry_h - oe2), (S)(f2c - e2));
-                }
-            } else {
-                c1 = (S)(hshift - oe1);
-                c2 = (S)(hshift - oe2);
-            }
-            S f1 = scan_maxplus(c1, jb.e1, inf_min, lane);
-            S f2 = scan_maxplus(c2, jb.e2, inf_min, lane);
-            carry_h = (S)__builtin_amdgcn_readlane((int)hpre, WAVE - 1);
-            f1c = (S)__builtin_amdgcn_readlane((int)f1, WAVE - 1);
-            f2c = (S)__builtin_amdgcn_readlane((int)f2, WAVE - 1);
-
-            KPROF_T(kc2);
-            S hf = smax(hpre, smax(f1, f2));
-            if (local_mode) hf = smax(hf, (S)0);
-            S e1n = smax((S)(e1v - e1), (S)(hf - oe1));
-            S e2n = smax((S)(e2v - e2), (S)(hf - oe2));
-            if (local_mode) { e1n = smax(e1n, (S)0); e2n = smax(e2n, (S)0); }
-            if (act) {
-                H[j - beg] = hf; E1r[j - beg] = e1n; E2r[j - beg] = e2n;
-                F1r[j - beg] = f1; F2r[j - beg] = f2;
-                if (cache_fits) {
-                    cw[j - beg] = hf;
-                    cw[BMAX + j - beg] = e1n;
-                    cw[2 * BMAX + j - beg] = e2n;
-                }
-                if (hf > lmax) { lmax = hf; lleft = j; lright = j; }
-                else if (hf == lmax) { lright = j; }
-            }
-#ifdef ABAMD_KPROF
-            {
-                unsigned long long kc3 = __builtin_readcyclecounter();
-                kp_g += kc1 - kc_prev; /* gather + query since chunk entry */
-                kp_s += kc2 - kc1;     /* shfl + F scans + carries */
-                kp_w += kc3 - kc2;     /* H/E fold + stores + local track */
-                kc_prev = kc3;
-            }
-#endif
-        }
-
-        KPROF_T(kt2);
-        if (cache_fits) { prev_ok = 1; prev_row = r; prev_beg = beg; prev_end = end; }
-        else prev_ok = 0;
-        buf_cur ^= 1;
-
-        /* row argmax reduce + adaptive band push (simd_abpoa_max_in_row /
-         * ada_max_i); local/extend also track the running best here */
-        if (jb.banded || local_mode || extend_mode) {
-            int mv = wave_red_max_i32((int)lmax);
-            int ll = ((int)lmax == mv && lleft >= 0) ? lleft : 0x7fffffff;
-            int rr = ((int)lmax == mv && lright >= 0) ? lright : -1;
-            ll = wave_red_min_i32(ll);
-            rr = wave_red_max_i32(rr);
-            if (local_mode) {
-                if (mv > run_best) { run_best = mv; run_best_i = r; run_best_j = ll; }
-            } else if (extend_mode) {
-                if (mv > run_best) {
-                    run_best = mv; run_best_i = r; run_best_j = rr;
-                    run_best_remain = row_remain;
-                } else if (jb.zdrop > 0) {
-                    int delta = run_best_remain - row_remain;
-                    int dd = delta - (rr - run_best_j); if (dd < 0) dd = -dd;
-                    if (run_best - mv > jb.zdrop + jb.e1 * dd) zdropped = 1;
-                }
-            }
-            push_ml = 0x7fffffff; push_mr = -0x7fffffff;
-            if (!zdropped && jb.banded) {
-                /* serial scan (mean out-degree ~1.05): the common edge
-                 * r -> r+1 updates the register push; skip-edges go to
-                 * memory (their readers prefetch strictly later) */
-                for (int k = oo0; k < oo1; ++k) {
-                    int o = jb.out_idx[k];
-                    if (o == r + 1) {
-                        if (rr + 1 > push_mr) push_mr = rr + 1;
-                        if (ll + 1 < push_ml) push_ml = ll + 1;
-                    } else if (lane == 0) {
-                        if (rr + 1 > jb.max_right[o]) jb.max_right[o] = rr + 1;
-                        if (ll + 1 < jb.max_left[o]) jb.max_left[o] = ll + 1;
-                    }
-                }
-            }
-            if (zdropped) break;
-        } else {
-            push_ml = 0x7fffffff; push_mr = -0x7fffffff;
-        }
-#ifdef ABAMD_KPROF
-        {
-            unsigned long long kt3 = __builtin_readcyclecounter();
-            kp_band += kt1 - kt0; kp_chunk += kt2 - kt1; kp_epi += kt3 - kt2;
-            kp_rows += 1; kp_chunks += (unsigned long long)((end - beg) / WAVE + 1);
-        }
-#endif
-    }
-#ifdef ABAMD_KPROF
-    if (lane == 0) {
-        atomicAdd(&abamd_kprof_acc[0], kp_rows);
-        atomicAdd(&abamd_kprof_acc[1], kp_chunks);
-        atomicAdd(&abamd_kprof_acc[2], kp_band);
-        atomicAdd(&abamd_kprof_acc[3], kp_chunk);
-        atomicAdd(&abamd_kprof_acc[4], kp_epi);
-        atomicAdd(&abamd_kprof_acc[7], 1ull);
-        atomicAdd(&abamd_kprof_acc[8], kp_g);
-        atomicAdd(&abamd_kprof_acc[9], kp_s);
-        atomicAdd(&abamd_kprof_acc[10], kp_w);
-    }
-#endif
-    KPROF_T(kt_bt0);
-
-    if (lane == 0) res->cells = used;
-
-    /* ---- final best + backtrack (lane 0) ---- */
-    if (lane != 0) return;
-
-    int32_t best_score = run_best;
-    int best_i = run_best_i, best_j = run_best_j;
-    if (jb.align_mode == 0) { /* global: max over the end row's predecessors */
-        best_score = jb.inf_min; best_i = 0; best_j = 0;
-        for (int k = jb.pre_off[n_rows - 1]; k < jb.pre_off[n_rows]; ++k) {
-            const int p = jb.pre_idx[k];
-            const abamd_row_meta_t pm = meta[p];
-            int e = pm.end < qlen ? pm.end : qlen;
-            const S *pH = arena + pm.off * 5;
-            int32_t sc = (e >= pm.beg) ? (int32_t)pH[e - pm.beg] : jb.inf_min;
-            if (sc > best_score) { best_score = sc; best_i = p; best_j = e; }
-        }
-    }
-    res->best_score = best_score;
-    res->best_i = best_i; res->best_j = best_j;
-    if (!jb.ret_cigar) return;
-
-    /* backtrack: transcription of simd_abpoa_cg_backtrack (:309-458) */
-    {
-        int bi = best_i, bj = best_j, start_i = best_i, start_j = best_j;
-        int cur_op = 0x1f, n_c = 0, status = ABAMD_JOB_OK;
-        int look_end = jb.put_gap_at_end, put_right = jb.put_gap_on_right;
-        int n_aln = 0, n_matched = 0;
-        uint64_t *cig = jb.cigar;
-        int id = jb.row_node_id[bi];
-        if (best_j < qlen) dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, qlen - best_j, -1, qlen - 1, &status);
-        while (bi > 0 && bj > 0 && status == ABAMD_JOB_OK) {
-            const abamd_row_meta_t bm = meta[bi];
-            const int rb = bm.beg, re = bm.end;
-            const int64_t bw = re - rb + 1;
-            const S *H = arena + bm.off * 5;
-            const S *E1r = H + bw, *E2r = E1r + bw, *F1r = E2r + bw, *F2r = F1r + bw;
-            const S Hj = (bj >= rb && bj <= re) ? H[bj - rb] : inf_min;
-            const S Hjm1 = (bj - 1 >= rb && bj - 1 <= re) ? H[bj - 1 - rb] : inf_min;
-            const S E1j = (bj >= rb && bj <= re) ? E1r[bj - rb] : inf_min;
-            const S E2j = (bj >= rb && bj <= re) ? E2r[bj - rb] : inf_min;
-            const S F1j = (bj >= rb && bj <= re) ? F1r[bj - rb] : inf_min;
-            const S F2j = (bj >= rb && bj <= re) ? F2r[bj - rb] : inf_min;
-            const S F1jm1 = (bj - 1 >= rb && bj - 1 <= re) ? F1r[bj - 1 - rb] : inf_min;
-            const S F2jm1 = (bj - 1 >= rb && bj - 1 <= re) ? F2r[bj - 1 - rb] : inf_min;
-            if (local_mode && Hj == 0) break;
-            start_i = bi; start_j = bj;
-            const int pq0 = jb.pre_off[bi], pq1 = jb.pre_off[bi + 1];
-            const S s = (S)mat_lds[m * jb.row_base[bi] + query[bj - 1]];
-            const int is_match = jb.row_base[bi] == query[bj - 1];
-            int hit = 0;
-            if (put_right == 0 && look_end == 0 && (cur_op & 0x1)) {
-                for (int k = pq0; k < pq1; ++k) {
-                    const int p = jb.pre_idx[k];
-                    const S ps = (S)jb.pre_ps[k];
-                    const abamd_row_meta_t pm = meta[p];
-                    if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
-                    const S *pH = arena + pm.off * 5;
-                    if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
-                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
-                        bi = p; --bj; id = jb.row_node_id[bi]; hit = 1;
-                        cur_op = 0x1f; ++n_aln; n_matched += is_match;
-                        break;
-                    }
-                }
-            }
-            if (!hit && (cur_op & 0x6)) { /* deletion */
-                for (int k = pq0; k < pq1; ++k) {
-                    const int p = jb.pre_idx[k];
-                    const S ps = (S)jb.pre_ps[k];
-                    const abamd_row_meta_t pm = meta[p];
-                    if (bj < pm.beg || bj > pm.end) continue;
-                    const int poffc = bj - pm.beg;
-                    const int64_t pbw = pm.end - pm.beg + 1;
-                    const S *pH = arena + pm.off * 5;
-                    const S *pE1 = pH + pbw, *pE2 = pE1 + pbw;
-                    if (cur_op & 0x2) {
-                        if (cur_op & 0x1) {
-                            if (Hj == (S)(pE1[poffc] + ps)) {
-                                cur_op = ((S)(pH[poffc] - oe1) == pE1[poffc]) ? (0x1 | 0x18) : 0x2;
-                                hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
-                                bi = p; id = jb.row_node_id[bi];
-                                if (look_end) look_end = 0;
-                                break;
-                            }
-                        } else {
-                            if (E1j == (S)(pE1[poffc] - e1 + ps)) {
-                                cur_op = ((S)(pH[poffc] - oe1) == pE1[poffc]) ? (0x1 | 0x18) : 0x2;
-                                hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
-                                bi = p; id = jb.row_node_id[bi];
-                                if (look_end) look_end = 0;
-                                break;
-                            }
-                        }
-                    }
-                    if (cur_op & 0x4) {
-                        if (cur_op & 0x1) {
-                            if (Hj == (S)(pE2[poffc] + ps)) {
-                                cur_op = ((S)(pH[poffc] - oe2) == pE2[poffc]) ? (0x1 | 0x18) : 0x4;
-                                hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
-                                bi = p; id = jb.row_node_id[bi];
-                                if (look_end) look_end = 0;
-                                break;
-                            }
-                        } else {
-                            if (E2j == (S)(pE2[poffc] - e2 + ps)) {
-                                cur_op = ((S)(pH[poffc] - oe2) == pE2[poffc]) ? (0x1 | 0x18) : 0x4;
-                                hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
-                                bi = p; id = jb.row_node_id[bi];
-                                if (look_end) look_end = 0;
-                                break;
-                            }
-                        }
-                    }
-                }
-            }
-            if (!hit && (cur_op & 0x18)) { /* insertion */
-                if (cur_op & 0x8) {
-                    if (cur_op & 0x1) {
-                        if (Hj == F1j) {
-                            if ((S)(Hjm1 - oe1) == F1j) { cur_op = 0x1 | 0x6; hit = 1; }
-                            else if ((S)(F1jm1 - e1) == F1j) { cur_op = 0x8; hit = 1; }
-                        }
-                    } else {
-                        if ((S)(Hjm1 - oe1) == F1j) { cur_op = 0x1 | 0x6; hit = 1; }
-                        else if ((S)(F1jm1 - e1) == F1j) { cur_op = 0x8; hit = 1; }
-                    }
-                }
-                if (!hit && (cur_op & 0x10)) {
-                    if (cur_op & 0x1) {
-                        if (Hj == F2j) {
-                            if ((S)(Hjm1 - oe2) == F2j) { cur_op = 0x1 | 0x6; hit = 1; }
-                            else if ((S)(F2jm1 - e2) == F2j) { cur_op = 0x10; hit = 1; }
-                        }
-                    } else {
-                        if ((S)(Hjm1 - oe2) == F2j) { cur_op = 0x1 | 0x6; hit = 1; }
-                        else if ((S)(F2jm1 - e2) == F2j) { cur_op = 0x10; hit = 1; }
-                    }
-                }
-                if (hit) {
-                    dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, 1, id, bj - 1, &status);
-                    --bj;
-                    if (look_end) look_end = 0;
-                    ++n_aln;
-                }
-            }
-            if (!hit && (cur_op & 0x1)) {
-                for (int k = pq0; k < pq1; ++k) {
-                    const int p = jb.pre_idx[k];
-                    const S ps = (S)jb.pre_ps[k];
-                    const abamd_row_meta_t pm = meta[p];
-                    if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
-                    const S *pH = arena + pm.off * 5;
-                    if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
-                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
-                        bi = p; --bj; id = jb.row_node_id[bi]; hit = 1;
-                        cur_op = 0x1f; ++n_aln; n_matched += is_match;
-                        look_end = 0;
-                        break;
-                    }
-                }
-            }
-            if (!hit) { status = ABAMD_JOB_BT_DEAD_END; break; }
-        }
-        if (status == ABAMD_JOB_OK && bj > 0)
-            dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, bj, -1, bj - 1, &status);
-        res->status = status;
-        res->n_cigar = n_c;
-        res->n_aln_bases = n_aln;
-        res->n_matched_bases = n_matched;
-        res->node_e = jb.row_node_id[best_i]; res->query_e = best_j - 1;
-        res->node_s = jb.row_node_id[start_i]; res->query_s = start_j - 1;
-    }
-#ifdef ABAMD_KPROF
-    atomicAdd(&abamd_kprof_acc[5], __builtin_readcyclecounter() - kt_bt0);
-#endif
-}
-
-/* ------------------------------------------------------------------ */
 /* Multi-wave convex kernel: ONE JOB PER 512-THREAD BLOCK.              */
 /*                                                                      */
 /* The one-wave kernel above is instruction-ISSUE bound: a lone wave    */
@@ -786,21 +220,21 @@ void cg_global_mw_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         if (tid == 0) { meta[0].beg = 0; meta[0].end = end0; meta[0].off = 0; }
         int64_t bw = end0 + 1;
         used = bw;
-        S *H = arena, *E1 = arena + bw, *E2 = arena + 2 * bw, *F1 = arena + 3 * bw, *F2 = arena + 4 * bw;
+        /* 3 planes only (H,E1,E2): the F planes are recomputed from stored H
+         * at backtrack time (validated: tools/f_recompute_experiment.py — a
+         * 40% cut of plane bytes and arena capacity) */
+        S *H = arena, *E1 = arena + bw, *E2 = arena + 2 * bw;
         S *c = &prev_lds[buf_cur][0];
         const int fits = end0 + 1 <= BMW;
         for (int j = tid; j <= end0; j += MWT) {
             S hv, e1v2, e2v2;
             if (local_mode) {
                 hv = 0; e1v2 = 0; e2v2 = 0;
-                F1[j] = 0; F2[j] = 0;
             } else if (j == 0) {
                 hv = 0; e1v2 = (S)(0 - oe1); e2v2 = (S)(0 - oe2);
-                F1[0] = inf_min; F2[0] = inf_min;
             } else {
                 S f1 = (S)(-(jb.o1 + jb.e1 * j));
                 S f2 = (S)(-(jb.o2 + jb.e2 * j));
-                F1[j] = f1; F2[j] = f2;
                 hv = smax(f1, f2);
                 e1v2 = inf_min; e2v2 = inf_min;
             }
@@ -880,7 +314,7 @@ void cg_global_mw_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         const int64_t off = used;
         if (tid == 0) { meta[r].beg = beg; meta[r].end = end; meta[r].off = off; }
         used += bw;
-        S *H = arena + off * 5, *E1r = H + bw, *E2r = E1r + bw, *F1r = E2r + bw, *F2r = F1r + bw;
+        S *H = arena + off * 3, *E1r = H + bw, *E2r = E1r + bw;
         const uint8_t base = jb.row_base[r];
         const int *mrow = &mat_lds[base * m];
         const int cache_fits = bw <= BMW;
@@ -930,7 +364,7 @@ void cg_global_mw_kernel(const abamd_gpu_job_t *__restrict__ jobs,
                 }
                 const abamd_row_meta_t pm = meta[p];
                 const int64_t pbw = pm.end - pm.beg + 1;
-                const S *__restrict__ pH = arena + pm.off * 5;
+                const S *__restrict__ pH = arena + pm.off * 3;
                 const S *__restrict__ pE1 = pH + pbw;
                 const S *__restrict__ pE2 = pE1 + pbw;
                 if (act) {
@@ -999,7 +433,6 @@ void cg_global_mw_kernel(const abamd_gpu_job_t *__restrict__ jobs,
             if (local_mode) { e1n = smax(e1n, (S)0); e2n = smax(e2n, (S)0); }
             if (act) {
                 H[j - beg] = hf; E1r[j - beg] = e1n; E2r[j - beg] = e2n;
-                F1r[j - beg] = f1; F2r[j - beg] = f2;
                 if (cache_fits) {
                     cw[j - beg] = hf;
                     cw[BMW + j - beg] = e1n;
@@ -1104,9 +537,9 @@ void cg_global_mw_kernel(const abamd_gpu_job_t *__restrict__ jobs,
     __syncthreads();
     if (tid == 0) res->cells = used;
 
-    /* ---- final best + backtrack (thread 0; simd_abpoa_cg_backtrack) ---- */
-    if (tid != 0) return;
-
+    /* ---- final best (computed redundantly by every thread: all inputs are
+     * uniform, and every thread must stay alive for the cooperative
+     * F-recompute below) ---- */
     int32_t best_score = run_best;
     int best_i = run_best_i, best_j = run_best_j;
     if (jb.align_mode == 0) {
@@ -1115,191 +548,266 @@ void cg_global_mw_kernel(const abamd_gpu_job_t *__restrict__ jobs,
             const int p = jb.pre_idx[k];
             const abamd_row_meta_t pm = meta[p];
             int e = pm.end < qlen ? pm.end : qlen;
-            const S *pH = arena + pm.off * 5;
+            const S *pH = arena + pm.off * 3;
             int32_t sc = (e >= pm.beg) ? (int32_t)pH[e - pm.beg] : jb.inf_min;
             if (sc > best_score) { best_score = sc; best_i = p; best_j = e; }
         }
     }
-    res->best_score = best_score;
-    res->best_i = best_i; res->best_j = best_j;
+    if (tid == 0) {
+        res->best_score = best_score;
+        res->best_i = best_i; res->best_j = best_j;
+    }
     if (!jb.ret_cigar) return;
-    {
-        int bi = best_i, bj = best_j, start_i = best_i, start_j = best_j;
-        int cur_op = 0x1f, n_c = 0, status = ABAMD_JOB_OK;
-        int look_end = jb.put_gap_at_end, put_right = jb.put_gap_on_right;
-        int n_aln = 0, n_matched = 0;
-        uint64_t *cig = jb.cigar;
-        int id = jb.row_node_id[bi];
+
+    /* ---- backtrack (simd_abpoa_cg_backtrack transcription) ----
+     * Thread 0 walks; the F planes are not stored, so when the insertion
+     * branch needs F values the walk pauses and the whole block recomputes
+     * the row's F window from the STORED (post-F) H plane:
+     *   F'[rb] = inf_min - oe;  F'[j] = max(F'[j-1] - e, H[j-1] - oe)
+     * — every backtrack decision is identical to stored-F (validated over
+     * ~2.5M entries incl. entry-direction flips, tools/f_recompute_experiment.py,
+     * and end-to-end by the parity suite). The window scan reuses the
+     * per-wave local-scan + cross-wave-carry machinery of the forward pass
+     * (same associativity/headroom argument). */
+    __shared__ int bt_go, bt_bi_s, bt_bj_s;
+    __shared__ S fbuf[2 * BMW];
+    /* walk state: only thread 0's copies advance */
+    int bi = best_i, bj = best_j, start_i = best_i, start_j = best_j;
+    int cur_op = 0x1f, n_c = 0, status = ABAMD_JOB_OK;
+    int look_end = jb.put_gap_at_end, put_right = jb.put_gap_on_right;
+    int n_aln = 0, n_matched = 0;
+    uint64_t *cig = jb.cigar;
+    int id = jb.row_node_id[bi];
+    int f_row = -1, f_wl = 0, f_hi = -1;
+    if (tid == 0) {
+        bt_go = 1;
         if (best_j < qlen) dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, qlen - best_j, -1, qlen - 1, &status);
-        while (bi > 0 && bj > 0 && status == ABAMD_JOB_OK) {
-            const abamd_row_meta_t bm = meta[bi];
+    }
+    for (;;) {
+        __syncthreads();
+        if (bt_go == 0) break;
+        if (bt_go == 2) {
+            /* recompute F for row bt_bi_s, window ending at bt_bj_s */
+            const int rbi = bt_bi_s;
+            const abamd_row_meta_t bm = meta[rbi];
             const int rb = bm.beg, re = bm.end;
-            const int64_t bw = re - rb + 1;
-            const S *H = arena + bm.off * 5;
-            const S *E1r = H + bw, *E2r = E1r + bw, *F1r = E2r + bw, *F2r = F1r + bw;
-            const S Hj = (bj >= rb && bj <= re) ? H[bj - rb] : inf_min;
-            const S Hjm1 = (bj - 1 >= rb && bj - 1 <= re) ? H[bj - 1 - rb] : inf_min;
-            const S E1j = (bj >= rb && bj <= re) ? E1r[bj - rb] : inf_min;
-            const S E2j = (bj >= rb && bj <= re) ? E2r[bj - rb] : inf_min;
-            const S F1j = (bj >= rb && bj <= re) ? F1r[bj - rb] : inf_min;
-            const S F2j = (bj >= rb && bj <= re) ? F2r[bj - rb] : inf_min;
-            const S F1jm1 = (bj - 1 >= rb && bj - 1 <= re) ? F1r[bj - 1 - rb] : inf_min;
-            const S F2jm1 = (bj - 1 >= rb && bj - 1 <= re) ? F2r[bj - 1 - rb] : inf_min;
-            if (local_mode && Hj == 0) break;
-            start_i = bi; start_j = bj;
-            const int pq0 = jb.pre_off[bi], pq1 = jb.pre_off[bi + 1];
-            const S s = (S)mat_lds[m * jb.row_base[bi] + query[bj - 1]];
-            const int is_match = jb.row_base[bi] == query[bj - 1];
-            int hit = 0;
-            if (put_right == 0 && look_end == 0 && (cur_op & 0x1)) {
-                for (int k = pq0; k < pq1; ++k) {
-                    const int p = jb.pre_idx[k];
-                    const S ps = (S)jb.pre_ps[k];
-                    const abamd_row_meta_t pm = meta[p];
-                    if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
-                    const S *pH = arena + pm.off * 5;
-                    if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
-                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
-                        bi = p; --bj; id = jb.row_node_id[bi]; hit = 1;
-                        cur_op = 0x1f; ++n_aln; n_matched += is_match;
-                        break;
+            const int hi = bt_bj_s < re ? bt_bj_s : re;
+            const int wl = (hi - (BMW - 2)) > rb ? hi - (BMW - 2) : rb;
+            const S *Hrow = arena + bm.off * 3;
+            S carry1 = inf_min, carry2 = inf_min; /* F' at position ss-1 */
+            for (int ss = rb; ss <= hi; ss += MWT) {
+                const int j = ss + tid;
+                const bool act2 = j <= hi;
+                S c1, c2;
+                if (!act2) { c1 = inf_min; c2 = inf_min; }
+                else if (j == rb) { c1 = (S)(inf_min - oe1); c2 = (S)(inf_min - oe2); }
+                else {
+                    const S hp = Hrow[j - 1 - rb];
+                    c1 = (S)(hp - oe1);
+                    c2 = (S)(hp - oe2);
+                }
+                S L1 = scan_maxplus(c1, jb.e1, inf_min, lane);
+                S L2 = scan_maxplus(c2, jb.e2, inf_min, lane);
+                if (lane == WAVE - 1) { sc_f1[wv] = L1; sc_f2[wv] = L2; }
+                __syncthreads();
+                /* F' entering this wave; no carry exists before the band
+                 * start (mirrors the forward pass: wave 0 of the first
+                 * superchunk takes its local scan verbatim) */
+                S C1 = carry1, C2 = carry2;
+                bool havec = ss != rb;
+                for (int ww = 0; ww < wv; ++ww) {
+                    if (havec) {
+                        C1 = smax(sc_f1[ww], (S)(C1 - (S)(WAVE * jb.e1)));
+                        C2 = smax(sc_f2[ww], (S)(C2 - (S)(WAVE * jb.e2)));
+                    } else {
+                        C1 = sc_f1[ww]; C2 = sc_f2[ww];
+                        havec = true;
                     }
                 }
+                S f1 = L1, f2 = L2;
+                if (havec) {
+                    f1 = smax(L1, (S)(C1 - (S)((lane + 1) * jb.e1)));
+                    f2 = smax(L2, (S)(C2 - (S)((lane + 1) * jb.e2)));
+                }
+                if (act2 && j >= wl) {
+                    fbuf[j - wl] = f1;
+                    fbuf[BMW + j - wl] = f2;
+                }
+                const bool more2 = ss + MWT <= hi;
+                if (more2 && j == ss + MWT - 1) { sc_carry[0] = f1; sc_carry[1] = f2; }
+                __syncthreads();
+                if (more2) { carry1 = sc_carry[0]; carry2 = sc_carry[1]; }
             }
-            if (!hit && (cur_op & 0x6)) { /* deletion */
-                for (int k = pq0; k < pq1; ++k) {
-                    const int p = jb.pre_idx[k];
-                    const S ps = (S)jb.pre_ps[k];
-                    const abamd_row_meta_t pm = meta[p];
-                    if (bj < pm.beg || bj > pm.end) continue;
-                    const int poffc = bj - pm.beg;
-                    const int64_t pbw = pm.end - pm.beg + 1;
-                    const S *pH = arena + pm.off * 5;
-                    const S *pE1 = pH + pbw, *pE2 = pE1 + pbw;
-                    if (cur_op & 0x2) {
-                        if (cur_op & 0x1) {
-                            if (Hj == (S)(pE1[poffc] + ps)) {
-                                cur_op = ((S)(pH[poffc] - oe1) == pE1[poffc]) ? (0x1 | 0x18) : 0x2;
-                                hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
-                                bi = p; id = jb.row_node_id[bi];
-                                if (look_end) look_end = 0;
-                                break;
+            f_row = rbi; f_wl = wl; f_hi = hi;
+            __syncthreads();
+        }
+        if (tid == 0) {
+            int pause = 0;
+            while (bi > 0 && bj > 0 && status == ABAMD_JOB_OK) {
+                const abamd_row_meta_t bm = meta[bi];
+                const int rb = bm.beg, re = bm.end;
+                const int64_t bw = re - rb + 1;
+                const S *H = arena + bm.off * 3;
+                const S *E1r = H + bw, *E2r = E1r + bw;
+                const S Hj = (bj >= rb && bj <= re) ? H[bj - rb] : inf_min;
+                const S Hjm1 = (bj - 1 >= rb && bj - 1 <= re) ? H[bj - 1 - rb] : inf_min;
+                const S E1j = (bj >= rb && bj <= re) ? E1r[bj - rb] : inf_min;
+                const S E2j = (bj >= rb && bj <= re) ? E2r[bj - rb] : inf_min;
+                if (local_mode && Hj == 0) break;
+                start_i = bi; start_j = bj;
+                const int pq0 = jb.pre_off[bi], pq1 = jb.pre_off[bi + 1];
+                const S s = (S)mat_lds[m * jb.row_base[bi] + query[bj - 1]];
+                const int is_match = jb.row_base[bi] == query[bj - 1];
+                int hit = 0;
+                if (put_right == 0 && look_end == 0 && (cur_op & 0x1)) {
+                    for (int k = pq0; k < pq1; ++k) {
+                        const int p = jb.pre_idx[k];
+                        const S ps = (S)jb.pre_ps[k];
+                        const abamd_row_meta_t pm = meta[p];
+                        if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
+                        const S *pH = arena + pm.off * 3;
+                        if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
+                            dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
+                            bi = p; --bj; id = jb.row_node_id[bi]; hit = 1;
+                            cur_op = 0x1f; ++n_aln; n_matched += is_match;
+                            break;
+                        }
+                    }
+                }
+                if (!hit && (cur_op & 0x6)) { /* deletion */
+                    for (int k = pq0; k < pq1; ++k) {
+                        const int p = jb.pre_idx[k];
+                        const S ps = (S)jb.pre_ps[k];
+                        const abamd_row_meta_t pm = meta[p];
+                        if (bj < pm.beg || bj > pm.end) continue;
+                        const int poffc = bj - pm.beg;
+                        const int64_t pbw = pm.end - pm.beg + 1;
+                        const S *pH = arena + pm.off * 3;
+                        const S *pE1 = pH + pbw, *pE2 = pE1 + pbw;
+                        if (cur_op & 0x2) {
+                            if (cur_op & 0x1) {
+                                if (Hj == (S)(pE1[poffc] + ps)) {
+                                    cur_op = ((S)(pH[poffc] - oe1) == pE1[poffc]) ? (0x1 | 0x18) : 0x2;
+                                    hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                                    bi = p; id = jb.row_node_id[bi];
+                                    if (look_end) look_end = 0;
+                                    break;
+                                }
+                            } else {
+                                if (E1j == (S)(pE1[poffc] - e1 + ps)) {
+                                    cur_op = ((S)(pH[poffc] - oe1) == pE1[poffc]) ? (0x1 | 0x18) : 0x2;
+                                    hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                                    bi = p; id = jb.row_node_id[bi];
+                                    if (look_end) look_end = 0;
+                                    break;
+                                }
                             }
-                        } else {
-                            if (E1j == (S)(pE1[poffc] - e1 + ps)) {
-                                cur_op = ((S)(pH[poffc] - oe1) == pE1[poffc]) ? (0x1 | 0x18) : 0x2;
-                                hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
-                                bi = p; id = jb.row_node_id[bi];
-                                if (look_end) look_end = 0;
-                                break;
+                        }
+                        if (cur_op & 0x4) {
+                            if (cur_op & 0x1) {
+                                if (Hj == (S)(pE2[poffc] + ps)) {
+                                    cur_op = ((S)(pH[poffc] - oe2) == pE2[poffc]) ? (0x1 | 0x18) : 0x4;
+                                    hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                                    bi = p; id = jb.row_node_id[bi];
+                                    if (look_end) look_end = 0;
+                                    break;
+                                }
+                            } else {
+                                if (E2j == (S)(pE2[poffc] - e2 + ps)) {
+                                    cur_op = ((S)(pH[poffc] - oe2) == pE2[poffc]) ? (0x1 | 0x18) : 0x4;
+                                    hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                                    bi = p; id = jb.row_node_id[bi];
+                                    if (look_end) look_end = 0;
+                                    break;
+                                }
                             }
                         }
                     }
-                    if (cur_op & 0x4) {
+                }
+                if (!hit && (cur_op & 0x18)) { /* insertion: needs recomputed F */
+                    const bool needj = (bj >= rb && bj <= re);
+                    const bool needj1 = (bj - 1 >= rb && bj - 1 <= re);
+                    const bool cover = (f_row == bi)
+                        && (!needj || (bj >= f_wl && bj <= f_hi))
+                        && (!needj1 || (bj - 1 >= f_wl && bj - 1 <= f_hi));
+                    if ((needj || needj1) && !cover) {
+                        bt_bi_s = bi; bt_bj_s = bj; bt_go = 2;
+                        pause = 1;
+                        break; /* re-enter this step after the block recomputes */
+                    }
+                    const S F1j = needj ? fbuf[bj - f_wl] : inf_min;
+                    const S F2j = needj ? fbuf[BMW + bj - f_wl] : inf_min;
+                    const S F1jm1 = needj1 ? fbuf[bj - 1 - f_wl] : inf_min;
+                    const S F2jm1 = needj1 ? fbuf[BMW + bj - 1 - f_wl] : inf_min;
+                    if (cur_op & 0x8) {
                         if (cur_op & 0x1) {
-                            if (Hj == (S)(pE2[poffc] + ps)) {
-                                cur_op = ((S)(pH[poffc] - oe2) == pE2[poffc]) ? (0x1 | 0x18) : 0x4;
-                                hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
-                                bi = p; id = jb.row_node_id[bi];
-                                if (look_end) look_end = 0;
-                                break;
+                            if (Hj == F1j) {
+                                if ((S)(Hjm1 - oe1) == F1j) { cur_op = 0x1 | 0x6; hit = 1; }
+                                else if ((S)(F1jm1 - e1) == F1j) { cur_op = 0x8; hit = 1; }
                             }
                         } else {
-                            if (E2j == (S)(pE2[poffc] - e2 + ps)) {
-                                cur_op = ((S)(pH[poffc] - oe2) == pE2[poffc]) ? (0x1 | 0x18) : 0x4;
-                                hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
-                                bi = p; id = jb.row_node_id[bi];
-                                if (look_end) look_end = 0;
-                                break;
-                            }
-                        }
-                    }
-                }
-            }
-            if (!hit && (cur_op & 0x18)) { /* insertion */
-                if (cur_op & 0x8) {
-                    if (cur_op & 0x1) {
-                        if (Hj == F1j) {
                             if ((S)(Hjm1 - oe1) == F1j) { cur_op = 0x1 | 0x6; hit = 1; }
                             else if ((S)(F1jm1 - e1) == F1j) { cur_op = 0x8; hit = 1; }
                         }
-                    } else {
-                        if ((S)(Hjm1 - oe1) == F1j) { cur_op = 0x1 | 0x6; hit = 1; }
-                        else if ((S)(F1jm1 - e1) == F1j) { cur_op = 0x8; hit = 1; }
                     }
-                }
-                if (!hit && (cur_op & 0x10)) {
-                    if (cur_op & 0x1) {
-                        if (Hj == F2j) {
+                    if (!hit && (cur_op & 0x10)) {
+                        if (cur_op & 0x1) {
+                            if (Hj == F2j) {
+                                if ((S)(Hjm1 - oe2) == F2j) { cur_op = 0x1 | 0x6; hit = 1; }
+                                else if ((S)(F2jm1 - e2) == F2j) { cur_op = 0x10; hit = 1; }
+                            }
+                        } else {
                             if ((S)(Hjm1 - oe2) == F2j) { cur_op = 0x1 | 0x6; hit = 1; }
                             else if ((S)(F2jm1 - e2) == F2j) { cur_op = 0x10; hit = 1; }
                         }
-                    } else {
-                        if ((S)(Hjm1 - oe2) == F2j) { cur_op = 0x1 | 0x6; hit = 1; }
-                        else if ((S)(F2jm1 - e2) == F2j) { cur_op = 0x10; hit = 1; }
+                    }
+                    if (hit) {
+                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, 1, id, bj - 1, &status);
+                        --bj;
+                        if (look_end) look_end = 0;
+                        ++n_aln;
                     }
                 }
-                if (hit) {
-                    dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, 1, id, bj - 1, &status);
-                    --bj;
-                    if (look_end) look_end = 0;
-                    ++n_aln;
-                }
-            }
-            if (!hit && (cur_op & 0x1)) {
-                for (int k = pq0; k < pq1; ++k) {
-                    const int p = jb.pre_idx[k];
-                    const S ps = (S)jb.pre_ps[k];
-                    const abamd_row_meta_t pm = meta[p];
-                    if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
-                    const S *pH = arena + pm.off * 5;
-                    if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
-                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
-                        bi = p; --bj; id = jb.row_node_id[bi]; hit = 1;
-                        cur_op = 0x1f; ++n_aln; n_matched += is_match;
-                        look_end = 0;
-                        break;
+                if (!hit && (cur_op & 0x1)) {
+                    for (int k = pq0; k < pq1; ++k) {
+                        const int p = jb.pre_idx[k];
+                        const S ps = (S)jb.pre_ps[k];
+                        const abamd_row_meta_t pm = meta[p];
+                        if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
+                        const S *pH = arena + pm.off * 3;
+                        if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
+                            dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
+                            bi = p; --bj; id = jb.row_node_id[bi]; hit = 1;
+                            cur_op = 0x1f; ++n_aln; n_matched += is_match;
+                            look_end = 0;
+                            break;
+                        }
                     }
                 }
+                if (!hit) { status = ABAMD_JOB_BT_DEAD_END; break; }
             }
-            if (!hit) { status = ABAMD_JOB_BT_DEAD_END; break; }
+            if (!pause) {
+                if (status == ABAMD_JOB_OK && bj > 0)
+                    dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, bj, -1, bj - 1, &status);
+                res->status = status;
+                res->n_cigar = n_c;
+                res->n_aln_bases = n_aln;
+                res->n_matched_bases = n_matched;
+                res->node_e = jb.row_node_id[best_i]; res->query_e = best_j - 1;
+                res->node_s = jb.row_node_id[start_i]; res->query_s = start_j - 1;
+                bt_go = 0;
+            }
         }
-        if (status == ABAMD_JOB_OK && bj > 0)
-            dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, bj, -1, bj - 1, &status);
-        res->status = status;
-        res->n_cigar = n_c;
-        res->n_aln_bases = n_aln;
-        res->n_matched_bases = n_matched;
-        res->node_e = jb.row_node_id[best_i]; res->query_e = best_j - 1;
-        res->node_s = jb.row_node_id[start_i]; res->query_s = start_j - 1;
     }
-}
-
-static int use_single_wave_kernel(void) {
-    static int v = -1;
-    if (v < 0) v = getenv("ABPOA_AMD_SW_KERNEL") != nullptr;
-    return v;
 }
 
 extern "C" void abamd_launch_cg_i16(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
                                     int n_jobs, void *stream) {
-    if (use_single_wave_kernel()) {
-        int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
-        hipLaunchKernelGGL((cg_global_kernel<int16_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
-                           (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
-        return;
-    }
     hipLaunchKernelGGL((cg_global_mw_kernel<int16_t>), dim3(n_jobs), dim3(MWT), 0,
                        (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
 }
 extern "C" void abamd_launch_cg_i32(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
                                     int n_jobs, void *stream) {
-    if (use_single_wave_kernel()) {
-        int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
-        hipLaunchKernelGGL((cg_global_kernel<int32_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
-                           (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
-        return;
-    }
     hipLaunchKernelGGL((cg_global_mw_kernel<int32_t>), dim3(n_jobs), dim3(MWT), 0,
                        (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
 }

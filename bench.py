@@ -28,7 +28,7 @@ sys.path.insert(0, ROOT)
 
 WORKLOAD = "1000 synthetic sets x 50 reads x 10 kbp, 10% ONT-style error, global convex gap, 1 MI355X (BASELINE.json configs[1])"
 HBM_PEAK_GBPS = 8000.0  # gfx950 spec peak (MI355X_MICROARCH.md)
-ALG_BYTES_PER_CELL = 10.0  # 5 planes x int16, written once (backtrack re-read counted separately)
+ALG_BYTES_PER_CELL = 6.0  # 3 planes (H,E1,E2) x int16, written once; F planes recomputed at backtrack
 
 
 def gen_sets(rng, n_sets, depth, qlen, p_sub=0.045, p_del=0.03, p_ins=0.025):
