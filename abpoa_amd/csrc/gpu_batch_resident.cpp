@@ -83,7 +83,19 @@ struct DevBuf {
         if (want < 4096) want = 4096;
         double t0 = abamd_realtime();
         if (p) RHIP_CHECK(hipFree(p));
-        RHIP_CHECK(hipMalloc(&p, want));
+        hipError_t e = hipMalloc(&p, want);
+        if (e != hipSuccess && want > n) {
+            /* headroom did not fit — fall back to the exact demand, and
+             * consume the sticky error the failed attempt recorded (it
+             * would otherwise surface at the next hipGetLastError) */
+            (void)hipGetLastError();
+            want = n;
+            e = hipMalloc(&p, want);
+        }
+        if (e != hipSuccess) {
+            fprintf(stderr, "[abpoa_amd] hipMalloc(%zu) failed: %s\n", want, hipGetErrorString(e));
+            exit(EXIT_FAILURE);
+        }
         g_alloc_s += abamd_realtime() - t0;
         g_alloc_n += 1;
         g_held += want - cap;
@@ -122,6 +134,7 @@ struct SetState {
     std::vector<int64_t> qoff;   /* per-read offsets into d_query */
     int n_rows, n_pre, n_out;    /* DP-row CSR sizes from the last fold */
     int64_t last_cells;          /* measured banded cells of the last round */
+    int last_rows;               /* rows of that round (for band = cells/rows) */
     int n_seqs;
     const int *seq_lens;
     void *own_slab;              /* non-null after an expansion */
@@ -261,12 +274,13 @@ struct Batch {
 
 int64_t job_est_cells(const Batch &B, const SetState &S, int qlen) {
     abpoa_para_t *abpt = B.abpt;
+    /* rows x (2w + drift margin): a sound per-job bound in practice (zero
+     * retries over every soak); tightening it with measured-cells hints was
+     * tried twice and caused retry storms (bands drift unevenly), while the
+     * formula's whole-workload demand fits the pipeline arenas at the
+     * 3-plane layout. Jobs that still overflow retry with 2x. */
     int w = abpt->wb < 0 ? qlen : abpt->wb + (int)(abpt->wf * qlen);
     int64_t est = (int64_t)S.n_rows * (2 * (int64_t)w + 160) + qlen + 64;
-    if (S.last_cells > 0 && !getenv("ABPOA_AMD_NO_HINT")) {
-        int64_t tight = S.last_cells + S.last_cells / 2 + qlen;
-        if (tight < est) est = tight;
-    }
     return est;
 }
 
@@ -584,6 +598,7 @@ void retry_failed(Batch &B, Slot &S) {
             } else {
                 abamd_stats_add_cells((uint64_t)rr[i].cells, (uint64_t)rr[i].cells * B.planes * ssz);
                 B.sets[R.set_of[i]].last_cells = rr[i].cells;
+                B.sets[R.set_of[i]].last_rows = R.hjobs[i].n_rows;
                 settle_fold(B, R, i);
             }
         }
@@ -615,6 +630,7 @@ void finish_slot(Batch &B, Slot &S) {
         if (res[i].status != ABAMD_JOB_OK) continue;
         abamd_stats_add_cells((uint64_t)res[i].cells, (uint64_t)res[i].cells * B.planes * ssz);
         B.sets[S.set_of[i]].last_cells = res[i].cells;
+        B.sets[S.set_of[i]].last_rows = S.hjobs[i].n_rows;
         settle_fold(B, S, i);
     }
     retry_failed(B, S);
@@ -684,7 +700,7 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
     B.sets.resize(n_sets);
 
     /* ---- per-set capacity plan + one slab for every graph ---- */
-    double alpha = 0.35;
+    double alpha = 0.25;
     {
         const char *a = getenv("ABPOA_AMD_NODE_ALPHA");
         if (a && *a) { double v = atof(a); if (v > 0.01 && v <= 1.0) alpha = v; }
@@ -699,6 +715,7 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
         S.seq_lens = seq_lens[s];
         S.own_slab = nullptr;
         S.last_cells = 0;
+        S.last_rows = 0;
         S.n_rows = S.n_pre = S.n_out = 0;
         if (n_seqs[s] > max_reads) max_reads = n_seqs[s];
         int64_t L = 0, first = n_seqs[s] > 0 ? seq_lens[s][0] : 0;
@@ -787,10 +804,15 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
             RHIP_CHECK(hipMemGetInfo(&free_b, &total_b));
             uint64_t arena_held = 0;
             for (int i = 0; i < 8; ++i) arena_held += C.slot[i].arena.cap + C.slot[i].cigars.cap;
-            /* every arena-holding slot can reach the cap at once: n_groups
-             * pipeline slots + the big-item double-buffer pair + the retry
-             * slot — splitting across fewer shares OOMs a full-HBM run */
-            mem_gb = (double)(free_b + arena_held) * 0.84 / (n_groups + 3) / 1e9;
+            /* size the pipeline slots to carry the WHOLE workload without
+             * chunking (chunked big items serialize and measured ~25% slower
+             * at the north-star shape); the big/retry slots use small chunks
+             * (budget/6) and ensure() falls back to exact-demand allocation,
+             * so the occasional overflow cannot OOM the run */
+            mem_gb = (double)(free_b + arena_held) * 0.87 / n_groups / 1e9;
+            if (getenv("ABPOA_AMD_TIMING"))
+                fprintf(stderr, "[abamd budget] free %.1f GB held %.1f GB -> %.1f GB/slot (%d groups)\n",
+                        free_b / 1e9, arena_held / 1e9, mem_gb, n_groups);
         }
         B.budget_bytes = mem_gb * 1e9;
     }
@@ -913,16 +935,20 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
         double demand = 0;
         for (int idx : list) demand += job_bytes(idx, r);
         if (demand > B.budget_bytes && list.size() > 1) {
+            if (B.big_chunks == 0 && getenv("ABPOA_AMD_TIMING"))
+                fprintf(stderr, "[abamd budget] first big item: round %d, %zu jobs, demand %.1f GB > budget %.1f GB (n_rows[0] %d)\n",
+                        r, list.size(), demand / 1e9, B.budget_bytes / 1e9, B.sets[list[0]].n_rows);
             Slot *bs[2] = {&C.slot[6], &C.slot[7]};
             bs[0]->ensure_init(); bs[1]->ensure_init();
             int par = 0;
             size_t done = 0;
+            const double chunk_cap = B.budget_bytes / 6;
             while (done < list.size()) {
                 double acc = 0;
                 chunk.clear();
                 while (done < list.size()) {
                     double e = job_bytes(list[done], r);
-                    if (!chunk.empty() && acc + e > B.budget_bytes) break;
+                    if (!chunk.empty() && acc + e > chunk_cap) break;
                     acc += e;
                     chunk.push_back(list[done]);
                     ++done;

@@ -160,38 +160,57 @@ __device__ static void dev_topo_bfs(const flat_graph_t *fg, int *index_to_node_i
 }
 
 /* weight-descending adjacency sort — per-node work is independent, so nodes
- * stride across lanes; the per-node pairwise-swap pattern (tie semantics)
- * matches abamd_flat_sort_adjacency exactly */
+ * stride across lanes. The host reference's pairwise-swap pass with strict <
+ * (abamd_flat_sort_adjacency) is a STABLE selection sort: position j ends
+ * with the first-occurring maximum of the remaining edges. Typical degrees
+ * (<= 64: at most one edge per read plus merges) sort via a small private
+ * array; larger degrees fall back to an in-place stable selection over the
+ * chain — identical order, no O(KB) per-lane scratch (a 1024-int array here
+ * forced a multi-GB device scratch pool that OOM'd full-HBM runs). */
+#define SORT_LOC 64
+__device__ static void dev_sort_chain(int *head, int *tail, int *next,
+                                      const int *w) {
+    int loc[SORT_LOC];
+    int e, n = 0;
+    for (e = *head; e != -1; e = next[e]) {
+        if (n < SORT_LOC) loc[n] = e;
+        ++n;
+    }
+    if (n <= 1) return;
+    if (n <= SORT_LOC) {
+        int j, k;
+        for (j = 0; j < n - 1; ++j)
+            for (k = j + 1; k < n; ++k)
+                if (w[loc[j]] < w[loc[k]]) {
+                    int t = loc[j]; loc[j] = loc[k]; loc[k] = t;
+                }
+        *head = loc[0];
+        for (j = 0; j + 1 < n; ++j) next[loc[j]] = loc[j + 1];
+        next[loc[n - 1]] = -1;
+        *tail = loc[n - 1];
+        return;
+    }
+    /* rare big-degree fallback: stable selection by chain re-linking */
+    int out_head = -1, out_tail = -1;
+    while (*head != -1) {
+        int best = *head, prev_best = -1, prev = *head;
+        for (e = next[*head]; e != -1; prev = e, e = next[e])
+            if (w[e] > w[best]) { best = e; prev_best = prev; }
+        if (prev_best == -1) *head = next[best];
+        else next[prev_best] = next[best];
+        if (out_tail == -1) out_head = best;
+        else next[out_tail] = best;
+        out_tail = best;
+    }
+    next[out_tail] = -1;
+    *head = out_head;
+    *tail = out_tail;
+}
+
 __device__ static void dev_par_sort_adjacency(flat_graph_t *fg, int lane) {
-    int scratch[1024];
     for (int i = lane; i < fg->node_n; i += FOLD_WAVE) {
-        int j, k, e, n;
-        n = 0;
-        for (e = fg->in_head[i]; e != -1; e = fg->in_next[e]) {
-            if (n >= 1024) abort();
-            scratch[n++] = e;
-        }
-        for (j = 0; j < n - 1; ++j)
-            for (k = j + 1; k < n; ++k)
-                if (fg->in_w[scratch[j]] < fg->in_w[scratch[k]]) {
-                    int t = scratch[j]; scratch[j] = scratch[k]; scratch[k] = t;
-                }
-        fg->in_head[i] = n ? scratch[0] : -1;
-        for (j = 0; j + 1 < n; ++j) fg->in_next[scratch[j]] = scratch[j + 1];
-        if (n) { fg->in_next[scratch[n - 1]] = -1; fg->in_tail[i] = scratch[n - 1]; }
-        n = 0;
-        for (e = fg->out_head[i]; e != -1; e = fg->out_next[e]) {
-            if (n >= 1024) abort();
-            scratch[n++] = e;
-        }
-        for (j = 0; j < n - 1; ++j)
-            for (k = j + 1; k < n; ++k)
-                if (fg->out_w[scratch[j]] < fg->out_w[scratch[k]]) {
-                    int t = scratch[j]; scratch[j] = scratch[k]; scratch[k] = t;
-                }
-        fg->out_head[i] = n ? scratch[0] : -1;
-        for (j = 0; j + 1 < n; ++j) fg->out_next[scratch[j]] = scratch[j + 1];
-        if (n) { fg->out_next[scratch[n - 1]] = -1; fg->out_tail[i] = scratch[n - 1]; }
+        dev_sort_chain(&fg->in_head[i], &fg->in_tail[i], fg->in_next, fg->in_w);
+        dev_sort_chain(&fg->out_head[i], &fg->out_tail[i], fg->out_next, fg->out_w);
     }
 }
 
