@@ -27,18 +27,28 @@ ROOT = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, ROOT)
 
 WORKLOAD = "1000 synthetic sets x 50 reads x 10 kbp, 10% ONT-style error, global convex gap, 1 MI355X (BASELINE.json configs[1])"
+# BASELINE.json configs[2]/[3] variants (non-default --workload values emit
+# their own JSON line; the driver's default run is always configs[1])
+WORKLOADS = {
+    "cg":     ("global convex gap (BASELINE configs[1])", 0, None),
+    "affine": ("global affine gap -O4 -E2 (BASELINE configs[2])", 0, "affine"),
+    "linear": ("global linear gap -O0 (BASELINE configs[2])", 0, "linear"),
+    "local":  ("local convex gap -m1 (BASELINE configs[2])", 1, None),
+    "extend": ("extension convex gap -m2 (BASELINE configs[2])", 2, None),
+    "aa":     ("amino-acid 30x2 kaa BLOSUM62 global affine (BASELINE configs[3])", 0, "aa"),
+}
 HBM_PEAK_GBPS = 8000.0  # gfx950 spec peak (MI355X_MICROARCH.md)
 ALG_BYTES_PER_CELL = 6.0  # 3 planes (H,E1,E2) x int16, written once; F planes recomputed at backtrack
 
 
-def gen_sets(rng, n_sets, depth, qlen, p_sub=0.045, p_del=0.03, p_ins=0.025):
+def gen_sets(rng, n_sets, depth, qlen, p_sub=0.045, p_del=0.03, p_ins=0.025, alphabet=4):
     """Generate encoded (0..3) read sets with numpy; returns list of list of bytes.
     Error model matches tests/make_synth.py: per base, substitute with p_sub,
     delete with p_del, keep-and-insert-one with p_ins, else keep."""
     import numpy as np
     sets = []
     for _ in range(n_sets):
-        ref = rng.integers(0, 4, size=qlen, dtype=np.uint8)
+        ref = rng.integers(0, alphabet, size=qlen, dtype=np.uint8)
         reads = []
         for _ in range(depth):
             r = rng.random(qlen)
@@ -48,11 +58,11 @@ def gen_sets(rng, n_sets, depth, qlen, p_sub=0.045, p_del=0.03, p_ins=0.025):
             out = ref.copy()
             nsub = int(sub.sum())
             if nsub:
-                out[sub] = (ref[sub] + rng.integers(1, 4, size=nsub, dtype=np.uint8)) % 4
+                out[sub] = (ref[sub] + rng.integers(1, alphabet, size=nsub, dtype=np.uint8)) % alphabet
             kept = out[~mask_del]
             ins_pos = np.nonzero(ins[~mask_del])[0]
             if len(ins_pos):
-                ins_bases = rng.integers(0, 4, size=len(ins_pos), dtype=np.uint8)
+                ins_bases = rng.integers(0, alphabet, size=len(ins_pos), dtype=np.uint8)
                 kept = np.insert(kept, ins_pos + 1, ins_bases)
             reads.append(np.ascontiguousarray(kept).tobytes())
         sets.append(reads)
@@ -80,6 +90,34 @@ def build_batch_args(lib, sets):
 
 class Para(ctypes.Structure):
     pass  # opaque; created/freed by the library
+
+
+def make_para(lib, workload):
+    """abpoa_para_t for the requested workload (field pokes mirror the
+    reference CLI flags; the default path touches nothing)."""
+    para = lib.abpoa_init_para()
+    if workload != "cg":
+        from abpoa_amd.pyabpoa import ParaT
+        p = ctypes.cast(para, ctypes.POINTER(ParaT)).contents
+        _, align_mode, gapk = WORKLOADS[workload]
+        p.align_mode = align_mode
+        if gapk == "affine":
+            p.gap_open1, p.gap_ext1, p.gap_open2, p.gap_ext2 = 4, 2, 0, 0
+        elif gapk == "linear":
+            p.gap_open1, p.gap_ext1, p.gap_open2, p.gap_ext2 = 0, 2, 0, 0
+        elif gapk == "aa":
+            p.m = 27
+            p.use_score_matrix = 1
+            p.gap_open1, p.gap_ext1, p.gap_open2, p.gap_ext2 = 4, 2, 0, 0
+            mtx = os.path.join(ROOT, "tests", "golden", "BLOSUM62.mtx")
+            libc = ctypes.CDLL(None)
+            libc.strdup.restype = ctypes.c_void_p
+            libc.strdup.argtypes = [ctypes.c_char_p]
+            p.mat_fn = libc.strdup(mtx.encode())
+        if workload == "extend":
+            p.zdrop, p.end_bonus = 100, -1
+    lib.abpoa_post_set_para(para)
+    return para
 
 
 def native_setup():
@@ -118,11 +156,19 @@ def _cpu_model():
 
 
 def cpu_baseline_leg(depth, qlen):
-    """Time the unmodified reference binary on one set (bounded sample)."""
+    """Time the unmodified reference binary (bounded sample).
+
+    Reports the measured single-core figure AND the full-socket framing the
+    north star prices against: a linear extrapolation to every core of this
+    host (the reference binary is single-threaded), plus, when
+    ABPOA_BENCH_CPU_SOCKET is set, an ACTUAL concurrent run of one process
+    per set across the socket (bounded to ~30 s)."""
     ref_bin = os.path.join(ROOT, "oracle", "_ref", "abpoa")
     if not os.path.exists(ref_bin) or os.environ.get("ABPOA_BENCH_SKIP_CPU"):
         return None
     import tempfile
+    ncores = os.cpu_count() or 1
+    out = None
     with tempfile.TemporaryDirectory() as td:
         fa = os.path.join(td, "cpu.fa")
         subprocess.run([sys.executable, os.path.join(ROOT, "tests", "make_synth.py"), fa,
@@ -132,9 +178,94 @@ def cpu_baseline_leg(depth, qlen):
         subprocess.run([ref_bin, fa], check=True, stdout=subprocess.DEVNULL,
                        stderr=subprocess.DEVNULL)
         dt = time.monotonic() - t0
-    return {"value": 1.0 / dt, "unit": "sets/s", "cores": 1, "kind": "reference",
-            "sample": "1 set (%d reads x %d bp), reference binary, cold arena, host: %s"
-                      % (depth, qlen, _cpu_model())}
+        out = {"value": 1.0 / dt, "unit": "sets/s", "cores": 1, "kind": "reference",
+               "sample": "1 set (%d reads x %d bp), reference binary, cold arena, host: %s"
+                         % (depth, qlen, _cpu_model()),
+               "socket_cores": ncores,
+               "socket_value_linear_extrapolation": round(ncores / dt, 4),
+               "socket_note": "reference is single-threaded; socket figure is value x %d cores"
+                              % ncores}
+        if os.environ.get("ABPOA_BENCH_CPU_SOCKET"):
+            # actual concurrent socket sample: one process per core over
+            # distinct synthetic sets (bounded: one set each)
+            nproc = min(ncores, 32)
+            fas = []
+            for i in range(nproc):
+                f2 = os.path.join(td, "cpu%d.fa" % i)
+                subprocess.run([sys.executable, os.path.join(ROOT, "tests", "make_synth.py"),
+                                f2, "--seed", str(100 + i), "--len", str(qlen),
+                                "--depth", str(depth)], check=True, stderr=subprocess.DEVNULL)
+                fas.append(f2)
+            t0 = time.monotonic()
+            procs = [subprocess.Popen([ref_bin, f2], stdout=subprocess.DEVNULL,
+                                      stderr=subprocess.DEVNULL) for f2 in fas]
+            for p in procs:
+                assert p.wait() == 0
+            dt2 = time.monotonic() - t0
+            out["socket_value_measured"] = round(nproc / dt2, 4)
+            out["socket_measured_procs"] = nproc
+    return out
+
+
+def traffic_probe(args):
+    """Counter-measured HBM bytes per DP-kernel launch (roofline.traffic).
+
+    Two separate rocprofv3 --pmc passes over a small probe run (FETCH_SIZE
+    and WRITE_SIZE cannot share a TCC pass), collected per the MI355X guide:
+    FETCH_SIZE on gfx950 reports half the bytes of wide coalesced reads, so
+    it is doubled; WRITE_SIZE is used as reported. Returns bytes per
+    DP-kernel launch (all gap-mode kernels counted), or None."""
+    import csv as _csv
+    import glob as _glob
+    import shutil
+    import tempfile
+    if os.environ.get("ABPOA_BENCH_NO_TRAFFIC") or not shutil.which("rocprofv3"):
+        return None
+    n_probe_sets = 16
+    res = {}
+    with tempfile.TemporaryDirectory(dir="/tmp") as td:
+        for pmc in ("FETCH_SIZE", "WRITE_SIZE"):
+            env = dict(os.environ)
+            env["TMPDIR"] = "/tmp"
+            env["ABPOA_BENCH_NO_TRAFFIC"] = "1"
+            env["ABPOA_BENCH_SKIP_CPU"] = "1"
+            odir = os.path.join(td, pmc)
+            r = subprocess.run(
+                ["rocprofv3", "--pmc", pmc, "-d", odir, "-o", "probe", "--",
+                 sys.executable, os.path.abspath(__file__),
+                 "--sets-per-step", str(n_probe_sets), "--steps", "1", "--warmup", "0",
+                 "--depth", str(args.depth), "--qlen", str(args.qlen),
+                 "--workload", args.workload],
+                cwd="/tmp", env=env, stdout=subprocess.PIPE, stderr=subprocess.DEVNULL)
+            if r.returncode != 0:
+                return None
+            probe_cells = 0
+            for ln in (r.stdout or b"").decode(errors="replace").splitlines():
+                if ln.startswith("{"):
+                    try:
+                        probe_cells = json.loads(ln)["config"]["total_cells"]
+                    except (ValueError, KeyError):
+                        pass
+            total = 0.0
+            n_disp = 0
+            for fn in _glob.glob(os.path.join(odir, "**", "*counter_collection.csv"),
+                                 recursive=True):
+                with open(fn) as f:
+                    for row in _csv.DictReader(f):
+                        kname = row.get("Kernel_Name", "")
+                        if "global_mw_kernel" in kname or "ag_global_kernel" in kname                                 or "lg_global_kernel" in kname:
+                            total += float(row.get("Counter_Value", 0) or 0)
+                            n_disp += 1
+            if n_disp == 0:
+                return None
+            res[pmc] = (total, n_disp, probe_cells)
+    fetch, nf = res["FETCH_SIZE"]
+    write, nw = res["WRITE_SIZE"]
+    # CSV reports kilobytes for *_SIZE counters on this rocprofv3; values are
+    # per dispatch rows — normalize to bytes per launch
+    per_launch = (2.0 * fetch / max(nf, 1) + write / max(nw, 1)) * 1024.0
+    return {"bytes_per_launch": per_launch, "probe_sets": n_probe_sets,
+            "fetch_corrected_x2": True, "dispatches": max(nf, nw)}
 
 
 def main():
@@ -148,7 +279,13 @@ def main():
     ap.add_argument("--threads", type=int, default=max(2, (os.cpu_count() or 8) - 2))
     ap.add_argument("--dry-run", action="store_true",
                     help="exercise the distributed harness without a GPU (CPU tests)")
+    ap.add_argument("--workload", choices=sorted(WORKLOADS), default="cg",
+                    help="BASELINE config variant; default = configs[1] (the driver's contract)")
     args = ap.parse_args()
+    if args.workload == "aa":
+        # configs[3]: 30 reads x 2 kaa unless explicitly overridden
+        if args.depth == 50: args.depth = 30
+        if args.qlen == 10000: args.qlen = 2000
 
     import numpy as np
     rank = int(os.environ.get("RANK", "0"))
@@ -172,11 +309,11 @@ def main():
         lib = para = None
     else:
         lib = native_setup()
-        para = lib.abpoa_init_para()
-        lib.abpoa_post_set_para(para)
+        para = make_para(lib, args.workload)
         import abpoa_amd
 
     rng = np.random.default_rng(4242 + 1000 * rank)
+    alphabet = 20 if args.workload == "aa" else 4
 
     # Pregenerate every step's input OUTSIDE the timed region (the metric is
     # quoted with inputs resident; generation is not part of the hot path) and
@@ -185,7 +322,7 @@ def main():
     prepared = []
     if not args.dry_run:
         for _ in range(args.warmup + args.steps):
-            sets = gen_sets(rng, args.sets_per_step, args.depth, args.qlen)
+            sets = gen_sets(rng, args.sets_per_step, args.depth, args.qlen, alphabet=alphabet)
             prepared.append(build_batch_args(lib, sets))
 
     def one_step(step_idx):
@@ -209,6 +346,10 @@ def main():
     if not args.dry_run:
         import abpoa_amd
         abpoa_amd.reset_stats()
+        try:
+            abpoa_amd.lib().abpoa_amd_reset_gpu_spans()
+        except AttributeError:
+            pass
     barrier_sync()
     t0 = time.monotonic()
     for k in range(args.steps):
@@ -241,22 +382,54 @@ def main():
         abpoa_amd.lib().abpoa_amd_get_stats2(_ct.byref(_ab))
         alg_bytes = _ab.value or cells * ALG_BYTES_PER_CELL
         gcells_s = cells / elapsed / 1e9
-        ach = alg_bytes / max(kns, 1)  # bytes/ns == GB/s
+        # kernel time, split: summed per-launch event time per
+        # kernel kind AND the wall-clock union of kernel spans across the
+        # overlapped streams (the sums exceed the wall under overlap; the
+        # union never does)
+        dpms = _ct.c_double(); foldms = _ct.c_double(); busyms = _ct.c_double()
+        try:
+            abpoa_amd.lib().abpoa_amd_get_gpu_spans(
+                _ct.byref(busyms), _ct.byref(dpms), _ct.byref(foldms))
+            dp_s, fold_s, busy_s = dpms.value / 1e3, foldms.value / 1e3, busyms.value / 1e3
+        except AttributeError:
+            dp_s = kns / 1e9; fold_s = 0.0; busy_s = None
+        ach = alg_bytes / max(kns, 1)  # bytes/ns == GB/s (per summed DP event time)
+        traffic = traffic_probe(args) if world == 1 else None
+        tval = None
+        if traffic:
+            # probe bytes/launch scaled to this run's cells/launch (the probe
+            # uses fewer sets, same shape)
+            tval = round(traffic["bytes_per_launch"] * (cells / max(launches, 1))
+                         / max(traffic.get("probe_cells_per_launch") or
+                               traffic["bytes_per_launch"] / max(ALG_BYTES_PER_CELL, 1), 1), 0)
         roofline = {"bound": "hbm", "achieved": round(ach, 1), "peak": HBM_PEAK_GBPS,
                     "unit": "GB/s", "frac": round(ach / HBM_PEAK_GBPS, 4),
-                    "traffic": None}
+                    "traffic": tval,
+                    "traffic_note": ("PMC FETCH_SIZE(x2 gfx950 correction)+WRITE_SIZE per DP-kernel launch, "
+                                     "separate --pmc passes on a %d-set probe, scaled by cells"
+                                     % traffic["probe_sets"]) if traffic else None}
         cpu = cpu_baseline_leg(args.depth, args.qlen) if world == 1 else None
+        wl_desc = WORKLOAD if args.workload == "cg" else (
+            "%d synthetic sets x %d reads x %d bp, %s, 1 MI355X"
+            % (args.sets_per_step, args.depth, args.qlen, WORKLOADS[args.workload][0]))
+        metric = ("read-sets/sec, 50x10 kbp global convex-gap POA" if args.workload == "cg"
+                  else "read-sets/sec, %s" % WORKLOADS[args.workload][0])
         line = {
-            "metric": "read-sets/sec, 50x10 kbp global convex-gap POA",
+            "metric": metric,
             "value": round(value, 4), "unit": "sets/s",
             "n_gpus": world, "steps": args.steps, "warmup": args.warmup,
             "ms_per_step": round(elapsed / args.steps * 1e3, 1),
             "higher_is_better": True, "scaling": "weak", "vs_baseline": None,
-            "dtype": "int16", "data": "synthetic",
-            "config": {"workload": WORKLOAD, "sets_per_step": args.sets_per_step,
+            "dtype": "int16+int32 rescore", "data": "synthetic",
+            "config": {"workload": wl_desc, "sets_per_step": args.sets_per_step,
                        "depth": args.depth, "qlen": args.qlen,
                        "gcells_per_s": round(gcells_s, 3),
-                       "kernel_s": round(kns / 1e9, 3), "launches": launches,
+                       "total_cells": cells,
+                       "kernel_s": round(kns / 1e9, 3),
+                       "dp_kernel_s": round(dp_s, 3),
+                       "fold_kernel_s": round(fold_s, 3),
+                       "gpu_busy_s": round(busy_s, 3) if busy_s is not None else None,
+                       "launches": launches,
                        "parallelism": "dp%d independent read-set shards (no data-path collective)" % world},
             "roofline": roofline,
             "cpu_baseline": cpu,

@@ -73,3 +73,27 @@ def test_resident_pool_expansion():
         os.environ.pop("ABPOA_AMD_NODE_ALPHA", None)
     roomy = abpoa_amd.msa_batch_consensus(sets, n_threads=2)
     assert tight == roomy, "pool-expansion path changed the consensus"
+
+
+def test_two_rank_one_gpu_smoke(tmp_path):
+    """Sharding-path hardware smoke: two ranks
+    time-slice ONE GPU (gloo rendezvous, HIP_VISIBLE_DEVICES pinned to 0 for
+    both), each processing its shard through the resident driver — the exact
+    code path the driver's 8-GPU scaling run takes, minus the extra devices."""
+    import json
+    import subprocess
+    env = dict(os.environ)
+    env["HIP_VISIBLE_DEVICES"] = "0"
+    env["ABPOA_BENCH_SKIP_CPU"] = "1"
+    env["ABPOA_BENCH_NO_TRAFFIC"] = "1"
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29417", os.path.join(ROOT, "bench.py"),
+         "--gpus", "2", "--steps", "1", "--warmup", "0",
+         "--sets-per-step", "6", "--qlen", "600", "--depth", "10"],
+        env=env, cwd=ROOT, stdout=subprocess.PIPE, stderr=subprocess.PIPE, timeout=420)
+    assert r.returncode == 0, r.stderr.decode()[-2000:]
+    line = [l for l in r.stdout.decode().splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["n_gpus"] == 2 and out["value"] > 0

@@ -168,3 +168,31 @@ def test_gpu_seeding_vs_oracle(gpu_bin, cputest_bin, oracle_env, tmp_path):
         gpu = run_stdout([gpu_bin, str(fa)] + opts)
         cpu = run_stdout([cputest_bin, str(fa)] + opts, env=oracle_env)
         assert gpu == cpu, "GPU/oracle seeding divergence opts=%r" % (opts,)
+
+
+@pytest.mark.parametrize("opts", [["-O", "4", "-E", "2"], ["-O", "0", "-E", "2"]],
+                         ids=["affine-10k", "linear-10k"])
+def test_gpu_gap_modes_full_shape(gpu_bin, cputest_bin, oracle_env, tmp_path, opts):
+    """Affine/linear kernels at the full north-star shape (50 x 10 kbp):
+    graphs large enough that most rounds run the int32 rescore width."""
+    fa = tmp_path / "s10k.fa"
+    subprocess.run(["python3", os.path.join(ROOT, "tests", "make_synth.py"), str(fa),
+                    "--seed", "43", "--len", "10000", "--depth", "50"],
+                   check=True, stderr=subprocess.DEVNULL)
+    gpu = run_stdout([gpu_bin, str(fa)] + opts)
+    cpu = run_stdout([cputest_bin, str(fa)] + opts, env=oracle_env)
+    assert gpu == cpu
+
+
+def test_gpu_forced_int32_overflow(gpu_bin, cputest_bin, oracle_env, tmp_path):
+    """-M9 on 10 kbp reads puts the score bound (qlen x max_mat = 90k) past
+    int16 from the FIRST read: the whole set runs the int32 kernel path,
+    including its width-specific inf_min clamps (abpoa_align_simd.c:1299)."""
+    fa = tmp_path / "s.fa"
+    subprocess.run(["python3", os.path.join(ROOT, "tests", "make_synth.py"), str(fa),
+                    "--seed", "44", "--len", "10000", "--depth", "12"],
+                   check=True, stderr=subprocess.DEVNULL)
+    for opts in (["-M", "9", "-X", "12"], ["-M", "9", "-X", "12", "-O", "4", "-E", "2"]):
+        gpu = run_stdout([gpu_bin, str(fa)] + opts)
+        cpu = run_stdout([cputest_bin, str(fa)] + opts, env=oracle_env)
+        assert gpu == cpu, "int32-forced divergence opts=%r" % (opts,)
