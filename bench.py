@@ -167,7 +167,18 @@ def cpu_baseline_leg(depth, qlen):
     if not os.path.exists(ref_bin) or os.environ.get("ABPOA_BENCH_SKIP_CPU"):
         return None
     import tempfile
-    ncores = os.cpu_count() or 1
+    nthreads = os.cpu_count() or 1
+    # one physical socket (the north star's framing): cores/socket from cpuinfo
+    socket_cores = None
+    try:
+        with open("/proc/cpuinfo") as f:
+            for ln in f:
+                if ln.startswith("cpu cores"):
+                    socket_cores = int(ln.split(":")[1])
+                    break
+    except (OSError, ValueError):
+        pass
+    ncores = socket_cores or nthreads
     out = None
     with tempfile.TemporaryDirectory() as td:
         fa = os.path.join(td, "cpu.fa")
@@ -183,8 +194,10 @@ def cpu_baseline_leg(depth, qlen):
                          % (depth, qlen, _cpu_model()),
                "socket_cores": ncores,
                "socket_value_linear_extrapolation": round(ncores / dt, 4),
-               "socket_note": "reference is single-threaded; socket figure is value x %d cores"
-                              % ncores}
+               "machine_threads": nthreads,
+               "socket_note": "reference is single-threaded; socket figure is value x %d "
+                              "physical cores of one socket (machine has %d logical CPUs)"
+                              % (ncores, nthreads)}
         if os.environ.get("ABPOA_BENCH_CPU_SOCKET"):
             # actual concurrent socket sample: one process per core over
             # distinct synthetic sets (bounded: one set each)
@@ -231,7 +244,8 @@ def traffic_probe(args):
             env["ABPOA_BENCH_SKIP_CPU"] = "1"
             odir = os.path.join(td, pmc)
             r = subprocess.run(
-                ["rocprofv3", "--pmc", pmc, "-d", odir, "-o", "probe", "--",
+                ["rocprofv3", "--pmc", pmc, "--output-format", "csv",
+                 "-d", odir, "-o", "probe", "--",
                  sys.executable, os.path.abspath(__file__),
                  "--sets-per-step", str(n_probe_sets), "--steps", "1", "--warmup", "0",
                  "--depth", str(args.depth), "--qlen", str(args.qlen),
