@@ -107,6 +107,11 @@ def make_para(lib, workload):
             p.gap_open1, p.gap_ext1, p.gap_open2, p.gap_ext2 = 0, 2, 0, 0
         elif gapk == "aa":
             p.m = 27
+            libc0 = ctypes.CDLL(None)
+            libc0.realloc.restype = ctypes.c_void_p
+            libc0.realloc.argtypes = [ctypes.c_void_p, ctypes.c_size_t]
+            newmat = libc0.realloc(ctypes.cast(p.mat, ctypes.c_void_p), 27 * 27 * 4)
+            p.mat = ctypes.cast(newmat, ctypes.POINTER(ctypes.c_int))
             p.use_score_matrix = 1
             p.gap_open1, p.gap_ext1, p.gap_open2, p.gap_ext2 = 4, 2, 0, 0
             mtx = os.path.join(ROOT, "tests", "golden", "BLOSUM62.mtx")
