@@ -446,6 +446,21 @@ void cg_global_mw_kernel(const abamd_gpu_job_t *__restrict__ jobs,
             if (more && j == ss + MWT - 1) {
                 sc_carry[0] = hpre; sc_carry[1] = f1; sc_carry[2] = f2;
             }
+            if (!more && (jb.banded || local_mode || extend_mode)) {
+                /* last superchunk: fold the per-wave argmax publish into this
+                 * barrier instead of paying a fourth barrier (and its store
+                 * drain) after the loop */
+                int mvw = wave_red_max_i32((int)lmax);
+                int llw = ((int)lmax == mvw && lleft >= 0) ? lleft : 0x7fffffff;
+                int rrw = ((int)lmax == mvw && lright >= 0) ? lright : -1;
+                llw = wave_red_min_i32(llw);
+                rrw = wave_red_max_i32(rrw);
+                if (lane == 0) {
+                    sc_red[wv] = mvw;
+                    sc_red[MWAVES + wv] = llw;
+                    sc_red[2 * MWAVES + wv] = rrw;
+                }
+            }
             __syncthreads(); /* also orders cw writes before the next row's reads */
             if (more) { carry_h = sc_carry[0]; f1c = sc_carry[1]; f2c = sc_carry[2]; }
 #ifdef ABAMD_KPROF
@@ -463,19 +478,8 @@ void cg_global_mw_kernel(const abamd_gpu_job_t *__restrict__ jobs,
         buf_cur ^= 1;
 
         if (jb.banded || local_mode || extend_mode) {
-            /* per-wave argmax, then a cross-wave combine every thread
-             * repeats from the 8 published triples (uniform result) */
-            int mvw = wave_red_max_i32((int)lmax);
-            int llw = ((int)lmax == mvw && lleft >= 0) ? lleft : 0x7fffffff;
-            int rrw = ((int)lmax == mvw && lright >= 0) ? lright : -1;
-            llw = wave_red_min_i32(llw);
-            rrw = wave_red_max_i32(rrw);
-            if (lane == 0) {
-                sc_red[wv] = mvw;
-                sc_red[MWAVES + wv] = llw;
-                sc_red[2 * MWAVES + wv] = rrw;
-            }
-            __syncthreads();
+            /* per-wave argmax published under the loop's last barrier; every
+             * thread repeats the 8-way combine from LDS (uniform result) */
             int mv = sc_red[0], ll = sc_red[MWAVES], rr = sc_red[2 * MWAVES];
             #pragma unroll
             for (int ww = 1; ww < MWAVES; ++ww) {
