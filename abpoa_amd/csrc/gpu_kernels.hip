@@ -1491,27 +1491,892 @@ void lg_global_kernel(const abamd_gpu_job_t *__restrict__ jobs,
     }
 }
 
+/* ------------------------------------------------------------------ */
+/* Multi-wave affine kernel (3 planes H,E1,F1): same 8-waves-per-job   */
+/* design as the convex kernel; the F chain feeds from H BEFORE the E  */
+/* fold (simd_abpoa_ag_dp, abpoa_align_simd.c:817-933) and the stored  */
+/* E is inf_min when the insertion won the cell (SIMDSetIfEqual).      */
+/* ------------------------------------------------------------------ */
+template <typename S>
+__global__ __launch_bounds__(MWT, 1)
+void ag_global_mw_kernel(const abamd_gpu_job_t *__restrict__ jobs,
+                         abamd_gpu_res_t *__restrict__ results, int n_jobs) {
+    const int jid = blockIdx.x;
+    if (jid >= n_jobs) return;
+    const int tid = threadIdx.x;
+    const int wv = tid / WAVE, lane = tid % WAVE;
+
+    __shared__ int mat_lds[27 * 27];
+    __shared__ S prev_lds[2][2 * BMW];
+    __shared__ S sc_h[MWAVES], sc_f1[MWAVES];
+    __shared__ int sc_red[3 * MWAVES];
+    __shared__ S sc_carry[2];
+
+    const abamd_gpu_job_t jb = jobs[jid];
+    abamd_gpu_res_t *res = &results[jid];
+    abamd_row_meta_t *__restrict__ meta = (abamd_row_meta_t*)jb.row_meta;
+
+    const int qlen = jb.qlen, n_rows = jb.n_rows, w = jb.w, m = jb.m;
+    const S inf_min = (S)jb.inf_min;
+    const S e1 = (S)jb.e1, oe1 = (S)jb.oe1;
+    const int local_mode = jb.align_mode == 1, extend_mode = jb.align_mode == 2;
+    int32_t run_best = jb.inf_min;
+    int run_best_i = 0, run_best_j = 0, run_best_remain = jb.max_remain[0];
+    int zdropped = 0;
+    const int end_remain = jb.max_remain[n_rows - 1];
+    S *arena = (S*)jb.arena;
+    const uint8_t *__restrict__ query = jb.query;
+
+    for (int i = tid; i < m * m; i += MWT) mat_lds[i] = jb.mat[i];
+    if (tid == 0) { res->status = ABAMD_JOB_OK; res->n_cigar = 0; }
+
+    for (int i = tid; i < n_rows; i += MWT) {
+        jb.max_left[i] = jb.node_n_init;
+        jb.max_right[i] = 0;
+    }
+    if (tid == 0) { jb.max_left[0] = 0; jb.max_right[0] = 0; }
+    for (int k = jb.out_off[0] + tid; k < jb.out_off[1]; k += MWT) {
+        int o = jb.out_idx[k];
+        jb.max_left[o] = 1; jb.max_right[o] = 1;
+    }
+    __syncthreads();
+
+    int buf_cur = 0;
+    int prev_ok = 0, prev_row = -1, prev_beg = 0, prev_end = -1;
+
+    /* first row (simd_abpoa_ag_first_dp, abpoa_align_simd.c:651-667) */
+    int64_t used;
+    {
+        int mr = jb.max_remain[0] - end_remain - 1;
+        int end0;
+        if (jb.banded) {
+            int t = jb.max_right[0] > qlen - mr ? jb.max_right[0] : qlen - mr;
+            end0 = (qlen < t + w) ? qlen : t + w;
+        } else end0 = qlen;
+        if (tid == 0) { meta[0].beg = 0; meta[0].end = end0; meta[0].off = 0; }
+        int64_t bw = end0 + 1;
+        used = bw;
+        S *H = arena, *E1 = arena + bw, *F1 = arena + 2 * bw;
+        S *c = &prev_lds[buf_cur][0];
+        const int fits = end0 + 1 <= BMW;
+        for (int j = tid; j <= end0; j += MWT) {
+            S hv, e1v2;
+            if (local_mode) {
+                hv = 0; e1v2 = 0; F1[j] = 0;
+            } else if (j == 0) {
+                hv = 0; e1v2 = (S)(0 - oe1);
+                F1[0] = inf_min;
+            } else {
+                S f1 = (S)(-(jb.o1 + jb.e1 * j));
+                F1[j] = f1;
+                hv = f1; e1v2 = inf_min;
+            }
+            H[j] = hv; E1[j] = e1v2;
+            if (fits) { c[j] = hv; c[BMW + j] = e1v2; }
+        }
+        if (fits) { prev_ok = 1; prev_row = 0; prev_beg = 0; prev_end = end0; }
+        buf_cur ^= 1;
+        __syncthreads();
+    }
+
+    int cur_pk0 = jb.pre_off[1];
+    int cur_pk1 = jb.pre_off[2 <= n_rows ? 2 : n_rows];
+    int cur_oo0 = jb.out_off[1];
+    int cur_oo1 = jb.out_off[2 <= n_rows ? 2 : n_rows];
+    int cur_remain = jb.max_remain[1];
+    int cur_ml_mem = jb.max_left[1];
+    int cur_mr_mem = jb.max_right[1];
+    int cur_pidx0 = jb.pre_idx[cur_pk0];
+    int cur_ps0 = jb.pre_ps[cur_pk0];
+    int push_ml = 0x7fffffff, push_mr = -0x7fffffff;
+    for (int r = 1; r < n_rows - 1; ++r) {
+        const int pk0 = cur_pk0, pk1 = cur_pk1;
+        const int oo0 = cur_oo0, oo1 = cur_oo1;
+        const int row_remain = cur_remain;
+        const int pidx0 = cur_pidx0;
+        const S ps0 = (S)cur_ps0;
+        const int ml_eff = cur_ml_mem < push_ml ? cur_ml_mem : push_ml;
+        const int mr_eff = cur_mr_mem > push_mr ? cur_mr_mem : push_mr;
+        {
+            const int nr = r + 1;
+            cur_pk0 = pk1;
+            cur_pk1 = jb.pre_off[nr + 1];
+            cur_oo0 = oo1;
+            cur_oo1 = jb.out_off[nr + 1];
+            cur_remain = jb.max_remain[nr];
+            cur_ml_mem = jb.max_left[nr];
+            cur_mr_mem = jb.max_right[nr];
+            cur_pidx0 = jb.pre_idx[cur_pk0];
+            cur_ps0 = jb.pre_ps[cur_pk0];
+        }
+        int beg, end;
+        {
+            int mr = row_remain - end_remain - 1;
+            if (jb.banded) {
+                int lo = ml_eff < qlen - mr ? ml_eff : qlen - mr;
+                beg = lo - w; if (beg < 0) beg = 0;
+                int hi = mr_eff > qlen - mr ? mr_eff : qlen - mr;
+                end = hi + w; if (end > qlen) end = qlen;
+                int min_pre_beg;
+                if (pk1 - pk0 == 1) {
+                    min_pre_beg = (prev_ok && pidx0 == prev_row) ? prev_beg : meta[pidx0].beg;
+                } else {
+                    min_pre_beg = 0x7fffffff;
+                    for (int k = pk0; k < pk1; ++k) {
+                        const int pidx = jb.pre_idx[k];
+                        int pb = (prev_ok && pidx == prev_row) ? prev_beg : meta[pidx].beg;
+                        if (pb < min_pre_beg) min_pre_beg = pb;
+                    }
+                }
+                if (beg < min_pre_beg) beg = min_pre_beg;
+            } else { beg = 0; end = qlen; }
+        }
+        const int64_t bw = end - beg + 1;
+        if (used + bw > jb.arena_cap) { if (tid == 0) res->status = ABAMD_JOB_ARENA_OVERFLOW; return; }
+        const int64_t off = used;
+        if (tid == 0) { meta[r].beg = beg; meta[r].end = end; meta[r].off = off; }
+        used += bw;
+        S *H = arena + off * 3, *E1r = H + bw, *F1r = E1r + bw;
+        const uint8_t base = jb.row_base[r];
+        const int *mrow = &mat_lds[base * m];
+        const int cache_fits = bw <= BMW;
+        S *cw = &prev_lds[buf_cur][0];
+        const S *cr = &prev_lds[buf_cur ^ 1][0];
+        const bool fast1 = (pk1 - pk0 == 1) && prev_ok && (pidx0 == prev_row);
+
+        S carry_hm = inf_min, f1c = inf_min;
+        S lmax = inf_min; int lleft = -1, lright = -1;
+
+        for (int ss = beg; ss <= end; ss += MWT) {
+            const int j = ss + tid;
+            const bool act = j <= end;
+            S h = inf_min, e1v = inf_min;
+            if (fast1) {
+                if (act) {
+                    if (local_mode && j == 0) { if (ps0 > h) h = ps0; }
+                    if (j - 1 >= prev_beg && j - 1 <= prev_end) {
+                        S v = (S)(cr[j - 1 - prev_beg] + ps0);
+                        if (v > h) h = v;
+                    }
+                    if (j >= prev_beg && j <= prev_end) {
+                        S v1 = (S)(cr[BMW + j - prev_beg] + ps0);
+                        if (v1 > e1v) e1v = v1;
+                    }
+                }
+            } else for (int k = pk0; k < pk1; ++k) {
+                const int p = jb.pre_idx[k];
+                const S ps = (S)jb.pre_ps[k];
+                if (prev_ok && p == prev_row) {
+                    if (act) {
+                        if (local_mode && j == 0) { if (ps > h) h = ps; }
+                        if (j - 1 >= prev_beg && j - 1 <= prev_end) {
+                            S v = (S)(cr[j - 1 - prev_beg] + ps);
+                            if (v > h) h = v;
+                        }
+                        if (j >= prev_beg && j <= prev_end) {
+                            S v1 = (S)(cr[BMW + j - prev_beg] + ps);
+                            if (v1 > e1v) e1v = v1;
+                        }
+                    }
+                    continue;
+                }
+                const abamd_row_meta_t pm = meta[p];
+                const int64_t pbw = pm.end - pm.beg + 1;
+                const S *__restrict__ pH = arena + pm.off * 3;
+                const S *__restrict__ pE1 = pH + pbw;
+                if (act) {
+                    if (local_mode && j == 0) { if (ps > h) h = ps; }
+                    if (j - 1 >= pm.beg && j - 1 <= pm.end) {
+                        S v = (S)(pH[j - 1 - pm.beg] + ps);
+                        if (v > h) h = v;
+                    }
+                    if (j >= pm.beg && j <= pm.end) {
+                        S v1 = (S)(pE1[j - pm.beg] + ps);
+                        if (v1 > e1v) e1v = v1;
+                    }
+                }
+            }
+            const S q = (S)((j == 0 || !act) ? 0 : mrow[query[j - 1]]);
+            S hm = (S)(h + q);   /* M+q: the F chain feeds from this, pre-E */
+            if (!act) hm = inf_min;
+
+            if (lane == WAVE - 1) sc_h[wv] = hm;
+            __syncthreads();
+            S hmshift = (S)__shfl_up((int)hm, 1);
+            S c1;
+            if (lane == 0) {
+                if (wv == 0) {
+                    if (ss == beg) c1 = (S)(inf_min - oe1);
+                    else c1 = smax((S)(carry_hm - oe1), (S)(f1c - e1));
+                } else {
+                    c1 = (S)(sc_h[wv - 1] - oe1);
+                }
+            } else c1 = (S)(hmshift - oe1);
+            S L1 = scan_maxplus(c1, jb.e1, inf_min, lane);
+            if (lane == WAVE - 1) sc_f1[wv] = L1;
+            __syncthreads();
+            S f1 = L1;
+            if (wv > 0) {
+                S C1 = sc_f1[0];
+                for (int ww = 1; ww < wv; ++ww)
+                    C1 = smax(sc_f1[ww], (S)(C1 - (S)(WAVE * jb.e1)));
+                f1 = smax(L1, (S)(C1 - (S)((lane + 1) * jb.e1)));
+            }
+
+            S tmp = smax(hm, e1v);
+            S hf = smax(tmp, f1);
+            if (local_mode) hf = smax(hf, (S)0);
+            S e1n = (hf == tmp) ? smax((S)(e1v - e1), (S)(hf - oe1))
+                                : (local_mode ? (S)0 : inf_min);
+            if (act) {
+                H[j - beg] = hf; E1r[j - beg] = e1n; F1r[j - beg] = f1;
+                if (cache_fits) { cw[j - beg] = hf; cw[BMW + j - beg] = e1n; }
+                if (hf > lmax) { lmax = hf; lleft = j; lright = j; }
+                else if (hf == lmax) { lright = j; }
+            }
+            const bool more = ss + MWT <= end;
+            if (more && j == ss + MWT - 1) {
+                sc_carry[0] = hm; sc_carry[1] = f1;
+            }
+            if (!more && (jb.banded || local_mode || extend_mode)) {
+                int mvw = wave_red_max_i32((int)lmax);
+                int llw = ((int)lmax == mvw && lleft >= 0) ? lleft : 0x7fffffff;
+                int rrw = ((int)lmax == mvw && lright >= 0) ? lright : -1;
+                llw = wave_red_min_i32(llw);
+                rrw = wave_red_max_i32(rrw);
+                if (lane == 0) {
+                    sc_red[wv] = mvw;
+                    sc_red[MWAVES + wv] = llw;
+                    sc_red[2 * MWAVES + wv] = rrw;
+                }
+            }
+            __syncthreads(); /* also orders cw writes before the next row's reads */
+            if (more) { carry_hm = sc_carry[0]; f1c = sc_carry[1]; }
+        }
+
+        if (cache_fits) { prev_ok = 1; prev_row = r; prev_beg = beg; prev_end = end; }
+        else prev_ok = 0;
+        buf_cur ^= 1;
+
+        if (jb.banded || local_mode || extend_mode) {
+            int mv = sc_red[0], ll = sc_red[MWAVES], rr = sc_red[2 * MWAVES];
+            #pragma unroll
+            for (int ww = 1; ww < MWAVES; ++ww) {
+                int m2 = sc_red[ww];
+                if (m2 > mv) { mv = m2; ll = sc_red[MWAVES + ww]; rr = sc_red[2 * MWAVES + ww]; }
+                else if (m2 == mv) {
+                    if (sc_red[MWAVES + ww] < ll) ll = sc_red[MWAVES + ww];
+                    if (sc_red[2 * MWAVES + ww] > rr) rr = sc_red[2 * MWAVES + ww];
+                }
+            }
+            if (local_mode) {
+                if (mv > run_best) { run_best = mv; run_best_i = r; run_best_j = ll; }
+            } else if (extend_mode) {
+                if (mv > run_best) {
+                    run_best = mv; run_best_i = r; run_best_j = rr;
+                    run_best_remain = row_remain;
+                } else if (jb.zdrop > 0) {
+                    int delta = run_best_remain - row_remain;
+                    int dd = delta - (rr - run_best_j); if (dd < 0) dd = -dd;
+                    if (run_best - mv > jb.zdrop + jb.e1 * dd) zdropped = 1;
+                }
+            }
+            push_ml = 0x7fffffff; push_mr = -0x7fffffff;
+            if (!zdropped && jb.banded) {
+                for (int k = oo0; k < oo1; ++k) {
+                    int o = jb.out_idx[k];
+                    if (o == r + 1) {
+                        if (rr + 1 > push_mr) push_mr = rr + 1;
+                        if (ll + 1 < push_ml) push_ml = ll + 1;
+                    } else {
+                        if (rr + 1 > jb.max_right[o]) jb.max_right[o] = rr + 1;
+                        if (ll + 1 < jb.max_left[o]) jb.max_left[o] = ll + 1;
+                    }
+                }
+            }
+            if (zdropped) break;
+        } else {
+            push_ml = 0x7fffffff; push_mr = -0x7fffffff;
+        }
+    }
+
+    __syncthreads();
+    if (tid == 0) res->cells = used;
+    if (tid != 0) return;
+
+    int32_t best_score = run_best;
+    int best_i = run_best_i, best_j = run_best_j;
+    if (jb.align_mode == 0) {
+        best_score = jb.inf_min; best_i = 0; best_j = 0;
+        for (int k = jb.pre_off[n_rows - 1]; k < jb.pre_off[n_rows]; ++k) {
+            const int p = jb.pre_idx[k];
+            const abamd_row_meta_t pm = meta[p];
+            int e = pm.end < qlen ? pm.end : qlen;
+            const S *pH = arena + pm.off * 3;
+            int32_t sc = (e >= pm.beg) ? (int32_t)pH[e - pm.beg] : jb.inf_min;
+            if (sc > best_score) { best_score = sc; best_i = p; best_j = e; }
+        }
+    }
+    res->best_score = best_score;
+    res->best_i = best_i; res->best_j = best_j;
+    if (!jb.ret_cigar) return;
+    { /* simd_abpoa_ag_backtrack (:196-307) */
+        int bi = best_i, bj = best_j, start_i = best_i, start_j = best_j;
+        int cur_op = 0x1f, n_c = 0, status = ABAMD_JOB_OK;
+        int look_end = jb.put_gap_at_end, put_right = jb.put_gap_on_right;
+        int n_aln = 0, n_matched = 0;
+        uint64_t *cig = jb.cigar;
+        int id = jb.row_node_id[bi];
+        if (best_j < qlen) dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, qlen - best_j, -1, qlen - 1, &status);
+        while (bi > 0 && bj > 0 && status == ABAMD_JOB_OK) {
+            const abamd_row_meta_t bm = meta[bi];
+            const int rb = bm.beg, re = bm.end;
+            const int64_t bw = re - rb + 1;
+            const S *H = arena + bm.off * 3;
+            const S *E1r = H + bw, *F1r = E1r + bw;
+            const S Hj = (bj >= rb && bj <= re) ? H[bj - rb] : inf_min;
+            const S Hjm1 = (bj - 1 >= rb && bj - 1 <= re) ? H[bj - 1 - rb] : inf_min;
+            const S E1j = (bj >= rb && bj <= re) ? E1r[bj - rb] : inf_min;
+            const S F1j = (bj >= rb && bj <= re) ? F1r[bj - rb] : inf_min;
+            const S F1jm1 = (bj - 1 >= rb && bj - 1 <= re) ? F1r[bj - 1 - rb] : inf_min;
+            if (local_mode && Hj == 0) break;
+            start_i = bi; start_j = bj;
+            const int pq0 = jb.pre_off[bi], pq1 = jb.pre_off[bi + 1];
+            const S s = (S)mat_lds[m * jb.row_base[bi] + query[bj - 1]];
+            const int is_match = jb.row_base[bi] == query[bj - 1];
+            int hit = 0;
+            if (put_right == 0 && look_end == 0 && (cur_op & 0x1)) {
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
+                    const S *pH = arena + pm.off * 3;
+                    if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
+                        cur_op = 0x1f; hit = 1;
+                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
+                        bi = p; --bj; id = jb.row_node_id[bi];
+                        ++n_aln; n_matched += is_match;
+                        break;
+                    }
+                }
+            }
+            if (!hit && (cur_op & 0x2)) { /* deletion */
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj < pm.beg || bj > pm.end) continue;
+                    const int poffc = bj - pm.beg;
+                    const int64_t pbw = pm.end - pm.beg + 1;
+                    const S *pH = arena + pm.off * 3;
+                    const S *pE1 = pH + pbw;
+                    if (cur_op & 0x1) {
+                        if (Hj == (S)(pE1[poffc] + ps)) {
+                            cur_op = ((S)(pH[poffc] - oe1) == pE1[poffc]) ? (0x1 | 0x18) : 0x2;
+                            hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                            bi = p; id = jb.row_node_id[bi];
+                            if (look_end) look_end = 0;
+                            break;
+                        }
+                    } else {
+                        if (E1j == (S)(pE1[poffc] - e1 + ps)) {
+                            cur_op = ((S)(pH[poffc] - oe1) == pE1[poffc]) ? (0x1 | 0x18) : 0x2;
+                            hit = 1; dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                            bi = p; id = jb.row_node_id[bi];
+                            if (look_end) look_end = 0;
+                            break;
+                        }
+                    }
+                }
+            }
+            if (!hit && (cur_op & 0x18)) { /* insertion */
+                if (cur_op & 0x1) {
+                    if (Hj == F1j) {
+                        if ((S)(Hjm1 - oe1) == F1j) { cur_op = 0x1 | 0x6; hit = 1; }
+                        else if ((S)(F1jm1 - e1) == F1j) { cur_op = 0x8; hit = 1; }
+                    }
+                } else {
+                    if ((S)(Hjm1 - oe1) == F1j) { cur_op = 0x1 | 0x6; hit = 1; }
+                    else if ((S)(F1jm1 - e1) == F1j) { cur_op = 0x8; hit = 1; }
+                }
+                if (hit) {
+                    dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, 1, id, bj - 1, &status);
+                    --bj;
+                    if (look_end) look_end = 0;
+                    ++n_aln;
+                }
+            }
+            if (!hit && (cur_op & 0x1)) {
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
+                    const S *pH = arena + pm.off * 3;
+                    if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
+                        cur_op = 0x1f; hit = 1;
+                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
+                        bi = p; --bj; id = jb.row_node_id[bi];
+                        ++n_aln; n_matched += is_match;
+                        look_end = 0;
+                        break;
+                    }
+                }
+            }
+            if (!hit) { status = ABAMD_JOB_BT_DEAD_END; break; }
+        }
+        if (status == ABAMD_JOB_OK && bj > 0)
+            dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, bj, -1, bj - 1, &status);
+        res->status = status;
+        res->n_cigar = n_c;
+        res->n_aln_bases = n_aln;
+        res->n_matched_bases = n_matched;
+        res->node_e = jb.row_node_id[best_i]; res->query_e = best_j - 1;
+        res->node_s = jb.row_node_id[start_i]; res->query_s = start_j - 1;
+    }
+}
+
+
+/* ------------------------------------------------------------------ */
+/* Multi-wave linear kernel (1 plane): candidates carry the query score */
+/* and the deletion term, then the insertion max-plus scan runs on H    */
+/* itself (simd_abpoa_lg_dp, abpoa_align_simd.c:727-815); the local     */
+/* clamp applies AFTER the scan and the cross-wave carries are          */
+/* PRE-clamp, exactly like the one-wave kernel's readlane carry.        */
+/* ------------------------------------------------------------------ */
+template <typename S>
+__global__ __launch_bounds__(MWT, 1)
+void lg_global_mw_kernel(const abamd_gpu_job_t *__restrict__ jobs,
+                         abamd_gpu_res_t *__restrict__ results, int n_jobs) {
+    const int jid = blockIdx.x;
+    if (jid >= n_jobs) return;
+    const int tid = threadIdx.x;
+    const int wv = tid / WAVE, lane = tid % WAVE;
+
+    __shared__ int mat_lds[27 * 27];
+    __shared__ S prev_lds[2][BMW];
+    __shared__ S sc_f1[MWAVES];
+    __shared__ int sc_red[3 * MWAVES];
+    __shared__ S sc_carry[1];
+
+    const abamd_gpu_job_t jb = jobs[jid];
+    abamd_gpu_res_t *res = &results[jid];
+    abamd_row_meta_t *__restrict__ meta = (abamd_row_meta_t*)jb.row_meta;
+
+    const int qlen = jb.qlen, n_rows = jb.n_rows, w = jb.w, m = jb.m;
+    const S inf_min = (S)jb.inf_min;
+    const S e1 = (S)jb.e1;
+    const int local_mode = jb.align_mode == 1, extend_mode = jb.align_mode == 2;
+    int32_t run_best = jb.inf_min;
+    int run_best_i = 0, run_best_j = 0, run_best_remain = jb.max_remain[0];
+    int zdropped = 0;
+    const int end_remain = jb.max_remain[n_rows - 1];
+    S *arena = (S*)jb.arena;
+    const uint8_t *__restrict__ query = jb.query;
+
+    for (int i = tid; i < m * m; i += MWT) mat_lds[i] = jb.mat[i];
+    if (tid == 0) { res->status = ABAMD_JOB_OK; res->n_cigar = 0; }
+
+    for (int i = tid; i < n_rows; i += MWT) {
+        jb.max_left[i] = jb.node_n_init;
+        jb.max_right[i] = 0;
+    }
+    if (tid == 0) { jb.max_left[0] = 0; jb.max_right[0] = 0; }
+    for (int k = jb.out_off[0] + tid; k < jb.out_off[1]; k += MWT) {
+        int o = jb.out_idx[k];
+        jb.max_left[o] = 1; jb.max_right[o] = 1;
+    }
+    __syncthreads();
+
+    int buf_cur = 0;
+    int prev_ok = 0, prev_row = -1, prev_beg = 0, prev_end = -1;
+
+    /* first row (simd_abpoa_lg_first_dp, abpoa_align_simd.c:635-649) */
+    int64_t used;
+    {
+        int mr = jb.max_remain[0] - end_remain - 1;
+        int end0;
+        if (jb.banded) {
+            int t = jb.max_right[0] > qlen - mr ? jb.max_right[0] : qlen - mr;
+            end0 = (qlen < t + w) ? qlen : t + w;
+        } else end0 = qlen;
+        if (tid == 0) { meta[0].beg = 0; meta[0].end = end0; meta[0].off = 0; }
+        used = end0 + 1;
+        S *H = arena;
+        S *c = &prev_lds[buf_cur][0];
+        const int fits = end0 + 1 <= BMW;
+        for (int j = tid; j <= end0; j += MWT) {
+            S hv = local_mode ? (S)0 : (S)(-jb.e1 * j);
+            H[j] = hv;
+            if (fits) c[j] = hv;
+        }
+        if (fits) { prev_ok = 1; prev_row = 0; prev_beg = 0; prev_end = end0; }
+        buf_cur ^= 1;
+        __syncthreads();
+    }
+
+    int cur_pk0 = jb.pre_off[1];
+    int cur_pk1 = jb.pre_off[2 <= n_rows ? 2 : n_rows];
+    int cur_oo0 = jb.out_off[1];
+    int cur_oo1 = jb.out_off[2 <= n_rows ? 2 : n_rows];
+    int cur_remain = jb.max_remain[1];
+    int cur_ml_mem = jb.max_left[1];
+    int cur_mr_mem = jb.max_right[1];
+    int cur_pidx0 = jb.pre_idx[cur_pk0];
+    int cur_ps0 = jb.pre_ps[cur_pk0];
+    int push_ml = 0x7fffffff, push_mr = -0x7fffffff;
+    for (int r = 1; r < n_rows - 1; ++r) {
+        const int pk0 = cur_pk0, pk1 = cur_pk1;
+        const int oo0 = cur_oo0, oo1 = cur_oo1;
+        const int row_remain = cur_remain;
+        const int pidx0 = cur_pidx0;
+        const S ps0 = (S)cur_ps0;
+        const int ml_eff = cur_ml_mem < push_ml ? cur_ml_mem : push_ml;
+        const int mr_eff = cur_mr_mem > push_mr ? cur_mr_mem : push_mr;
+        {
+            const int nr = r + 1;
+            cur_pk0 = pk1;
+            cur_pk1 = jb.pre_off[nr + 1];
+            cur_oo0 = oo1;
+            cur_oo1 = jb.out_off[nr + 1];
+            cur_remain = jb.max_remain[nr];
+            cur_ml_mem = jb.max_left[nr];
+            cur_mr_mem = jb.max_right[nr];
+            cur_pidx0 = jb.pre_idx[cur_pk0];
+            cur_ps0 = jb.pre_ps[cur_pk0];
+        }
+        int beg, end;
+        {
+            int mr = row_remain - end_remain - 1;
+            if (jb.banded) {
+                int lo = ml_eff < qlen - mr ? ml_eff : qlen - mr;
+                beg = lo - w; if (beg < 0) beg = 0;
+                int hi = mr_eff > qlen - mr ? mr_eff : qlen - mr;
+                end = hi + w; if (end > qlen) end = qlen;
+                int min_pre_beg;
+                if (pk1 - pk0 == 1) {
+                    min_pre_beg = (prev_ok && pidx0 == prev_row) ? prev_beg : meta[pidx0].beg;
+                } else {
+                    min_pre_beg = 0x7fffffff;
+                    for (int k = pk0; k < pk1; ++k) {
+                        const int pidx = jb.pre_idx[k];
+                        int pb = (prev_ok && pidx == prev_row) ? prev_beg : meta[pidx].beg;
+                        if (pb < min_pre_beg) min_pre_beg = pb;
+                    }
+                }
+                if (beg < min_pre_beg) beg = min_pre_beg;
+            } else { beg = 0; end = qlen; }
+        }
+        const int64_t bw = end - beg + 1;
+        if (used + bw > jb.arena_cap) { if (tid == 0) res->status = ABAMD_JOB_ARENA_OVERFLOW; return; }
+        const int64_t off = used;
+        if (tid == 0) { meta[r].beg = beg; meta[r].end = end; meta[r].off = off; }
+        used += bw;
+        S *H = arena + off;
+        const uint8_t base = jb.row_base[r];
+        const int *mrow = &mat_lds[base * m];
+        const int cache_fits = bw <= BMW;
+        S *cw = &prev_lds[buf_cur][0];
+        const S *cr = &prev_lds[buf_cur ^ 1][0];
+        const bool fast1 = (pk1 - pk0 == 1) && prev_ok && (pidx0 == prev_row);
+
+        S carry_h = inf_min;                 /* PRE-clamp scan value at ss-1 */
+        S lmax = inf_min; int lleft = -1, lright = -1;
+
+        for (int ss = beg; ss <= end; ss += MWT) {
+            const int j = ss + tid;
+            const bool act = j <= end;
+            const S q = (S)((j == 0 || !act) ? 0 : mrow[query[j - 1]]);
+            S h = inf_min;
+            if (fast1) {
+                if (act) {
+                    if (local_mode && j == 0) { S v = (S)(ps0 + q); if (v > h) h = v; }
+                    if (j - 1 >= prev_beg && j - 1 <= prev_end) {
+                        S v = (S)(cr[j - 1 - prev_beg] + ps0 + q);
+                        if (v > h) h = v;
+                    }
+                    if (j >= prev_beg && j <= prev_end) {
+                        S v = (S)(cr[j - prev_beg] + ps0 - e1);
+                        if (v > h) h = v;
+                    }
+                }
+            } else for (int k = pk0; k < pk1; ++k) {
+                const int p = jb.pre_idx[k];
+                const S ps = (S)jb.pre_ps[k];
+                if (prev_ok && p == prev_row) {
+                    if (act) {
+                        if (local_mode && j == 0) { S v = (S)(ps + q); if (v > h) h = v; }
+                        if (j - 1 >= prev_beg && j - 1 <= prev_end) {
+                            S v = (S)(cr[j - 1 - prev_beg] + ps + q);
+                            if (v > h) h = v;
+                        }
+                        if (j >= prev_beg && j <= prev_end) {
+                            S v = (S)(cr[j - prev_beg] + ps - e1);
+                            if (v > h) h = v;
+                        }
+                    }
+                    continue;
+                }
+                const abamd_row_meta_t pm = meta[p];
+                const S *__restrict__ pH = arena + pm.off;
+                if (act) {
+                    if (local_mode && j == 0) { S v = (S)(ps + q); if (v > h) h = v; }
+                    if (j - 1 >= pm.beg && j - 1 <= pm.end) {
+                        S v = (S)(pH[j - 1 - pm.beg] + ps + q);
+                        if (v > h) h = v;
+                    }
+                    if (j >= pm.beg && j <= pm.end) {
+                        S v = (S)(pH[j - pm.beg] + ps - e1);
+                        if (v > h) h = v;
+                    }
+                }
+            }
+            if (!act) h = inf_min;
+
+            /* insertion scan ON H: per-wave local scan + cross-wave carry
+             * (the merged-candidate + scan form of the one-wave kernel
+             * regroups to exactly this under wrapping arithmetic) */
+            S L1 = scan_maxplus(h, jb.e1, inf_min, lane);
+            if (lane == WAVE - 1) sc_f1[wv] = L1;
+            __syncthreads();
+            S hsc = L1;
+            {
+                S C1 = carry_h;
+                bool havec = ss != beg;
+                for (int ww = 0; ww < wv; ++ww) {
+                    if (havec) C1 = smax(sc_f1[ww], (S)(C1 - (S)(WAVE * jb.e1)));
+                    else { C1 = sc_f1[ww]; havec = true; }
+                }
+                if (havec) hsc = smax(L1, (S)(C1 - (S)((lane + 1) * jb.e1)));
+            }
+            S hfin = local_mode ? smax(hsc, (S)0) : hsc; /* clamp AFTER scan+carry */
+            if (act) {
+                H[j - beg] = hfin;
+                if (cache_fits) cw[j - beg] = hfin;
+                if (hfin > lmax) { lmax = hfin; lleft = j; lright = j; }
+                else if (hfin == lmax) { lright = j; }
+            }
+            const bool more = ss + MWT <= end;
+            if (more && j == ss + MWT - 1) sc_carry[0] = hsc; /* PRE-clamp */
+            if (!more && (jb.banded || local_mode || extend_mode)) {
+                int mvw = wave_red_max_i32((int)lmax);
+                int llw = ((int)lmax == mvw && lleft >= 0) ? lleft : 0x7fffffff;
+                int rrw = ((int)lmax == mvw && lright >= 0) ? lright : -1;
+                llw = wave_red_min_i32(llw);
+                rrw = wave_red_max_i32(rrw);
+                if (lane == 0) {
+                    sc_red[wv] = mvw;
+                    sc_red[MWAVES + wv] = llw;
+                    sc_red[2 * MWAVES + wv] = rrw;
+                }
+            }
+            __syncthreads(); /* also orders cw writes before the next row's reads */
+            if (more) carry_h = sc_carry[0];
+        }
+
+        if (cache_fits) { prev_ok = 1; prev_row = r; prev_beg = beg; prev_end = end; }
+        else prev_ok = 0;
+        buf_cur ^= 1;
+
+        if (jb.banded || local_mode || extend_mode) {
+            int mv = sc_red[0], ll = sc_red[MWAVES], rr = sc_red[2 * MWAVES];
+            #pragma unroll
+            for (int ww = 1; ww < MWAVES; ++ww) {
+                int m2 = sc_red[ww];
+                if (m2 > mv) { mv = m2; ll = sc_red[MWAVES + ww]; rr = sc_red[2 * MWAVES + ww]; }
+                else if (m2 == mv) {
+                    if (sc_red[MWAVES + ww] < ll) ll = sc_red[MWAVES + ww];
+                    if (sc_red[2 * MWAVES + ww] > rr) rr = sc_red[2 * MWAVES + ww];
+                }
+            }
+            if (local_mode) {
+                if (mv > run_best) { run_best = mv; run_best_i = r; run_best_j = ll; }
+            } else if (extend_mode) {
+                if (mv > run_best) {
+                    run_best = mv; run_best_i = r; run_best_j = rr;
+                    run_best_remain = row_remain;
+                } else if (jb.zdrop > 0) {
+                    int delta = run_best_remain - row_remain;
+                    int dd = delta - (rr - run_best_j); if (dd < 0) dd = -dd;
+                    if (run_best - mv > jb.zdrop + jb.e1 * dd) zdropped = 1;
+                }
+            }
+            push_ml = 0x7fffffff; push_mr = -0x7fffffff;
+            if (!zdropped && jb.banded) {
+                for (int k = oo0; k < oo1; ++k) {
+                    int o = jb.out_idx[k];
+                    if (o == r + 1) {
+                        if (rr + 1 > push_mr) push_mr = rr + 1;
+                        if (ll + 1 < push_ml) push_ml = ll + 1;
+                    } else {
+                        if (rr + 1 > jb.max_right[o]) jb.max_right[o] = rr + 1;
+                        if (ll + 1 < jb.max_left[o]) jb.max_left[o] = ll + 1;
+                    }
+                }
+            }
+            if (zdropped) break;
+        } else {
+            push_ml = 0x7fffffff; push_mr = -0x7fffffff;
+        }
+    }
+
+    __syncthreads();
+    if (tid == 0) res->cells = used;
+    if (tid != 0) return;
+
+    int32_t best_score = run_best;
+    int best_i = run_best_i, best_j = run_best_j;
+    if (jb.align_mode == 0) {
+        best_score = jb.inf_min; best_i = 0; best_j = 0;
+        for (int k = jb.pre_off[n_rows - 1]; k < jb.pre_off[n_rows]; ++k) {
+            const int p = jb.pre_idx[k];
+            const abamd_row_meta_t pm = meta[p];
+            int e = pm.end < qlen ? pm.end : qlen;
+            const S *pH = arena + pm.off;
+            int32_t sc = (e >= pm.beg) ? (int32_t)pH[e - pm.beg] : jb.inf_min;
+            if (sc > best_score) { best_score = sc; best_i = p; best_j = e; }
+        }
+    }
+    res->best_score = best_score;
+    res->best_i = best_i; res->best_j = best_j;
+    if (!jb.ret_cigar) return;
+
+    { /* simd_abpoa_lg_backtrack (:116-194) */
+        int bi = best_i, bj = best_j, start_i = best_i, start_j = best_j;
+        int n_c = 0, status = ABAMD_JOB_OK;
+        int look_end = jb.put_gap_at_end, put_right = jb.put_gap_on_right;
+        int n_aln = 0, n_matched = 0;
+        uint64_t *cig = jb.cigar;
+        int id = jb.row_node_id[bi];
+        if (best_j < qlen) dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, qlen - best_j, -1, qlen - 1, &status);
+        while (bi > 0 && bj > 0 && status == ABAMD_JOB_OK) {
+            const abamd_row_meta_t bm = meta[bi];
+            const int rb = bm.beg, re = bm.end;
+            const S *H = arena + bm.off;
+            const S Hj = (bj >= rb && bj <= re) ? H[bj - rb] : inf_min;
+            const S Hjm1 = (bj - 1 >= rb && bj - 1 <= re) ? H[bj - 1 - rb] : inf_min;
+            if (local_mode && Hj == 0) break;
+            start_i = bi; start_j = bj;
+            const int pq0 = jb.pre_off[bi], pq1 = jb.pre_off[bi + 1];
+            const S s = (S)mat_lds[m * jb.row_base[bi] + query[bj - 1]];
+            const int is_match = jb.row_base[bi] == query[bj - 1];
+            int hit = 0;
+            if (put_right == 0 && look_end == 0) {
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
+                    const S *pH = arena + pm.off;
+                    if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
+                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
+                        bi = p; --bj; id = jb.row_node_id[bi]; hit = 1;
+                        ++n_aln; n_matched += is_match;
+                        break;
+                    }
+                }
+            }
+            if (!hit) { /* deletion */
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj < pm.beg || bj > pm.end) continue;
+                    const S *pH = arena + pm.off;
+                    if ((S)(pH[bj - pm.beg] - e1 + ps) == Hj) {
+                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 2, 1, id, bj - 1, &status);
+                        bi = p; id = jb.row_node_id[bi]; hit = 1;
+                        if (look_end) look_end = 0;
+                        break;
+                    }
+                }
+            }
+            if (!hit) { /* insertion */
+                if ((S)(Hjm1 - e1) == Hj) {
+                    dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, 1, id, bj - 1, &status);
+                    --bj;
+                    if (look_end) look_end = 0;
+                    hit = 1; ++n_aln;
+                }
+            }
+            if (!hit) {
+                for (int k = pq0; k < pq1; ++k) {
+                    const int p = jb.pre_idx[k];
+                    const S ps = (S)jb.pre_ps[k];
+                    const abamd_row_meta_t pm = meta[p];
+                    if (bj - 1 < pm.beg || bj - 1 > pm.end) continue;
+                    const S *pH = arena + pm.off;
+                    if ((S)(pH[bj - 1 - pm.beg] + s + ps) == Hj) {
+                        dev_push_cigar(cig, &n_c, jb.cigar_cap, 0, 1, id, bj - 1, &status);
+                        bi = p; --bj; id = jb.row_node_id[bi]; hit = 1;
+                        ++n_aln; n_matched += is_match;
+                        look_end = 0;
+                        break;
+                    }
+                }
+            }
+            if (!hit) { status = ABAMD_JOB_BT_DEAD_END; break; }
+        }
+        if (status == ABAMD_JOB_OK && bj > 0)
+            dev_push_cigar(cig, &n_c, jb.cigar_cap, 1, bj, -1, bj - 1, &status);
+        res->status = status;
+        res->n_cigar = n_c;
+        res->n_aln_bases = n_aln;
+        res->n_matched_bases = n_matched;
+        res->node_e = jb.row_node_id[best_i]; res->query_e = best_j - 1;
+        res->node_s = jb.row_node_id[start_i]; res->query_s = start_j - 1;
+    }
+}
+
+/* one-wave ag/lg kernels kept as an A/B fallback (ABPOA_AMD_SW_AGLG=1) */
+static int use_sw_aglg(void) {
+    static int v = -1;
+    if (v < 0) v = getenv("ABPOA_AMD_SW_AGLG") != nullptr;
+    return v;
+}
+
 extern "C" void abamd_launch_ag_i16(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
                                     int n_jobs, void *stream) {
-    int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
-    hipLaunchKernelGGL((ag_global_kernel<int16_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
+    if (use_sw_aglg()) {
+        int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
+        hipLaunchKernelGGL((ag_global_kernel<int16_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
+                           (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
+        return;
+    }
+    hipLaunchKernelGGL((ag_global_mw_kernel<int16_t>), dim3(n_jobs), dim3(MWT), 0,
                        (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
 }
 extern "C" void abamd_launch_ag_i32(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
                                     int n_jobs, void *stream) {
-    int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
-    hipLaunchKernelGGL((ag_global_kernel<int32_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
+    if (use_sw_aglg()) {
+        int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
+        hipLaunchKernelGGL((ag_global_kernel<int32_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
+                           (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
+        return;
+    }
+    hipLaunchKernelGGL((ag_global_mw_kernel<int32_t>), dim3(n_jobs), dim3(MWT), 0,
                        (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
 }
 extern "C" void abamd_launch_lg_i16(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
                                     int n_jobs, void *stream) {
-    int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
-    hipLaunchKernelGGL((lg_global_kernel<int16_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
+    if (use_sw_aglg()) {
+        int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
+        hipLaunchKernelGGL((lg_global_kernel<int16_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
+                           (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
+        return;
+    }
+    hipLaunchKernelGGL((lg_global_mw_kernel<int16_t>), dim3(n_jobs), dim3(MWT), 0,
                        (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
 }
 extern "C" void abamd_launch_lg_i32(const abamd_gpu_job_t *dev_jobs, abamd_gpu_res_t *dev_res,
                                     int n_jobs, void *stream) {
-    int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
-    hipLaunchKernelGGL((lg_global_kernel<int32_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
+    if (use_sw_aglg()) {
+        int blocks = (n_jobs + JOBS_PER_BLOCK - 1) / JOBS_PER_BLOCK;
+        hipLaunchKernelGGL((lg_global_kernel<int32_t>), dim3(blocks), dim3(WAVE * JOBS_PER_BLOCK), 0,
+                           (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
+        return;
+    }
+    hipLaunchKernelGGL((lg_global_mw_kernel<int32_t>), dim3(n_jobs), dim3(MWT), 0,
                        (hipStream_t)stream, dev_jobs, dev_res, n_jobs);
 }
