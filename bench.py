@@ -9,11 +9,15 @@ aggregate over all ranks (one process per GPU; sets shard across ranks with
 no data-path collective: weak scaling).
 
 Emits ONE JSON line from rank 0, including:
-  roofline:     dominant kernel (cg_global_kernel) algorithmic HBM bytes/s
-                measured with HIP events on the library stream (10 B per DP
-                cell: 5 int16 planes written once, re-read once by backtrack)
-  cpu_baseline: the unmodified reference binary (oracle/_ref/abpoa) timed on
-                this box's host cores on a bounded sample.
+  roofline:     dominant kernel (cg_global_mw_kernel) algorithmic HBM bytes/s
+                from HIP events on the library streams (6 B per DP cell at
+                int16: 3 planes H/E1/E2 written once; F planes recomputed at
+                backtrack), plus counter-measured traffic from two separate
+                rocprofv3 --pmc probe passes (gfx950 FETCH correction)
+  cpu_baseline: the unmodified reference binary (oracle/_ref/abpoa) on this
+                box's host cores: 1-core measured, one-socket linear
+                extrapolation, and (ABPOA_BENCH_CPU_SOCKET) a measured
+                concurrent batch run.
 """
 import argparse
 import ctypes

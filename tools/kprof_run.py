@@ -42,18 +42,6 @@ def main():
     k = fetch(L, "abamd_kprof_fetch")
     f = fetch(L, "abamd_kprof_fold_fetch")
 
-    rows, chunks, band, chunk, epi, bt, _, jobs = k[:8]
-    kg, ks, kw = k[8:11]
-    tot = band + chunk + epi
-    print("== cg DP kernel (per lane-0 job; %d jobs, %d rows, %.1f chunks/row) ==" %
-          (jobs, rows, chunks / max(rows, 1)))
-    for name, v in (("band+prefetch", band), ("chunk loop", chunk), ("epilogue", epi)):
-        print("  %-14s %12d cyc  %5.1f%%  (%.0f cyc/row)" %
-              (name, v, 100.0 * v / max(tot, 1), v / max(rows, 1)))
-    print("  %-14s %12d cyc  (%.0f cyc/chunk in-loop)" %
-          ("backtrack", bt, chunk / max(chunks, 1)))
-    for name, v in (("  gather+query", kg), ("  shfl+Fscan", ks), ("  fold+stores", kw)):
-        print("  %-14s %12d cyc  (%.0f cyc/chunk)" % (name, v, v / max(chunks, 1)))
     mwa, mwb, mwc, mwtot, mwrows = k[11], k[12], k[13], k[14], k[15]
     if mwrows:
         print("== cg MULTI-WAVE kernel (%d rows, %.0f cyc/row total) ==" %
