@@ -179,6 +179,53 @@ int main(int argc, char **argv) {
                     die("max_remain", i, id);
                 if (id == ABPOA_SINK_NODE_ID) break;
             }
+            {   /* flat -> pointer import (the resident driver's consensus
+                 * hand-off, abamd_graph_from_flat): must reproduce the live
+                 * graph field-for-field, topo arrays included */
+                abpoa_t *ab2 = abpoa_init();
+                void *slab = abamd_graph_from_flat(ab2, &fg, abpt, rid_n);
+                abpoa_graph_t *ga = ab->abg, *gb = ab2->abg;
+                int id2, k2, sink_idx = ga->node_id_to_index[ABPOA_SINK_NODE_ID];
+                if (ga->node_n != gb->node_n) die("import node_n", i, -1);
+                for (id2 = 0; id2 < ga->node_n; ++id2) {
+                    abpoa_node_t *va = &ga->node[id2], *vb = &gb->node[id2];
+                    if (va->base != vb->base || va->n_read != vb->n_read ||
+                        va->n_span_read != vb->n_span_read) die("import node fields", i, id2);
+                    if (va->in_edge_n != vb->in_edge_n || va->out_edge_n != vb->out_edge_n ||
+                        va->aligned_node_n != vb->aligned_node_n) die("import counts", i, id2);
+                    for (k2 = 0; k2 < va->in_edge_n; ++k2)
+                        if (va->in_id[k2] != vb->in_id[k2] ||
+                            va->in_edge_weight[k2] != vb->in_edge_weight[k2])
+                            die("import in edges", i, id2);
+                    for (k2 = 0; k2 < va->out_edge_n; ++k2) {
+                        if (va->out_id[k2] != vb->out_id[k2] ||
+                            va->out_edge_weight[k2] != vb->out_edge_weight[k2])
+                            die("import out edges", i, id2);
+                        if (rid_n > 0 && va->read_ids_n > 0) {
+                            int k3;
+                            if (vb->read_ids_n != rid_n) die("import read_ids_n", i, id2);
+                            for (k3 = 0; k3 < rid_n; ++k3)
+                                if (va->read_ids[k2][k3] != vb->read_ids[k2][k3])
+                                    die("import read bitsets", i, id2);
+                        }
+                    }
+                    for (k2 = 0; k2 < va->aligned_node_n; ++k2)
+                        if (va->aligned_node_id[k2] != vb->aligned_node_id[k2])
+                            die("import aligned", i, id2);
+                }
+                for (k2 = 0; k2 <= sink_idx; ++k2) {
+                    if (ga->index_to_node_id[k2] != gb->index_to_node_id[k2])
+                        die("import topo", i, k2);
+                    id2 = ga->index_to_node_id[k2];
+                    if (ga->node_id_to_index[id2] != gb->node_id_to_index[id2])
+                        die("import n2i", i, id2);
+                    if (ga->node_id_to_max_remain && gb->node_id_to_max_remain &&
+                        ga->node_id_to_max_remain[id2] != gb->node_id_to_max_remain[id2])
+                        die("import remain", i, id2);
+                }
+                abamd_graph_arena_release(ab2, slab);
+                abpoa_free(ab2);
+            }
             free(i2n); free(n2i); free(rem); free(scr);
         }
     }
