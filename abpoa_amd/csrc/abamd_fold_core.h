@@ -56,7 +56,8 @@ void abamd_flat_apply_alignment(flat_graph_t *fg, int beg_node_id, int end_node_
  * path (abamd_graph.c). `ab` must be fresh. All node arrays are carved from
  * ONE returned slab; call abamd_graph_arena_release(ab, slab) before
  * abpoa_free(ab). */
-void *abamd_graph_from_flat(abpoa_t *ab, const flat_graph_t *fg, abpoa_para_t *abpt, int read_ids_n);
+void *abamd_graph_from_flat(abpoa_t *ab, const flat_graph_t *fg, abpoa_para_t *abpt, int read_ids_n,
+                            const int *i2n, const int *n2i, const int *remain);
 void abamd_graph_arena_release(abpoa_t *ab, void *slab);
 #endif
 

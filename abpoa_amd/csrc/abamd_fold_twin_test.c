@@ -183,7 +183,7 @@ int main(int argc, char **argv) {
                  * hand-off, abamd_graph_from_flat): must reproduce the live
                  * graph field-for-field, topo arrays included */
                 abpoa_t *ab2 = abpoa_init();
-                void *slab = abamd_graph_from_flat(ab2, &fg, abpt, rid_n);
+                void *slab = abamd_graph_from_flat(ab2, &fg, abpt, rid_n, i2n, n2i, rem);
                 abpoa_graph_t *ga = ab->abg, *gb = ab2->abg;
                 int id2, k2, sink_idx = ga->node_id_to_index[ABPOA_SINK_NODE_ID];
                 if (ga->node_n != gb->node_n) die("import node_n", i, -1);

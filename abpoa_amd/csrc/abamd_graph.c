@@ -675,7 +675,8 @@ void abpoa_subgraph_nodes(abpoa_t *ab, abpoa_para_t *abpt, int inc_beg, int inc_
  * copies per-out-edge read bitsets (fg->rid_n must equal read_ids_n). */
 #include "abamd_fold_core.h"
 
-void *abamd_graph_from_flat(abpoa_t *ab, const flat_graph_t *fg, abpoa_para_t *abpt, int read_ids_n) {
+void *abamd_graph_from_flat(abpoa_t *ab, const flat_graph_t *fg, abpoa_para_t *abpt, int read_ids_n,
+                            const int *i2n, const int *n2i, const int *remain) {
     abpoa_graph_t *g = ab->abg;
     int id, e;
     if (read_ids_n > 0 && fg->rid_n != read_ids_n)
@@ -757,6 +758,18 @@ void *abamd_graph_from_flat(abpoa_t *ab, const flat_graph_t *fg, abpoa_para_t *a
     g->node_m = fg->node_n;
     g->is_topological_sorted = g->is_called_cons = g->is_set_msa_rank = 0;
     abpoa_topological_sort(g, abpt);
+    if (i2n && n2i) {
+        /* The reference's index arrays come from the BFS over the PRE-sort
+         * adjacency of the LAST fold; the re-derivation above ran over the
+         * already-sorted adjacency, which can order ready groups
+         * differently. Restore the faithful arrays the device computed in
+         * the live pass order (the sort itself is idempotent, so the
+         * adjacency state is untouched). */
+        memcpy(g->index_to_node_id, i2n, (size_t)fg->node_n * sizeof(int));
+        memcpy(g->node_id_to_index, n2i, (size_t)fg->node_n * sizeof(int));
+        if (remain && g->node_id_to_max_remain)
+            memcpy(g->node_id_to_max_remain, remain, (size_t)fg->node_n * sizeof(int));
+    }
     return slab;
 }
 

@@ -994,6 +994,10 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
     /* ---- download graphs, rebuild pointer graphs, consensus ---- */
     double t_cons0 = abamd_realtime();
     std::vector<flat_graph_t> hflat(n_sets);
+    /* device-derived index/remain arrays travel with the graph: they carry
+     * the reference's BFS-on-pre-sort-adjacency order, which a host
+     * re-derivation over the (sorted) imported adjacency cannot reproduce */
+    std::vector<int*> hi2n(n_sets), hn2i(n_sets), hrem(n_sets);
     for (int s = 0; s < n_sets; ++s) {
         const flat_graph_t &d = B.sets[s].g;
         flat_graph_t &h = hflat[s];
@@ -1033,6 +1037,13 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
         if (d.rid_n > 0) d2h(h.rid_pool, d.rid_pool, 8 * (size_t)d.edge_n_out * d.rid_n);
         d2h(h.aln_id, d.aln_id, 4 * (size_t)d.aln_n);
         d2h(h.aln_next, d.aln_next, 4 * (size_t)d.aln_n);
+        hi2n[s] = (int*)abamd_malloc(4 * nn + 4);
+        hn2i[s] = (int*)abamd_malloc(4 * nn + 4);
+        hrem[s] = (int*)abamd_malloc(4 * nn + 4);
+        d2h(hi2n[s], B.sets[s].i2n, 4 * nn);
+        d2h(hn2i[s], B.sets[s].n2i, 4 * nn);
+        if (B.use_remain) d2h(hrem[s], B.sets[s].rem, 4 * nn);
+        else memset(hrem[s], 0, 4 * nn + 4);
     }
     RHIP_CHECK(hipStreamSynchronize(S0.stream));
     double t_dl = abamd_realtime() - t_cons0;
@@ -1041,11 +1052,13 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
         std::atomic<int> next{0};
         Batch *B;
         std::vector<flat_graph_t> *hflat;
+        std::vector<int*> *hi2n, *hn2i, *hrem;
         std::vector<abpoa_t*> abs;
         std::vector<void*> slabs;
         abpoa_para_t *abpt;
     } cc;
     cc.B = &B; cc.hflat = &hflat; cc.abpt = abpt;
+    cc.hi2n = &hi2n; cc.hn2i = &hn2i; cc.hrem = &hrem;
     cc.abs.assign(n_sets, nullptr);
     cc.slabs.assign(n_sets, nullptr);
     auto cons_worker = [](void *p, int, int) {
@@ -1054,7 +1067,8 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
             int s = c.next.fetch_add(1);
             if (s >= c.B->n_sets) break;
             abpoa_t *ab = abpoa_init();
-            c.slabs[s] = abamd_graph_from_flat(ab, &(*c.hflat)[s], c.abpt, (*c.hflat)[s].rid_n);
+            c.slabs[s] = abamd_graph_from_flat(ab, &(*c.hflat)[s], c.abpt, (*c.hflat)[s].rid_n,
+                                               (*c.hi2n)[s], (*c.hn2i)[s], (*c.hrem)[s]);
             ab->abs->n_seq = c.B->sets[s].n_seqs;
             abpoa_generate_consensus(ab, c.abpt);
             c.abs[s] = ab;
@@ -1072,6 +1086,7 @@ extern "C" int abpoa_amd_msa_batch_resident(abpoa_para_t *abpt, int n_sets, cons
         free(h.in_to); free(h.in_w); free(h.in_next);
         free(h.out_to); free(h.out_w); free(h.out_next);
         free(h.rid_pool); free(h.aln_id); free(h.aln_next);
+        free(hi2n[s]); free(hn2i[s]); free(hrem[s]);
     }
     for (int s = 0; s < n_sets; ++s)
         if (B.sets[s].own_slab) RHIP_CHECK(hipFree(B.sets[s].own_slab));
