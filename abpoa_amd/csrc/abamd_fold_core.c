@@ -18,6 +18,7 @@
 #include <string.h>
 #include "abpoa_amd.h"
 #include "abamd_util.h"
+#include <math.h>
 #include "abamd_fold_core.h"
 
 void abamd_flat_init(flat_graph_t *fg, int node_cap, int edge_cap, int aln_cap, int rid_n) {
@@ -57,3 +58,4 @@ void abamd_flat_free(flat_graph_t *fg) {
 
 
 #include "abamd_fold_core.inc"
+#include "abamd_cons_core.inc"

@@ -50,6 +50,13 @@ void abamd_flat_apply_alignment(flat_graph_t *fg, int beg_node_id, int end_node_
                                 int *qpos_to_node_id, int n_cigar, const abpoa_cigar_t *cig,
                                 int read_id, int add_read_id, int inc_both_ends);
 
+/* single-cluster heaviest-bundle consensus over the flat layout
+ * (abamd_cons_core.inc); returns cons_len */
+int abamd_flat_hb_consensus(const flat_graph_t *fg, int n_seq,
+                            int *scratch, int *score, int *max_out,
+                            int *cons_id, uint8_t *cons_base,
+                            int *cons_cov, int *cons_phred);
+
 #ifdef ABPOA_AMD_H
 /* Rebuild a pointer graph from a (host copy of a) flat graph and topo-sort
  * it — the device-resident batch driver's hand-off to the host consensus

@@ -226,6 +226,34 @@ int main(int argc, char **argv) {
                 abamd_graph_arena_release(ab2, slab);
                 abpoa_free(ab2);
             }
+            {   /* flat heaviest-bundle consensus (abamd_cons_core.inc, the
+                 * round-3 device-consensus core) vs the live pointer-graph
+                 * consensus on the same state */
+                int *score = (int*)abamd_malloc((size_t)n * sizeof(int));
+                int *mout = (int*)abamd_malloc((size_t)n * sizeof(int));
+                int *cid = (int*)abamd_malloc((size_t)n * sizeof(int));
+                uint8_t *cb = (uint8_t*)abamd_malloc((size_t)n);
+                int *cc = (int*)abamd_malloc((size_t)n * sizeof(int));
+                int *cp = (int*)abamd_malloc((size_t)n * sizeof(int));
+                int clen = abamd_flat_hb_consensus(&fg, n_seq, scr, score, mout,
+                                                   cid, cb, cc, cp);
+                abpoa_clean_msa_cons(ab);
+                ab->abg->is_called_cons = 0;
+                abpoa_generate_consensus(ab, abpt);
+                abpoa_cons_t *abc = ab->abc;
+                int k4;
+                if (abc->n_cons != 1 || abc->cons_len[0] != clen)
+                    die("hb cons length", i, clen);
+                for (k4 = 0; k4 < clen; ++k4) {
+                    if (cid[k4] != abc->cons_node_ids[0][k4]) die("hb cons node", i, k4);
+                    if (cb[k4] != abc->cons_base[0][k4]) die("hb cons base", i, k4);
+                    if (cc[k4] != abc->cons_cov[0][k4]) die("hb cons cov", i, k4);
+                    if (cp[k4] != abc->cons_phred_score[0][k4]) die("hb cons phred", i, k4);
+                }
+                abpoa_clean_msa_cons(ab);
+                ab->abg->is_called_cons = 0;
+                free(score); free(mout); free(cid); free(cb); free(cc); free(cp);
+            }
             free(i2n); free(n2i); free(rem); free(scr);
         }
     }

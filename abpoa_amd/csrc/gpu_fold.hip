@@ -42,6 +42,7 @@ typedef uint64_t abpoa_cigar_t;
 #define ABAMD_FC_FAIL(msg) do { abort(); } while (0)
 #define ABAMD_FC_FAIL_RET(val, msg) do { abort(); } while (0)
 #include "abamd_fold_core.inc"
+#include "abamd_cons_core.inc"
 
 /* ------------------------------------------------------------------ */
 /* Round-1 validation kernel (kept: the GPU twin test drives it).      */
