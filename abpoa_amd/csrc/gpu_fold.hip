@@ -476,6 +476,33 @@ void abamd_fold_round_kernel(abamd_fold_round_job_t *jobs, int n_jobs) {
     }
 }
 
+
+/* Round-3 skeleton: per-set single-cluster consensus on device (compile-
+ * checked; not wired — the resident driver still downloads graphs and runs
+ * host consensus). One block per set, the twin-proven core on lane 0;
+ * phred parity (double pow/log10 on device libm) must be validated on
+ * hardware before wiring. */
+typedef struct {
+    flat_graph_t g;
+    int n_seq;
+    int *scratch, *score, *max_out;
+    int *cons_id; uint8_t *cons_base; int *cons_cov, *cons_phred;
+    int *cons_len_out;
+} abamd_cons_job_t;
+
+extern "C" __global__ void abamd_cons_kernel(abamd_cons_job_t *jobs, int n_jobs) {
+    int j = blockIdx.x;
+    if (j >= n_jobs) return;
+    abamd_cons_job_t *job = &jobs[j];
+    if (threadIdx.x == 0) {
+        flat_graph_t gl = job->g;
+        *job->cons_len_out = abamd_flat_hb_consensus(&gl, job->n_seq,
+                                                     job->scratch, job->score, job->max_out,
+                                                     job->cons_id, job->cons_base,
+                                                     job->cons_cov, job->cons_phred);
+    }
+}
+
 extern "C" void abamd_launch_fold_round(const abamd_fold_round_job_t *dev_jobs,
                                         int n_jobs, void *stream) {
     hipLaunchKernelGGL(abamd_fold_round_kernel, dim3(n_jobs), dim3(64), 0,

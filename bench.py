@@ -263,14 +263,18 @@ def traffic_probe(args):
             env["ABPOA_BENCH_NO_TRAFFIC"] = "1"
             env["ABPOA_BENCH_SKIP_CPU"] = "1"
             odir = os.path.join(td, pmc)
-            r = subprocess.run(
-                ["rocprofv3", "--pmc", pmc, "--output-format", "csv",
-                 "-d", odir, "-o", "probe", "--",
-                 sys.executable, os.path.abspath(__file__),
-                 "--sets-per-step", str(n_probe_sets), "--steps", "1", "--warmup", "0",
-                 "--depth", str(args.depth), "--qlen", str(args.qlen),
-                 "--workload", args.workload],
-                cwd="/tmp", env=env, stdout=subprocess.PIPE, stderr=subprocess.DEVNULL)
+            try:
+                r = subprocess.run(
+                    ["rocprofv3", "--pmc", pmc, "--output-format", "csv",
+                     "-d", odir, "-o", "probe", "--",
+                     sys.executable, os.path.abspath(__file__),
+                     "--sets-per-step", str(n_probe_sets), "--steps", "1", "--warmup", "0",
+                     "--depth", str(args.depth), "--qlen", str(args.qlen),
+                     "--workload", args.workload],
+                    cwd="/tmp", env=env, stdout=subprocess.PIPE,
+                    stderr=subprocess.DEVNULL, timeout=420)
+            except (OSError, subprocess.TimeoutExpired):
+                return None
             if r.returncode != 0:
                 return None
             probe_cells = 0
@@ -430,7 +434,10 @@ def main():
         except AttributeError:
             dp_s = kns / 1e9; fold_s = 0.0; busy_s = None
         ach = alg_bytes / max(kns, 1)  # bytes/ns == GB/s (per summed DP event time)
-        traffic = traffic_probe(args) if world == 1 else None
+        try:
+            traffic = traffic_probe(args) if world == 1 else None
+        except Exception:
+            traffic = None  # traffic is evidence, never a reason to fail the run
         tval = None
         if traffic:
             # counter-measured bytes/cell (probe) x this run's cells/launch
