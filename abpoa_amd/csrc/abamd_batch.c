@@ -1,12 +1,16 @@
-/* Batched multi-set POA driver.
+/* Batched multi-set POA driver — HOST-FOLD FALLBACK + dispatcher.
  *
  * The reference processes one read set at a time (abpoa_poa,
- * abpoa_align.c:313-353); a single alignment cannot fill a 256-CU GPU, so the
- * MI355X-native driver advances MANY independent read sets in lockstep
- * "rounds": in round r every unfinished set aligns its r-th read against its
- * own graph in ONE batched kernel launch (thousands of wavefronts), then the
- * CIGAR folds + topological re-sorts run on host threads. Per-set results are
- * identical to the sequential path: sets are fully independent
+ * abpoa_align.c:313-353); a single alignment cannot fill a 256-CU GPU, so
+ * the MI355X-native driver advances MANY independent read sets in lockstep
+ * "rounds". The PRODUCT path is the device-resident driver
+ * (gpu_batch_resident.cpp: graphs live in HBM, the fold kernel consumes
+ * CIGARs in place); abpoa_amd_msa_batch dispatches to it whenever the
+ * configuration allows. THIS file keeps the round-1 host-fold pipeline as
+ * the fallback (configs the flat pools don't carry: inc_path_score,
+ * per-read quality weights; ABPOA_AMD_HOST_FOLD=1 forces it for A/B) and
+ * as the CPU-test route (gpu_stub builds always land here). Per-set
+ * results are identical on every path: sets are fully independent
  * (SURVEY.md §5/§8e).
  */
 #include <pthread.h>

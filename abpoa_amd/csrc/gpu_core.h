@@ -4,15 +4,19 @@
  * the topologically-sorted reachable rows of the partial order graph
  * (reference semantics: abpoa_align_simd.c; clean cell-granularity
  * formulation as in oracle/ref_core.c — the two are checked bit-equal by
- * tests/). Rows are COMPACTED on the host: only reachable (index_map) rows
- * travel, with predecessor lists remapped to compact row indices.
+ * tests/). Row CSRs come either from the host pack (seam path: only
+ * reachable index_map rows travel, predecessor lists remapped to compact
+ * row indices) or straight from the device fold (resident batch path).
  *
- * Device execution: one workgroup of 64 threads (one wavefront) per job walks
- * the rows in topological order; the band of each row is computed on device
- * from the adaptive band state (max_left/max_right/max_remain), scores for
- * the band are kept in registers in 64-cell chunks, the F (insertion) chain
- * uses a wavefront log-scan, and the five banded planes (H,E1,E2,F1,F2)
- * stream to the HBM arena for the backtrack. Lane 0 then backtracks in-kernel.
+ * Device execution (round 2): ONE 512-THREAD BLOCK (8 wavefronts) per job
+ * walks the rows in topological order; the band of each row is computed on
+ * device from the adaptive band state (max_left/max_right/max_remain) with
+ * the per-row scalars rolled one row ahead, all 8 waves cover the band at
+ * once (per-wave DPP F-scans + a cross-wave carry recurrence), and the
+ * banded planes stream to the HBM arena for the backtrack (convex: 3
+ * planes H/E1/E2, F recomputed at backtrack; affine: H/E1/F1; linear: H).
+ * Thread 0 backtracks in-kernel, pausing for block-wide F-window
+ * recomputes on the convex path.
  */
 #ifndef ABAMD_GPU_CORE_H
 #define ABAMD_GPU_CORE_H

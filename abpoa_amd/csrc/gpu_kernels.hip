@@ -11,12 +11,13 @@
  *     HBM arena with coalesced 2/4-byte loads.
  *   - the F (insertion) recurrence is a wavefront max-plus log-scan
  *     (__shfl_up, 6 doubling steps) with a sequential carry between chunks,
- *   - the five planes (H,E1,E2,F1,F2) stream to the HBM arena with coalesced
- *     stores (re-read once by the in-kernel backtrack),
+ *   - the banded planes stream to the HBM arena with coalesced stores
+ *     (convex: H,E1,E2 only — the F planes are recomputed from stored H at
+ *     backtrack time; affine: H,E1,F1; linear: H),
  *   - the row argmax (adaptive band steering) is a wavefront reduction,
  *   - per-row band metadata is one 16-byte abamd_row_meta_t load.
  * Integer max-plus throughout: MFMA does not apply; the target bound is the
- * 10 B/cell HBM plane traffic.
+ * 6 B/cell (convex int16) HBM plane traffic.
  *
  * Numerics are bit-identical to oracle/ref_core.c (and the reference x86
  * build): int16 arithmetic wraps (non-saturating _mm*_add_epi16 semantics),

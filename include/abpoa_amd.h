@@ -217,8 +217,9 @@ int abpoa_amd_msa_batch(abpoa_para_t *abpt, int n_sets, const int *n_seqs,
  * total DP cells computed by the device core and total device-kernel
  * nanoseconds measured with HIP events on the library's stream. */
 void abpoa_amd_get_stats(uint64_t *dp_cells, uint64_t *kernel_ns, uint64_t *n_launches);
-/* algorithmic HBM bytes of the plane streams (5 planes x score width x cells),
- * summed per launch at the launch's actual score width */
+/* algorithmic HBM bytes of the plane streams (stored planes x score width x
+ * cells: convex 3, affine 3, linear 1), summed per launch at the launch's
+ * actual score width */
 void abpoa_amd_get_stats2(uint64_t *alg_bytes);
 void abpoa_amd_reset_stats(void);
 
