@@ -56,6 +56,26 @@ def test_resident_matches_hostfold():
         assert resident == hostfold, "resident/host-fold mismatch (cons_algrm=%s)" % algrm
 
 
+def test_resident_multiword_readids():
+    """Depth-80 MF-consensus sets: per-edge read-id bitsets span TWO 64-bit
+    words (rid_n = 2) in the device-resident pools — byte-compare against
+    the host-fold driver."""
+    sys.path.insert(0, ROOT)
+    import abpoa_amd
+    import importlib
+    import numpy as np
+    bench = importlib.import_module("bench")
+    rng = np.random.default_rng(88)
+    sets = bench.gen_sets(rng, 3, depth=80, qlen=400)
+    resident = abpoa_amd.msa_batch_consensus(sets, n_threads=2, cons_algrm="MF")
+    os.environ["ABPOA_AMD_HOST_FOLD"] = "1"
+    try:
+        hostfold = abpoa_amd.msa_batch_consensus(sets, n_threads=2, cons_algrm="MF")
+    finally:
+        os.environ.pop("ABPOA_AMD_HOST_FOLD", None)
+    assert resident == hostfold, "multi-word read-id bitset divergence"
+
+
 def test_resident_pool_expansion():
     """Tiny ABPOA_AMD_NODE_ALPHA forces graph-pool overflows mid-run: the
     expand + device-to-device move + refold path must still be byte-exact."""

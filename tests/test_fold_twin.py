@@ -30,7 +30,9 @@ def test_twin_golden():
 
 
 def test_twin_synthetic(tmp_path):
-    for seed, length, depth in ((1, 300, 12), (2, 900, 20), (3, 1800, 30), (4, 4000, 40)):
+    # depth 80 exercises multi-word per-edge read-id bitsets (rid_n = 2)
+    for seed, length, depth in ((1, 300, 12), (2, 900, 20), (3, 1800, 30), (4, 4000, 40),
+                                (5, 600, 80)):
         fa = str(tmp_path / ("s%d.fa" % seed))
         subprocess.run(["python3", os.path.join(ROOT, "tests", "make_synth.py"), fa,
                         "--seed", str(seed), "--len", str(length), "--depth", str(depth)],
